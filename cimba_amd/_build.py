@@ -1,0 +1,102 @@
+"""In-tree build of the cimba_amd native extension (hipcc, gfx950).
+
+The extension is built IN-TREE (cimba_amd/_C*.so) so the built artifact
+travels with repo snapshots to GPU boxes.  hipcc cross-compiles gfx950
+device code on CPU-only hosts; no GPU is needed to build.
+"""
+import os
+import subprocess
+import sys
+import sysconfig
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(ROOT, "csrc")
+GFX_ARCH = os.environ.get("CIMBA_GFX_ARCH", "gfx950")
+
+SOURCES = [
+    os.path.join(CSRC, "host", "support.cpp"),
+    os.path.join(CSRC, "bindings.cpp"),
+    os.path.join(CSRC, "hip", "deskernel.hip"),
+]
+
+
+def _hipcc():
+    for cand in (os.environ.get("HIPCC"), "/opt/rocm/bin/hipcc", "hipcc"):
+        if not cand:
+            continue
+        if os.path.sep in cand:
+            if os.path.exists(cand):
+                return cand
+        else:
+            from shutil import which
+            w = which(cand)
+            if w:
+                return w
+    raise RuntimeError("hipcc not found")
+
+
+def ext_path():
+    suffix = sysconfig.get_config_var("EXT_SUFFIX") or ".so"
+    return os.path.join(ROOT, "_C" + suffix)
+
+
+def _sources_mtime():
+    latest = 0.0
+    for base, _dirs, files in os.walk(CSRC):
+        for f in files:
+            if f.endswith((".cpp", ".hpp", ".hip", ".h")):
+                latest = max(latest, os.path.getmtime(os.path.join(base, f)))
+    latest = max(latest, os.path.getmtime(os.path.abspath(__file__)))
+    return latest
+
+
+def needs_build():
+    so = ext_path()
+    if not os.path.exists(so):
+        return True
+    return os.path.getmtime(so) < _sources_mtime()
+
+
+def build(verbose=True, force=False):
+    if not force and not needs_build():
+        return ext_path()
+    import pybind11
+
+    hipcc = _hipcc()
+    py_inc = sysconfig.get_paths()["include"]
+    pb_inc = pybind11.get_include()
+    objdir = os.path.join(ROOT, "_objs")
+    os.makedirs(objdir, exist_ok=True)
+
+    cflags = [
+        "-O3", "-std=c++17", "-fPIC", f"--offload-arch={GFX_ARCH}",
+        "-ffp-contract=off",  # bit-identical double math host<->device
+        f"-I{CSRC}", f"-I{os.path.join(CSRC, 'include')}",
+        f"-I{py_inc}", f"-I{pb_inc}",
+        "-Wall", "-Wno-unused-function",
+    ]
+
+    import concurrent.futures
+
+    def compile_one(src):
+        obj = os.path.join(objdir, os.path.basename(src) + ".o")
+        cmd = [hipcc, "-c", src, "-o", obj] + cflags
+        if verbose:
+            print("[cimba_amd build]", " ".join(cmd), flush=True)
+        subprocess.run(cmd, check=True)
+        return obj
+
+    with concurrent.futures.ThreadPoolExecutor(max_workers=len(SOURCES)) as ex:
+        objs = list(ex.map(compile_one, SOURCES))
+
+    so = ext_path()
+    cmd = [hipcc, "-shared", "-fPIC", "-o", so] + objs
+    if verbose:
+        print("[cimba_amd build]", " ".join(cmd), flush=True)
+    subprocess.run(cmd, check=True)
+    return so
+
+
+if __name__ == "__main__":
+    build(force="--force" in sys.argv)
+    print("built", ext_path())

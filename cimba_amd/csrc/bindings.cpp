@@ -1,0 +1,261 @@
+// pybind11 bindings for cimba_amd.
+//
+// The Python layer is orchestration only (bench harness, distributed
+// experiment fan-out via torch.distributed/RCCL); the engine, executive and
+// kernels are native C++/HIP (SURVEY.md §2.1 native-code census).
+#include <pybind11/pybind11.h>
+#include <pybind11/numpy.h>
+#include <pybind11/stl.h>
+
+#include "cimba/runner.hpp"
+#include "cimba/stats.hpp"
+#include "../models/mm1.hpp"
+
+#include <string>
+#include <vector>
+
+namespace py = pybind11;
+using namespace cmb;
+using cmb_models::MM1;
+
+// from hip/deskernel.hip
+extern "C" {
+struct Mm1GpuOut {
+    double elapsed_ms;
+    uint64_t total_events;
+    uint64_t total_objs;
+    double total_wait;
+    uint64_t trials_ok;
+    int32_t first_bad_status;
+    int32_t pad_;
+};
+int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
+                      uint64_t num_objects, uint64_t seed, int device,
+                      double until, uint64_t max_events, Mm1GpuOut* out);
+int cimba_gpu_device_count(int* n);
+int cimba_gpu_sync(void);
+}
+
+static py::dict mm1_host(uint64_t ntrials, uint64_t num_objects, double arr_rate,
+                         double srv_rate, uint64_t seed, int threads) {
+    MM1::Params p{1.0 / arr_rate, 1.0 / srv_rate, num_objects};
+    std::vector<MM1::Result> res(ntrials);
+    {
+        py::gil_scoped_release nogil;
+        run_host<MM1>(p, seed, ntrials, threads, res.data());
+    }
+    uint64_t ev = 0, objs = 0, ok = 0;
+    double wait = 0.0;
+    int32_t bad = 0;
+    py::list per_trial_avg;
+    for (auto& r : res) {
+        ev += r.events;
+        objs += r.obj_cnt;
+        wait += r.sum_wait;
+        if (r.status == 0)
+            ++ok;
+        else if (!bad)
+            bad = r.status;
+        per_trial_avg.append(r.obj_cnt ? r.sum_wait / (double)r.obj_cnt : 0.0);
+    }
+    py::dict d;
+    d["total_events"] = ev;
+    d["total_objects"] = objs;
+    d["total_wait"] = wait;
+    d["trials_ok"] = ok;
+    d["first_bad_status"] = bad;
+    d["avg_system_time"] = objs ? wait / (double)objs : 0.0;
+    d["per_trial_avg"] = per_trial_avg;
+    return d;
+}
+
+static py::dict mm1_gpu(uint64_t ntrials, uint64_t num_objects, double arr_rate,
+                        double srv_rate, uint64_t seed, int device) {
+    Mm1GpuOut o;
+    int rc;
+    {
+        py::gil_scoped_release nogil;
+        rc = cimba_mm1_gpu_run(ntrials, 1.0 / arr_rate, 1.0 / srv_rate,
+                               num_objects, seed, device, 1.0e308,
+                               UINT64_C(0xFFFFFFFFFFFFFFFF), &o);
+    }
+    if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
+    py::dict d;
+    d["elapsed_ms"] = o.elapsed_ms;
+    d["total_events"] = o.total_events;
+    d["total_objects"] = o.total_objs;
+    d["total_wait"] = o.total_wait;
+    d["trials_ok"] = o.trials_ok;
+    d["first_bad_status"] = o.first_bad_status;
+    d["avg_system_time"] = o.total_objs ? o.total_wait / (double)o.total_objs : 0.0;
+    d["events_per_sec"] =
+        o.elapsed_ms > 0 ? (double)o.total_events / (o.elapsed_ms * 1e-3) : 0.0;
+    return d;
+}
+
+static int gpu_device_count() {
+    int n = 0;
+    (void)cimba_gpu_device_count(&n);
+    return n;
+}
+
+// ---- RNG sampling for statistical tests -----------------------------------
+
+static py::array_t<double> rng_sample(const std::string& dist,
+                                      std::vector<double> a, uint64_t n,
+                                      uint64_t seed) {
+    py::array_t<double> out((py::ssize_t)n);
+    double* o = out.mutable_data();
+    Rng r;
+    r.seed(seed);
+    auto P = [&](size_t i) { return i < a.size() ? a[i] : 0.0; };
+    if (dist == "u64") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = (double)(r.next() >> 11);
+    } else if (dist == "uniform") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.uniform(P(0), P(1));
+    } else if (dist == "std_normal") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.std_normal();
+    } else if (dist == "normal") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.normal(P(0), P(1));
+    } else if (dist == "std_exponential") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.std_exponential();
+    } else if (dist == "exponential") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.exponential(P(0));
+    } else if (dist == "lognormal") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.lognormal(P(0), P(1));
+    } else if (dist == "logistic") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.logistic(P(0), P(1));
+    } else if (dist == "cauchy") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.cauchy(P(0), P(1));
+    } else if (dist == "rayleigh") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.rayleigh(P(0));
+    } else if (dist == "weibull") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.weibull(P(0), P(1));
+    } else if (dist == "pareto") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.pareto(P(0), P(1));
+    } else if (dist == "triangular") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.triangular(P(0), P(1), P(2));
+    } else if (dist == "pert") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.pert(P(0), P(1), P(2));
+    } else if (dist == "std_gamma") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.std_gamma(P(0));
+    } else if (dist == "gamma") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.gamma(P(0), P(1));
+    } else if (dist == "erlang") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.erlang((int64_t)P(0), P(1));
+    } else if (dist == "hypoexponential") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.hypoexponential(P(0), P(1));
+    } else if (dist == "hyperexponential") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.hyperexponential(P(0), P(1), P(2));
+    } else if (dist == "std_beta") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.std_beta(P(0), P(1));
+    } else if (dist == "beta") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.beta(P(0), P(1), P(2), P(3));
+    } else if (dist == "chisquared") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.chisquared(P(0));
+    } else if (dist == "std_t_dist") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.std_t_dist(P(0));
+    } else if (dist == "t_dist") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.t_dist(P(0), P(1), P(2));
+    } else if (dist == "f_dist") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = r.f_dist(P(0), P(1));
+    } else if (dist == "bernoulli") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = (double)r.bernoulli(P(0));
+    } else if (dist == "geometric") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = (double)r.geometric(P(0));
+    } else if (dist == "poisson") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = (double)r.poisson(P(0));
+    } else if (dist == "binomial") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = (double)r.binomial((int64_t)P(0), P(1));
+    } else if (dist == "negative_binomial") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = (double)r.negative_binomial(P(0), P(1));
+    } else if (dist == "discrete_uniform") {
+        for (uint64_t i = 0; i < n; ++i)
+            o[i] = (double)r.discrete_uniform((int64_t)P(0), (int64_t)P(1));
+    } else if (dist == "dice") {
+        for (uint64_t i = 0; i < n; ++i) o[i] = (double)r.dice((int64_t)P(0));
+    } else if (dist == "discrete_nonuniform") {
+        for (uint64_t i = 0; i < n; ++i)
+            o[i] = (double)r.discrete_nonuniform(a.data(), (int64_t)a.size());
+    } else if (dist == "alias") {
+        std::vector<double> prob(a.size());
+        std::vector<int32_t> alias(a.size());
+        std::vector<int32_t> scratch(2 * a.size());
+        alias_build(a.data(), (int64_t)a.size(), prob.data(), alias.data(),
+                    scratch.data());
+        AliasTable t{prob.data(), alias.data(), (int64_t)a.size()};
+        for (uint64_t i = 0; i < n; ++i) o[i] = (double)t.sample(r);
+    } else {
+        throw std::invalid_argument("unknown distribution: " + dist);
+    }
+    return out;
+}
+
+static uint64_t py_fmix64(uint64_t x) { return fmix64(x); }
+static uint64_t py_sfc64_raw(uint64_t seed, uint64_t skip) {
+    Rng r;
+    r.seed(seed);
+    for (uint64_t i = 0; i < skip; ++i) (void)r.next();
+    return r.next();
+}
+
+PYBIND11_MODULE(_C, m) {
+    m.doc() = "cimba_amd native engine (MI355X / gfx950)";
+
+    m.def("mm1_host", &mm1_host, py::arg("ntrials"), py::arg("num_objects"),
+          py::arg("arr_rate") = 0.9, py::arg("srv_rate") = 1.0,
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("threads") = 0);
+    m.def("mm1_gpu", &mm1_gpu, py::arg("ntrials"), py::arg("num_objects"),
+          py::arg("arr_rate") = 0.9, py::arg("srv_rate") = 1.0,
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0);
+    m.def("gpu_device_count", &gpu_device_count);
+    m.def("gpu_sync", []() { return cimba_gpu_sync(); });
+
+    m.def("rng_sample", &rng_sample, py::arg("dist"), py::arg("params"),
+          py::arg("n"), py::arg("seed") = 1ULL);
+    m.def("fmix64", &py_fmix64);
+    m.def("sfc64_raw", &py_sfc64_raw, py::arg("seed"), py::arg("skip") = 0);
+    m.def("engine_sizeof_mm1", []() { return sizeof(Engine<MM1>); });
+
+    py::class_<DataSummary>(m, "DataSummary")
+        .def(py::init([]() {
+            DataSummary s;
+            s.reset();
+            return s;
+        }))
+        .def("add", &DataSummary::add)
+        .def("merge", &DataSummary::merge)
+        .def("count", &DataSummary::count)
+        .def("mean", [](const DataSummary& s) { return s.mean; })
+        .def("minimum", [](const DataSummary& s) { return s.mn; })
+        .def("maximum", [](const DataSummary& s) { return s.mx; })
+        .def("variance", &DataSummary::variance)
+        .def("stddev", &DataSummary::stddev)
+        .def("skewness", &DataSummary::skewness)
+        .def("kurtosis", &DataSummary::kurtosis)
+        .def("raw", [](const DataSummary& s) {
+            return py::make_tuple(s.n, s.mean, s.m2, s.m3, s.m4, s.mn, s.mx);
+        })
+        .def_static("from_raw", [](double n, double mean, double m2, double m3,
+                                   double m4, double mn, double mx) {
+            DataSummary s;
+            s.n = n; s.mean = mean; s.m2 = m2; s.m3 = m3; s.m4 = m4;
+            s.mn = mn; s.mx = mx;
+            return s;
+        });
+
+    py::class_<WtdSummary>(m, "WtdSummary")
+        .def(py::init([]() {
+            WtdSummary s;
+            s.reset();
+            return s;
+        }))
+        .def("add", &WtdSummary::add)
+        .def("merge", &WtdSummary::merge)
+        .def("mean", [](const WtdSummary& s) { return s.mean; })
+        .def("sumw", [](const WtdSummary& s) { return s.sumw; })
+        .def("variance", &WtdSummary::variance)
+        .def("stddev", &WtdSummary::stddev)
+        .def("minimum", [](const WtdSummary& s) { return s.mn; })
+        .def("maximum", [](const WtdSummary& s) { return s.mx; });
+}

@@ -1,0 +1,1031 @@
+// cimba_amd simulation engine: clock + future-event list + state-machine
+// processes + process-interaction toolkit, in one POD aggregate that runs
+// identically as host code (CPU executive, unit tests) and as HIP device
+// code with the whole trial resident in LDS (one trial per wavefront).
+//
+// This is the MI355X-native redesign of the reference's L2-L4 layers
+// (SURVEY.md §1):
+//  - reference src/cmb_event.c (thread-local clock + hash-heap event queue,
+//    dispatch loop cmb_event.c:370-410)            -> Engine::dispatch_one()
+//  - reference src/cmi_coroutine.c + the x86-64 context-switch assembly
+//    (port/x86-64/linux/cmi_coroutine_context.asm) -> protothread-style
+//    resumable process functions (CMB_BEGIN/CMB_HOLD/.../CMB_END below):
+//    every blocking call site is a numbered resumption point, the
+//    "coroutine switch rewritten as a state-machine scheduler" of
+//    BASELINE.json's north star.  There is no stack to switch - a process
+//    is (pc, frame) and the scheduler is the event loop itself.
+//  - reference src/cmb_process.c (hold/wait/interrupt/stop + the
+//    enqueue-intent -> yield -> revalidate -> resume-with-signal discipline,
+//    cmb_process.c:481-508)                        -> Proc + await_cleanup()
+//  - reference src/cmb_resourceguard.c (demand-predicate wait queue,
+//    stale-grant handling :186-228)                -> Guard + guard_signal()
+//  - reference src/cmb_resource.c / cmb_resourcepool.c / cmb_buffer.c /
+//    cmb_objectqueue.c / cmb_priorityqueue.c / cmb_condition.c -> the
+//    fixed-capacity toolkit structs in this header.
+//
+// Everything is bounded-capacity (Cfg) so a trial's engine fits in LDS;
+// capacity overflow aborts the trial with a status code (the per-block
+// trial-abort flag of SURVEY.md §5.3) rather than growing.
+#pragma once
+
+#include "config.hpp"
+#include "hashheap.hpp"
+#include "rng.hpp"
+#include "stats.hpp"
+
+namespace cmb {
+
+// ---- signals (same contract as reference include/cmb_process.h:60-100) ----
+typedef int64_t sig_t;
+constexpr sig_t SIG_SUCCESS = 0;
+constexpr sig_t SIG_PREEMPTED = -1;
+constexpr sig_t SIG_INTERRUPTED = -2;
+constexpr sig_t SIG_STOPPED = -3;
+constexpr sig_t SIG_CANCELLED = -4;
+constexpr sig_t SIG_TIMEOUT = -5;
+
+// ---- engine status codes (trial abort reasons) ----
+enum EngStatus : int32_t {
+    ST_OK = 0,
+    ST_HEAP_FULL = 1,
+    ST_QUEUE_FULL = 2,
+    ST_BAD_STATE = 3,
+    ST_GUARD_OVERFLOW = 4,
+    ST_USER_ABORT = 5,
+    ST_EVENT_LIMIT = 6,
+};
+
+// ---- internal event kinds; model-defined kinds start at EV_USER ----
+enum EvKind : uint16_t {
+    EV_PROC_START = 1,
+    EV_TIMER = 2,   // a=pidx, c=timer slot, b=signal
+    EV_GRANT = 3,   // a=pidx, c=guard id,  b=wait key
+    EV_RESUME = 4,  // a=pidx, c=epoch,     b=signal
+    EV_USER = 16,
+};
+
+// ---- await kinds ----
+enum AwaitKind : uint8_t {
+    AW_NONE = 0,
+    AW_TIME = 1,   // blocking hold; key = timer-0 event handle
+    AW_GUARD = 2,  // guard wait;    key = wait_key
+    AW_PROC = 3,   // wait_process;  key = target pidx
+    AW_EVENT = 4,  // wait_event;    key = event handle
+};
+
+// ---- demand predicate kinds (guard wait conditions) ----
+enum DemandKind : uint8_t {
+    DEM_QSPACE = 1,  // ctx = queue idx: space in object queue
+    DEM_QOBJ = 2,    // ctx = queue idx: object available
+    DEM_RES = 3,     // ctx = resource idx: resource free
+    DEM_POOL = 4,    // ctx = pool idx: any units free
+    DEM_BUF_GE = 5,  // ctx = buf idx | amount<<8: level >= amount
+    DEM_BUF_SP = 6,  // ctx = buf idx | amount<<8: space >= amount
+    DEM_PQOBJ = 7,   // ctx = pq idx: object available
+    DEM_PQSP = 8,    // ctx = pq idx: space available
+    DEM_USER = 32,   // >= DEM_USER: Model::demand(E, pidx, kind, ctx)
+};
+
+enum ProcState : uint8_t {
+    PS_UNINIT = 0,
+    PS_READY = 1,    // initialized, not started
+    PS_RUNNING = 2,  // started (running or blocked)
+    PS_FINISHED = 3,
+};
+
+constexpr int16_t PC_DONE = -1;
+
+// ---------------------------------------------------------------------------
+// Process record.  This replaces the reference's cmb_process + embedded
+// cmi_coroutine (include/cmb_process.h:118-128): the coroutine stack becomes
+// (pc, model frame); awaits/resources/waiters lists become the fields below.
+// ---------------------------------------------------------------------------
+template <int NT>
+struct ProcRec {
+    int16_t pc;
+    uint8_t state;
+    uint8_t func;       // model process-function id
+    int16_t priority;
+    uint8_t await_kind;
+    uint8_t g_granted;  // guard grant consumed by the next try_*
+    uint32_t await_key;
+    uint32_t epoch;     // bumped at every blocking setup; staleness key for EV_RESUME
+    sig_t sig;          // signal delivered at resume (CMB_SIG())
+    // guard-wait bookkeeping (valid while await_kind == AW_GUARD)
+    int16_t gid;
+    int16_t gnext;      // intrusive waiter list link
+    uint8_t demand_kind;
+    uint8_t pad0_;
+    uint16_t pad1_;
+    uint32_t demand_ctx;
+    uint32_t wseq;      // FIFO tie-break within a guard
+    double entry_t;     // guard entry time (ordering: pri desc, entry asc, seq asc)
+    uint32_t timers[NT];  // pending timer event handles; 0 = free slot
+    int16_t waiters_head;  // procs waiting for me to finish (wait_process)
+    int16_t pnext;         // my link in another proc's waiter list
+};
+
+// ---------------------------------------------------------------------------
+// Interaction toolkit (fixed capacity, all POD)
+// ---------------------------------------------------------------------------
+
+struct Guard {
+    int16_t head;      // waiter list (proc indices via gnext), -1 = empty
+    int16_t observer;  // condition index observing this guard, -1 = none
+    CMB_FORCEINLINE bool empty() const { return head < 0; }
+};
+
+// FIFO object queue of 64-bit payloads (reference cmb_objectqueue: FIFO of
+// void*, two guards, capacity limit, length history).
+template <int CAP>
+struct ObjQueue {
+    uint64_t ring[CAP];
+    int32_t head, len;
+    int32_t limit;       // runtime capacity (<= CAP); CMB_UNLIMITED -> CAP
+    int16_t g_front;     // getters wait here
+    int16_t g_rear;      // putters wait here
+    uint8_t recording;
+    WtdSummary len_stats;  // time-weighted queue length
+    double t_last;
+};
+
+// one-holder resource (reference cmb_resource: binary semaphore with
+// priority queue + preemption)
+struct Resource {
+    int16_t holder;  // proc idx or -1
+    int16_t gid;
+    uint8_t recording;
+    WtdSummary busy;  // time-weighted utilization (0/1)
+    double t_last;
+};
+
+// counting semaphore (reference cmb_resourcepool)
+struct Pool {
+    int32_t capacity;
+    int32_t in_use;
+    int16_t gid;
+    uint8_t recording;
+    WtdSummary use_stats;  // time-weighted units in use
+    double t_last;
+};
+
+// producer/consumer level store (reference cmb_buffer)
+struct Buffer {
+    int64_t level;
+    int64_t capacity;
+    int16_t g_get;  // getters wait here (front)
+    int16_t g_put;  // putters wait here (rear)
+    uint8_t recording;
+    WtdSummary level_stats;
+    double t_last;
+};
+
+// priority-ordered object queue (reference cmb_priorityqueue: objects
+// retrieved in (priority desc, FIFO) order)
+template <int CAP>
+struct PrioQueue {
+    uint64_t key[CAP];  // (32767-pri)<<32 | seq
+    uint64_t val[CAP];
+    int32_t len;
+    int32_t limit;
+    uint32_t seq;
+    int16_t g_front, g_rear;
+    uint8_t recording;
+    WtdSummary len_stats;
+    double t_last;
+};
+
+// condition variable (reference cmb_condition: wait on arbitrary predicate;
+// signal wakes EVERY satisfied waiter; may observe other guards)
+struct Condition {
+    int16_t gid;
+};
+
+constexpr int32_t CMB_UNLIMITED = 0x7FFFFFFF;
+
+// Default model hooks: models inherit and override what they use.
+struct ModelBase {
+    template <class E_>
+    CMB_HD static bool demand(E_&, int /*pidx*/, uint8_t /*kind*/, uint32_t /*ctx*/) {
+        return false;
+    }
+    template <class E_>
+    CMB_HD static void on_event(E_&, const EvEntry&) {}
+};
+
+template <int N>
+struct ArrOf {
+    static constexpr int n = (N > 0) ? N : 1;
+};
+
+// ---------------------------------------------------------------------------
+// The engine
+// ---------------------------------------------------------------------------
+template <class Model>
+struct Engine {
+    using Cfg = typename Model::Cfg;
+    using Params = typename Model::Params;
+    using Frame = typename Model::Frame;
+    using ProcT = ProcRec<Cfg::TIMERS>;
+    using Self = Engine<Model>;
+
+    static constexpr int NQ = Cfg::NUM_QUEUES;
+    static constexpr int NR = Cfg::NUM_RES;
+    static constexpr int NP = Cfg::NUM_POOLS;
+    static constexpr int NB = Cfg::NUM_BUFS;
+    static constexpr int NPQ = Cfg::NUM_PQ;
+    static constexpr int NC = Cfg::NUM_COND;
+    static constexpr int NGUARD =
+        2 * NQ + NR + NP + 2 * NB + 2 * NPQ + NC > 0
+            ? 2 * NQ + NR + NP + 2 * NB + 2 * NPQ + NC
+            : 1;
+
+    // ---- state ----
+    double now;
+    uint64_t ev_dispatched;   // the benchmark metric: events executed
+    uint64_t seq;             // event FIFO counter
+    uint32_t next_handle;
+    int32_t status;
+    int32_t n_event_waiters;  // gates the wait_event scan
+    uint32_t trial_index;
+    const Params* params;
+    Rng rng;
+
+    HashHeap<Cfg::MAX_EV> evq;
+    ProcT procs[Cfg::MAX_PROC];
+    Frame frames[Cfg::MAX_PROC];
+    Guard guards[NGUARD];
+    ObjQueue<Cfg::QCAP> queues[ArrOf<NQ>::n];
+    Resource resources[ArrOf<NR>::n];
+    Pool pools[ArrOf<NP>::n];
+    Buffer buffers[ArrOf<NB>::n];
+    PrioQueue<Cfg::PQCAP> pqueues[ArrOf<NPQ>::n];
+    Condition conds[ArrOf<NC>::n];
+
+    // ---- lifecycle --------------------------------------------------------
+
+    CMB_HD void init(const Params* p, uint64_t trial_seed, uint32_t tidx,
+                     double start_time = 0.0) {
+        now = start_time;
+        ev_dispatched = 0;
+        seq = 0;
+        next_handle = 1;
+        status = ST_OK;
+        n_event_waiters = 0;
+        trial_index = tidx;
+        params = p;
+        rng.seed(trial_seed);
+        evq.reset();
+        for (int i = 0; i < Cfg::MAX_PROC; ++i) {
+            ProcT& pr = procs[i];
+            pr.pc = 0;
+            pr.state = PS_UNINIT;
+            pr.func = 0;
+            pr.priority = 0;
+            pr.await_kind = AW_NONE;
+            pr.g_granted = 0;
+            pr.await_key = 0;
+            pr.epoch = 0;
+            pr.sig = SIG_SUCCESS;
+            pr.gid = -1;
+            pr.gnext = -1;
+            pr.waiters_head = -1;
+            pr.pnext = -1;
+            for (int t = 0; t < Cfg::TIMERS; ++t) pr.timers[t] = 0;
+        }
+        for (int i = 0; i < NGUARD; ++i) { guards[i].head = -1; guards[i].observer = -1; }
+        int g = 0;
+        for (int i = 0; i < NQ; ++i) {
+            ObjQueue<Cfg::QCAP>& q = queues[i];
+            q.head = 0; q.len = 0; q.limit = Cfg::QCAP;
+            q.g_front = (int16_t)g++; q.g_rear = (int16_t)g++;
+            q.recording = 0; q.len_stats.reset(); q.t_last = now;
+        }
+        for (int i = 0; i < NR; ++i) {
+            resources[i].holder = -1; resources[i].gid = (int16_t)g++;
+            resources[i].recording = 0; resources[i].busy.reset();
+            resources[i].t_last = now;
+        }
+        for (int i = 0; i < NP; ++i) {
+            pools[i].capacity = 1; pools[i].in_use = 0; pools[i].gid = (int16_t)g++;
+            pools[i].recording = 0; pools[i].use_stats.reset(); pools[i].t_last = now;
+        }
+        for (int i = 0; i < NB; ++i) {
+            buffers[i].level = 0; buffers[i].capacity = CMB_UNLIMITED;
+            buffers[i].g_get = (int16_t)g++; buffers[i].g_put = (int16_t)g++;
+            buffers[i].recording = 0; buffers[i].level_stats.reset();
+            buffers[i].t_last = now;
+        }
+        for (int i = 0; i < NPQ; ++i) {
+            PrioQueue<Cfg::PQCAP>& q = pqueues[i];
+            q.len = 0; q.limit = Cfg::PQCAP; q.seq = 0;
+            q.g_front = (int16_t)g++; q.g_rear = (int16_t)g++;
+            q.recording = 0; q.len_stats.reset(); q.t_last = now;
+        }
+        for (int i = 0; i < NC; ++i) conds[i].gid = (int16_t)g++;
+    }
+
+    CMB_FORCEINLINE void fail(int32_t st) {
+        if (status == ST_OK) status = st;
+    }
+
+    CMB_FORCEINLINE int pidx_of(const ProcT* p) const { return (int)(p - procs); }
+
+    // ---- event scheduling (reference cmb_event_schedule, cmb_event.c:199) --
+
+    CMB_FORCEINLINE uint32_t schedule(uint16_t kind, uint16_t a, uint32_t c,
+                                      uint64_t b, double t, int priority) {
+        EvEntry ev;
+        ev.t = t;
+        ev.pseq = ev_pseq(priority, seq++);
+        ev.b = b;
+        ev.handle = next_handle++;
+        if (next_handle == 0) next_handle = 1;
+        ev.kind = kind;
+        ev.a = a;
+        ev.c = c;
+        if (!evq.push(ev)) {
+            fail(ST_HEAP_FULL);
+            return 0;
+        }
+        return ev.handle;
+    }
+
+    CMB_FORCEINLINE bool event_cancel(uint32_t handle) {
+        EvEntry out;
+        if (!evq.cancel(handle, &out)) return false;
+        if (n_event_waiters) wake_event_waiters(handle, SIG_CANCELLED);
+        return true;
+    }
+
+    CMB_FORCEINLINE bool event_reschedule(uint32_t handle, double t, int priority) {
+        return evq.reschedule(handle, t, ev_pseq(priority, seq++));
+    }
+
+    // ---- timers (reference cmb_process_timer_add/set/cancel/clear,
+    //      cmb_process.c:514-580; slot 0 reserved for hold/timeout) ---------
+
+    CMB_FORCEINLINE bool timer_add(ProcT& p, int slot, double delay, sig_t sg) {
+        cmb_assert_debug(slot >= 0 && slot < Cfg::TIMERS);
+        if (p.timers[slot]) return false;  // slot busy
+        const uint32_t h = schedule(EV_TIMER, (uint16_t)pidx_of(&p), (uint32_t)slot,
+                                    (uint64_t)sg, now + delay, p.priority);
+        p.timers[slot] = h;
+        return h != 0;
+    }
+
+    CMB_FORCEINLINE void timer_cancel(ProcT& p, int slot) {
+        if (p.timers[slot]) {
+            evq.cancel(p.timers[slot]);
+            p.timers[slot] = 0;
+        }
+    }
+
+    // ---- process control ---------------------------------------------------
+
+    CMB_HD void proc_init(int pidx, uint8_t func, int priority) {
+        ProcT& p = procs[pidx];
+        cmb_assert_debug(p.state == PS_UNINIT);
+        p.state = PS_READY;
+        p.func = (uint8_t)func;
+        p.priority = (int16_t)priority;
+        p.pc = 0;
+    }
+
+    // non-blocking start: schedules a start event (reference
+    // cmb_process_start, cmb_process.c:247-254)
+    CMB_HD void proc_start(int pidx, double delay = 0.0) {
+        ProcT& p = procs[pidx];
+        cmb_assert_debug(p.state == PS_READY);
+        schedule(EV_PROC_START, (uint16_t)pidx, 0, 0, now + delay, p.priority);
+    }
+
+    // non-blocking interrupt: wake the target's current wait with `sg`
+    // (reference cmb_process_interrupt)
+    CMB_HD void proc_interrupt(int pidx, sig_t sg) {
+        ProcT& p = procs[pidx];
+        if (p.state != PS_RUNNING || p.await_kind == AW_NONE) return;
+        schedule(EV_RESUME, (uint16_t)pidx, p.epoch, (uint64_t)sg, now, p.priority);
+    }
+
+    // kill a process: drop resources, cancel awaitables, wake waiters
+    // (reference kill path, cmb_process.c:979-1004)
+    CMB_HD void proc_stop(int pidx) {
+        ProcT& p = procs[pidx];
+        if (p.state == PS_FINISHED || p.state == PS_UNINIT) return;
+        for (int t = 0; t < Cfg::TIMERS; ++t) timer_cancel(p, t);
+        if (p.await_kind == AW_GUARD) guard_unlink(p);
+        if (p.await_kind == AW_PROC) proc_waiter_unlink(p);
+        if (p.await_kind == AW_EVENT) --n_event_waiters;
+        p.await_kind = AW_NONE;
+        drop_held(pidx);
+        finish_common(p, SIG_STOPPED);
+    }
+
+    CMB_HD void proc_priority_set(int pidx, int priority) {
+        procs[pidx].priority = (int16_t)priority;
+    }
+
+    // called by CMB_END / early exit
+    CMB_FORCEINLINE void proc_finish(ProcT& p) {
+        p.pc = PC_DONE;
+        drop_held(pidx_of(&p));
+        finish_common(p, SIG_SUCCESS);
+    }
+
+    CMB_FORCEINLINE void finish_common(ProcT& p, sig_t waiter_sig) {
+        p.state = PS_FINISHED;
+        int16_t w = p.waiters_head;
+        p.waiters_head = -1;
+        while (w >= 0) {
+            ProcT& wp = procs[w];
+            const int16_t nxt = wp.pnext;
+            wp.pnext = -1;
+            schedule(EV_RESUME, (uint16_t)w, wp.epoch, (uint64_t)waiter_sig, now,
+                     wp.priority);
+            w = nxt;
+        }
+    }
+
+    // release everything a killed/finished process still holds (reference
+    // cmi_holdable drop polymorphism, src/cmi_holdable.h:53-78); bounded
+    // scans over the small toolkit arrays replace the intrusive list.
+    CMB_HD void drop_held(int pidx) {
+        for (int r = 0; r < NR; ++r) {
+            if (resources[r].holder == (int16_t)pidx) resource_release(r);
+        }
+        // pools: per-holder amounts are not tracked (see docs/PARITY.md);
+        // a killed process's pool units are released by the model's cleanup.
+    }
+
+    // ---- the await/resume discipline --------------------------------------
+    // reference pattern: (enqueue intent) -> yield ->
+    // (scheduled wakeup revalidates intent) -> resume with signal.
+
+    CMB_FORCEINLINE void await_setup(ProcT& p, uint8_t kind, uint32_t key) {
+        p.await_kind = kind;
+        p.await_key = key;
+        p.epoch++;
+    }
+
+    // runs at every resumption point before user code continues: clears any
+    // residue of the primary await when the wake came from elsewhere
+    // (timer/interrupt), mirroring the reference's revalidation discipline.
+    CMB_FORCEINLINE void await_cleanup(ProcT& p) {
+        switch (p.await_kind) {
+            case AW_NONE: return;
+            case AW_TIME:
+                // woken by something other than the hold timer: cancel it
+                // (reference cmb_process.c:467-471)
+                timer_cancel(p, 0);
+                break;
+            case AW_GUARD:
+                guard_unlink(p);
+                break;
+            case AW_PROC:
+                proc_waiter_unlink(p);
+                break;
+            case AW_EVENT:
+                --n_event_waiters;
+                break;
+        }
+        p.await_kind = AW_NONE;
+    }
+
+    // blocking-call setups (used by the CMB_* macros) ---------------------
+
+    CMB_FORCEINLINE void hold_setup(ProcT& p, double dur) {
+        timer_add(p, 0, dur, SIG_SUCCESS);
+        await_setup(p, AW_TIME, p.timers[0]);
+    }
+
+    CMB_FORCEINLINE void timeout_arm(ProcT& p, double dur) {
+        timer_add(p, 0, dur, SIG_TIMEOUT);
+    }
+    CMB_FORCEINLINE void timeout_disarm(ProcT& p) { timer_cancel(p, 0); }
+
+    CMB_FORCEINLINE bool wait_proc_setup(ProcT& p, int target) {
+        ProcT& t = procs[target];
+        if (t.state == PS_FINISHED) return false;  // no wait needed
+        p.pnext = t.waiters_head;
+        t.waiters_head = (int16_t)pidx_of(&p);
+        await_setup(p, AW_PROC, (uint32_t)target);
+        return true;
+    }
+
+    CMB_FORCEINLINE void proc_waiter_unlink(ProcT& p) {
+        ProcT& t = procs[p.await_key];
+        int16_t* link = &t.waiters_head;
+        const int16_t me = (int16_t)pidx_of(&p);
+        while (*link >= 0) {
+            if (*link == me) {
+                *link = p.pnext;
+                p.pnext = -1;
+                return;
+            }
+            link = &procs[*link].pnext;
+        }
+    }
+
+    CMB_FORCEINLINE void wait_event_setup(ProcT& p, uint32_t handle) {
+        await_setup(p, AW_EVENT, handle);
+        ++n_event_waiters;
+    }
+
+    CMB_HD void wake_event_waiters(uint32_t handle, sig_t sg) {
+        for (int i = 0; i < Cfg::MAX_PROC; ++i) {
+            ProcT& p = procs[i];
+            if (p.await_kind == AW_EVENT && p.await_key == handle) {
+                schedule(EV_RESUME, (uint16_t)i, p.epoch, (uint64_t)sg, now,
+                         p.priority);
+            }
+        }
+    }
+
+    // ---- guards (reference cmb_resourceguard.c) ---------------------------
+
+    CMB_FORCEINLINE void guard_wait(ProcT& p, int gid, uint8_t demand,
+                                    uint32_t ctx) {
+        p.gid = (int16_t)gid;
+        p.demand_kind = demand;
+        p.demand_ctx = ctx;
+        p.entry_t = now;
+        p.wseq = (uint32_t)seq++;
+        p.gnext = guards[gid].head;
+        guards[gid].head = (int16_t)pidx_of(&p);
+        await_setup(p, AW_GUARD, (uint32_t)p.wseq);
+    }
+
+    CMB_FORCEINLINE void guard_unlink(ProcT& p) {
+        Guard& g = guards[p.gid];
+        int16_t* link = &g.head;
+        const int16_t me = (int16_t)pidx_of(&p);
+        while (*link >= 0) {
+            if (*link == me) {
+                *link = p.gnext;
+                p.gnext = -1;
+                return;
+            }
+            link = &procs[*link].gnext;
+        }
+    }
+
+    // front waiter: max priority, then earliest entry time, then lowest seq
+    // (reference cmb_resourceguard.c:66-89 ordering)
+    CMB_FORCEINLINE int guard_front(int gid) const {
+        int best = -1;
+        for (int16_t i = guards[gid].head; i >= 0; i = procs[i].gnext) {
+            if (best < 0) { best = i; continue; }
+            const ProcT& a = procs[i];
+            const ProcT& b = procs[best];
+            if (a.priority > b.priority ||
+                (a.priority == b.priority &&
+                 (a.entry_t < b.entry_t ||
+                  (a.entry_t == b.entry_t && a.wseq < b.wseq))))
+                best = i;
+        }
+        return best;
+    }
+
+    CMB_HD bool eval_demand(const ProcT& p) {
+        const uint32_t ctx = p.demand_ctx;
+        switch (p.demand_kind) {
+            case DEM_QSPACE: return queues[ctx].len < queues[ctx].limit;
+            case DEM_QOBJ: return queues[ctx].len > 0;
+            case DEM_RES: return resources[ctx].holder < 0;
+            case DEM_POOL: return pools[ctx].in_use < pools[ctx].capacity;
+            case DEM_BUF_GE:
+                return buffers[ctx & 0xFF].level >= (int64_t)(ctx >> 8);
+            case DEM_BUF_SP:
+                return buffers[ctx & 0xFF].capacity - buffers[ctx & 0xFF].level >=
+                       (int64_t)(ctx >> 8);
+            case DEM_PQOBJ: return pqueues[ctx].len > 0;
+            case DEM_PQSP: return pqueues[ctx].len < pqueues[ctx].limit;
+            default:
+                return Model::demand(*this, (int)(&p - procs), p.demand_kind, ctx);
+        }
+    }
+
+    // evaluate the front waiter's demand; if satisfied, schedule a grant
+    // event (reference cmb_resourceguard.c:240-260 semantics: the grant is
+    // a hint — the woken process revalidates in its acquire loop)
+    CMB_HD bool guard_signal(int gid) {
+        const int w = guard_front(gid);
+        bool granted = false;
+        if (w >= 0 && eval_demand(procs[w])) {
+            ProcT& p = procs[w];
+            schedule(EV_GRANT, (uint16_t)w, (uint32_t)gid, (uint64_t)p.wseq, now,
+                     p.priority);
+            granted = true;
+        }
+        const int16_t obs = guards[gid].observer;
+        if (obs >= 0) condition_signal(obs);
+        return granted;
+    }
+
+    // condition signal: wake EVERY satisfied waiter (reference
+    // include/cmb_condition.h:17-24)
+    CMB_HD uint64_t condition_signal(int ci) {
+        const int gid = conds[ci].gid;
+        uint64_t cnt = 0;
+        for (int16_t i = guards[gid].head; i >= 0; i = procs[i].gnext) {
+            if (eval_demand(procs[i])) {
+                ProcT& p = procs[i];
+                schedule(EV_GRANT, (uint16_t)i, (uint32_t)gid, (uint64_t)p.wseq,
+                         now, p.priority);
+                ++cnt;
+            }
+        }
+        return cnt;
+    }
+
+    // ---- toolkit operations ------------------------------------------------
+
+    CMB_FORCEINLINE void q_record(ObjQueue<Cfg::QCAP>& q) {
+        if (q.recording) {
+            q.len_stats.add((double)q.len, now - q.t_last);
+            q.t_last = now;
+        }
+    }
+
+    CMB_HD bool q_try_put(int qi, ProcT& p, uint64_t val) {
+        ObjQueue<Cfg::QCAP>& q = queues[qi];
+        const bool may = p.g_granted || guards[q.g_rear].empty();
+        p.g_granted = 0;
+        if (!may || q.len >= q.limit) return false;
+        if (q.len >= Cfg::QCAP) { fail(ST_QUEUE_FULL); return false; }
+        q_record(q);
+        q.ring[(q.head + q.len) % Cfg::QCAP] = val;
+        ++q.len;
+        guard_signal(q.g_front);
+        return true;
+    }
+
+    CMB_HD bool q_try_get(int qi, ProcT& p, uint64_t* out) {
+        ObjQueue<Cfg::QCAP>& q = queues[qi];
+        const bool may = p.g_granted || guards[q.g_front].empty();
+        p.g_granted = 0;
+        if (!may || q.len == 0) return false;
+        q_record(q);
+        *out = q.ring[q.head];
+        q.head = (q.head + 1) % Cfg::QCAP;
+        --q.len;
+        guard_signal(q.g_rear);
+        return true;
+    }
+
+    CMB_FORCEINLINE int64_t q_length(int qi) const { return queues[qi].len; }
+
+    CMB_HD bool pq_try_put(int qi, ProcT& p, uint64_t val, int priority) {
+        PrioQueue<Cfg::PQCAP>& q = pqueues[qi];
+        const bool may = p.g_granted || guards[q.g_rear].empty();
+        p.g_granted = 0;
+        if (!may || q.len >= q.limit) return false;
+        // binary heap push keyed (pri desc, seq asc)
+        const uint64_t key =
+            ((uint64_t)(uint16_t)(32767 - priority) << 32) | q.seq++;
+        int32_t i = q.len++;
+        while (i > 0) {
+            const int32_t par = (i - 1) >> 1;
+            if (q.key[par] <= key) break;
+            q.key[i] = q.key[par];
+            q.val[i] = q.val[par];
+            i = par;
+        }
+        q.key[i] = key;
+        q.val[i] = val;
+        guard_signal(q.g_front);
+        return true;
+    }
+
+    CMB_HD bool pq_try_get(int qi, ProcT& p, uint64_t* out) {
+        PrioQueue<Cfg::PQCAP>& q = pqueues[qi];
+        const bool may = p.g_granted || guards[q.g_front].empty();
+        p.g_granted = 0;
+        if (!may || q.len == 0) return false;
+        *out = q.val[0];
+        --q.len;
+        const uint64_t key = q.key[q.len];
+        const uint64_t val = q.val[q.len];
+        int32_t i = 0;
+        for (;;) {
+            int32_t c = 2 * i + 1;
+            if (c >= q.len) break;
+            if (c + 1 < q.len && q.key[c + 1] < q.key[c]) ++c;
+            if (q.key[c] >= key) break;
+            q.key[i] = q.key[c];
+            q.val[i] = q.val[c];
+            i = c;
+        }
+        q.key[i] = key;
+        q.val[i] = val;
+        guard_signal(q.g_rear);
+        return true;
+    }
+
+    CMB_HD bool res_try_acquire(int ri, ProcT& p) {
+        Resource& r = resources[ri];
+        const bool may = p.g_granted || guards[r.gid].empty();
+        p.g_granted = 0;
+        if (!may || r.holder >= 0) return false;
+        if (r.recording) {
+            r.busy.add(0.0, now - r.t_last);
+            r.t_last = now;
+        }
+        r.holder = (int16_t)pidx_of(&p);
+        return true;
+    }
+
+    CMB_HD void resource_release(int ri) {
+        Resource& r = resources[ri];
+        cmb_assert_debug(r.holder >= 0);
+        if (r.recording) {
+            r.busy.add(1.0, now - r.t_last);
+            r.t_last = now;
+        }
+        r.holder = -1;
+        guard_signal(r.gid);
+    }
+
+    CMB_HD int32_t pool_try_take(int pi, ProcT& p, int32_t want) {
+        Pool& pl = pools[pi];
+        const bool may = p.g_granted || guards[pl.gid].empty();
+        p.g_granted = 0;
+        if (!may) return 0;
+        const int32_t free_units = pl.capacity - pl.in_use;
+        const int32_t take = free_units < want ? free_units : want;
+        if (take > 0) {
+            if (pl.recording) {
+                pl.use_stats.add((double)pl.in_use, now - pl.t_last);
+                pl.t_last = now;
+            }
+            pl.in_use += take;
+        }
+        return take;
+    }
+
+    CMB_HD void pool_release(int pi, int32_t amount) {
+        Pool& pl = pools[pi];
+        cmb_assert_debug(pl.in_use >= amount);
+        if (pl.recording) {
+            pl.use_stats.add((double)pl.in_use, now - pl.t_last);
+            pl.t_last = now;
+        }
+        pl.in_use -= amount;
+        guard_signal(pl.gid);
+    }
+
+    CMB_HD bool buf_try_get(int bi, ProcT& p, int64_t amount) {
+        Buffer& b = buffers[bi];
+        const bool may = p.g_granted || guards[b.g_get].empty();
+        p.g_granted = 0;
+        if (!may || b.level < amount) return false;
+        if (b.recording) {
+            b.level_stats.add((double)b.level, now - b.t_last);
+            b.t_last = now;
+        }
+        b.level -= amount;
+        guard_signal(b.g_put);
+        return true;
+    }
+
+    CMB_HD bool buf_try_put(int bi, ProcT& p, int64_t amount) {
+        Buffer& b = buffers[bi];
+        const bool may = p.g_granted || guards[b.g_put].empty();
+        p.g_granted = 0;
+        if (!may || b.capacity - b.level < amount) return false;
+        if (b.recording) {
+            b.level_stats.add((double)b.level, now - b.t_last);
+            b.t_last = now;
+        }
+        b.level += amount;
+        guard_signal(b.g_get);
+        return true;
+    }
+
+    // ---- dispatch loop (reference cmb_event.c:370-410) ---------------------
+
+    CMB_FORCEINLINE void resume_proc(int pidx, sig_t sg) {
+        ProcT& p = procs[pidx];
+        p.sig = sg;
+        Model::step(*this, pidx);
+    }
+
+    CMB_HD bool dispatch_one() {
+        if (status != ST_OK || evq.empty()) return false;
+        const EvEntry ev = evq.pop();
+        now = ev.t;
+        ++ev_dispatched;
+        if (n_event_waiters) wake_event_waiters(ev.handle, SIG_SUCCESS);
+        switch (ev.kind) {
+            case EV_PROC_START: {
+                ProcT& p = procs[ev.a];
+                if (p.state != PS_READY) break;  // stale (stopped before start)
+                p.state = PS_RUNNING;
+                resume_proc(ev.a, SIG_SUCCESS);
+                break;
+            }
+            case EV_TIMER: {
+                ProcT& p = procs[ev.a];
+                if (p.timers[ev.c] != ev.handle) break;  // stale
+                p.timers[ev.c] = 0;
+                if (p.state != PS_RUNNING || p.await_kind == AW_NONE) break;
+                if (p.await_kind == AW_TIME && p.await_key == ev.handle) {
+                    // normal hold wake
+                    p.await_kind = AW_NONE;
+                }
+                // otherwise: interrupting timer (timeout) — leave the
+                // primary await for await_cleanup at the resumption point
+                resume_proc(ev.a, (sig_t)ev.b);
+                break;
+            }
+            case EV_GRANT: {
+                ProcT& p = procs[ev.a];
+                if (p.state == PS_RUNNING && p.await_kind == AW_GUARD &&
+                    p.wseq == (uint32_t)ev.b && p.gid == (int16_t)ev.c) {
+                    guard_unlink(p);
+                    p.await_kind = AW_NONE;
+                    p.g_granted = 1;
+                    resume_proc(ev.a, SIG_SUCCESS);
+                } else {
+                    // stale grant: the signal must not be lost — re-evaluate
+                    // the guard (reference cmb_resourceguard.c:163-180
+                    // pass-the-grant-on semantics)
+                    guard_signal((int)ev.c);
+                }
+                break;
+            }
+            case EV_RESUME: {
+                ProcT& p = procs[ev.a];
+                if (p.state != PS_RUNNING || p.await_kind == AW_NONE ||
+                    p.epoch != ev.c)
+                    break;  // stale
+                if (p.await_kind == AW_TIME || p.await_kind == AW_PROC ||
+                    p.await_kind == AW_EVENT || p.await_kind == AW_GUARD) {
+                    // leave residue for await_cleanup
+                }
+                resume_proc(ev.a, (sig_t)ev.b);
+                break;
+            }
+            default:
+                Model::on_event(*this, ev);
+                break;
+        }
+        return true;
+    }
+
+    // run until the event queue drains (reference cmb_event_queue_execute,
+    // cmb_event.c:402) or a limit is hit
+    CMB_HD void run(double until, uint64_t max_events) {
+        while (status == ST_OK && !evq.empty()) {
+            if (evq.top().t > until) {
+                now = until;
+                break;
+            }
+            dispatch_one();
+            if (ev_dispatched >= max_events) {
+                fail(ST_EVENT_LIMIT);
+                break;
+            }
+        }
+    }
+};
+
+// ---------------------------------------------------------------------------
+// Protothread macros — the blocking-call surface.  A model process function
+// has the shape:
+//
+//   template <class E_>
+//   CMB_HD static void body(E_& E, typename E_::ProcT* self) {
+//       auto& f = E.frames[E.pidx_of(self)].xxx;   // persistent locals
+//       CMB_BEGIN();
+//       ...
+//       CMB_HOLD(E.rng.exponential(m));            // blocking call
+//       if (CMB_SIG() != cmb::SIG_SUCCESS) ...     // signal inspection
+//       ...
+//       CMB_END();
+//   }
+//
+// Constraints (documented in docs/PARITY.md): one CMB_* blocking macro per
+// source line; locals that live across a blocking call go in the frame.
+// ---------------------------------------------------------------------------
+
+#define CMB_BEGIN() switch (self->pc) { case 0:
+#define CMB_END() \
+    }             \
+    E.proc_finish(*self); \
+    return;
+
+#define CMB_SIG() (self->sig)
+
+// internal: yield and name the resumption point
+#define CMB_YIELD_()            \
+    self->pc = (int16_t)__LINE__; \
+    return;                     \
+    case __LINE__:              \
+        E.await_cleanup(*self);
+
+// hold for `dur` sim time (reference cmb_process_hold, cmb_process.c:452)
+#define CMB_HOLD(dur)                 \
+    do {                              \
+        E.hold_setup(*self, (dur));   \
+        CMB_YIELD_();                 \
+    } while (0)
+
+// wait for another process to finish (reference cmb_process_wait_process)
+#define CMB_WAIT_PROCESS(tgt)                      \
+    do {                                           \
+        if (E.wait_proc_setup(*self, (tgt))) {     \
+            CMB_YIELD_();                          \
+        } else {                                   \
+            self->sig = cmb::SIG_SUCCESS;          \
+        }                                          \
+    } while (0)
+
+// wait for a scheduled event to execute (reference cmb_process_wait_event)
+#define CMB_WAIT_EVENT(handle)                  \
+    do {                                        \
+        E.wait_event_setup(*self, (handle));    \
+        CMB_YIELD_();                           \
+    } while (0)
+
+// generic guarded-retry loop body shared by the toolkit macros
+#define CMB_GUARDED_(try_expr, gid_expr, demand, ctx)           \
+    for (;;) {                                                  \
+        if (try_expr) {                                         \
+            self->sig = cmb::SIG_SUCCESS;                       \
+            break;                                              \
+        }                                                       \
+        if (E.status != cmb::ST_OK) { self->sig = cmb::SIG_CANCELLED; break; } \
+        E.guard_wait(*self, (gid_expr), (demand), (ctx));       \
+        CMB_YIELD_();                                           \
+        if (self->sig != cmb::SIG_SUCCESS) break;               \
+        self->g_granted = 1;                                    \
+    }
+
+// object queue put/get (reference cmb_objectqueue_put/get)
+#define CMB_QPUT(qi, val)                                              \
+    CMB_GUARDED_(E.q_try_put((qi), *self, (val)), E.queues[qi].g_rear, \
+                 cmb::DEM_QSPACE, (uint32_t)(qi))
+
+#define CMB_QGET(qi, outp)                                              \
+    CMB_GUARDED_(E.q_try_get((qi), *self, (outp)), E.queues[qi].g_front, \
+                 cmb::DEM_QOBJ, (uint32_t)(qi))
+
+// priority queue put/get (reference cmb_priorityqueue_put/get)
+#define CMB_PQPUT(qi, val, pri)                                \
+    CMB_GUARDED_(E.pq_try_put((qi), *self, (val), (pri)),      \
+                 E.pqueues[qi].g_rear, cmb::DEM_PQSP, (uint32_t)(qi))
+
+#define CMB_PQGET(qi, outp)                                     \
+    CMB_GUARDED_(E.pq_try_get((qi), *self, (outp)),             \
+                 E.pqueues[qi].g_front, cmb::DEM_PQOBJ, (uint32_t)(qi))
+
+// resource acquire/release (reference cmb_resource_acquire/release,
+// cmb_resource.c:235-277)
+#define CMB_RES_ACQUIRE(ri)                                         \
+    CMB_GUARDED_(E.res_try_acquire((ri), *self), E.resources[ri].gid, \
+                 cmb::DEM_RES, (uint32_t)(ri))
+
+#define CMB_RES_RELEASE(ri) E.resource_release((ri))
+
+// pool acquire: greedy partial acquisition (reference
+// include/cmb_resourcepool.h:15-19); `remvar` is a frame lvalue tracking
+// the amount still wanted
+#define CMB_POOL_ACQUIRE(pi, amount, remvar)                          \
+    do {                                                              \
+        (remvar) = (amount);                                          \
+        for (;;) {                                                    \
+            (remvar) -= E.pool_try_take((pi), *self, (remvar));       \
+            if ((remvar) <= 0) { self->sig = cmb::SIG_SUCCESS; break; } \
+            if (E.status != cmb::ST_OK) { self->sig = cmb::SIG_CANCELLED; break; } \
+            E.guard_wait(*self, E.pools[pi].gid, cmb::DEM_POOL,       \
+                         (uint32_t)(pi));                             \
+            CMB_YIELD_();                                             \
+            if (self->sig != cmb::SIG_SUCCESS) break;                 \
+            self->g_granted = 1;                                      \
+        }                                                             \
+    } while (0)
+
+#define CMB_POOL_RELEASE(pi, amount) E.pool_release((pi), (amount))
+
+// buffer get/put of `amount` units (reference cmb_buffer_get/put)
+#define CMB_BUF_GET(bi, amount)                                           \
+    CMB_GUARDED_(E.buf_try_get((bi), *self, (amount)), E.buffers[bi].g_get, \
+                 cmb::DEM_BUF_GE,                                         \
+                 (uint32_t)(bi) | ((uint32_t)(amount) << 8))
+
+#define CMB_BUF_PUT(bi, amount)                                           \
+    CMB_GUARDED_(E.buf_try_put((bi), *self, (amount)), E.buffers[bi].g_put, \
+                 cmb::DEM_BUF_SP,                                         \
+                 (uint32_t)(bi) | ((uint32_t)(amount) << 8))
+
+// condition wait on a user demand predicate (reference cmb_condition_wait)
+#define CMB_COND_WAIT(ci, demand_id, ctx)                                \
+    do {                                                                 \
+        E.guard_wait(*self, E.conds[ci].gid, (uint8_t)(demand_id),       \
+                     (uint32_t)(ctx));                                   \
+        CMB_YIELD_();                                                    \
+    } while (0)
+
+}  // namespace cmb — note: macros are file-scope; namespace closed after

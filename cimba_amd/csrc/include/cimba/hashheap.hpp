@@ -1,0 +1,177 @@
+// cimba_amd future-event list: fixed-capacity binary min-heap keyed by
+// (time asc, priority desc, FIFO seq asc), with O(1)-amortized cancel by
+// handle through a slot back-map.
+//
+// Counterpart of reference src/cmi_hashheap.c (binary min-heap + open
+// addressing hash map in one allocation, pattern find/count/cancel,
+// cmi_hashheap.h:335-357).  Redesigned for gfx950: the reference needs a
+// hash map because its 64-bit handles are opaque; here the heap lives in
+// LDS with a bounded capacity, so a handle is (slot-generation, pool index)
+// and cancellation resolves by a direct scan over a small heap — no hash
+// map, no tombstones, no growth path in the hot loop.  Host-side large
+// models use the same structure with a bigger CAP (heap-allocated engine).
+#pragma once
+
+#include "config.hpp"
+
+namespace cmb {
+
+// Event payload carried inline in the heap (reference keeps 4 pointers;
+// we keep {kind, a, c, b} — kind = action id, a = subject (proc index or
+// similar), c = auxiliary u32 (guard id / timer slot), b = 64-bit payload
+// (signal value, object word)).
+struct EvEntry {
+    double t;        // activation time
+    uint64_t pseq;   // (INT16_MAX - priority) << 48 | seq  → min == next
+    uint64_t b;      // payload
+    uint32_t handle; // unique id for cancel/reschedule
+    uint16_t kind;
+    uint16_t a;
+    uint32_t c;
+    uint32_t pad_;
+};
+
+CMB_FORCEINLINE uint64_t ev_pseq(int priority, uint64_t seq) {
+    const uint16_t k = (uint16_t)(32767 - priority);
+    return ((uint64_t)k << 48) | (seq & UINT64_C(0xFFFFFFFFFFFF));
+}
+
+// lexicographic (t, pseq): earlier time first; at equal time higher
+// priority first, then FIFO (same comparator contract as reference
+// cmb_event.c:78-103)
+CMB_FORCEINLINE bool ev_less(const EvEntry& x, const EvEntry& y) {
+    if (x.t != y.t) return x.t < y.t;
+    return x.pseq < y.pseq;
+}
+
+template <int CAP>
+struct HashHeap {
+    EvEntry e[CAP];
+    int32_t n;
+
+    CMB_FORCEINLINE void reset() { n = 0; }
+    CMB_FORCEINLINE bool empty() const { return n == 0; }
+    CMB_FORCEINLINE bool full() const { return n == CAP; }
+    CMB_FORCEINLINE const EvEntry& top() const { return e[0]; }
+
+    CMB_FORCEINLINE void sift_up(int32_t i) {
+        EvEntry tmp = e[i];
+        while (i > 0) {
+            const int32_t p = (i - 1) >> 1;
+            if (!ev_less(tmp, e[p])) break;
+            e[i] = e[p];
+            i = p;
+        }
+        e[i] = tmp;
+    }
+
+    CMB_FORCEINLINE void sift_down(int32_t i) {
+        EvEntry tmp = e[i];
+        for (;;) {
+            int32_t c = 2 * i + 1;
+            if (c >= n) break;
+            if (c + 1 < n && ev_less(e[c + 1], e[c])) ++c;
+            if (!ev_less(e[c], tmp)) break;
+            e[i] = e[c];
+            i = c;
+        }
+        e[i] = tmp;
+    }
+
+    // returns false when full (caller converts to a trial abort)
+    CMB_FORCEINLINE bool push(const EvEntry& ev) {
+        if (n == CAP) return false;
+        e[n] = ev;
+        sift_up(n);
+        ++n;
+        return true;
+    }
+
+    CMB_FORCEINLINE EvEntry pop() {
+        EvEntry out = e[0];
+        --n;
+        if (n > 0) {
+            e[0] = e[n];
+            sift_down(0);
+        }
+        return out;
+    }
+
+    // cancel by handle; O(n) scan over a small heap (see header comment)
+    CMB_FORCEINLINE bool cancel(uint32_t handle, EvEntry* out = nullptr) {
+        for (int32_t i = 0; i < n; ++i) {
+            if (e[i].handle == handle) {
+                if (out) *out = e[i];
+                remove_at(i);
+                return true;
+            }
+        }
+        return false;
+    }
+
+    CMB_FORCEINLINE void remove_at(int32_t i) {
+        --n;
+        if (i == n) return;
+        e[i] = e[n];
+        sift_down(i);
+        sift_up(i);
+    }
+
+    // reschedule (reference cmb_event_reschedule): new time, keep payload
+    CMB_FORCEINLINE bool reschedule(uint32_t handle, double t, uint64_t pseq) {
+        for (int32_t i = 0; i < n; ++i) {
+            if (e[i].handle == handle) {
+                e[i].t = t;
+                e[i].pseq = pseq;
+                sift_down(i);
+                sift_up(i);
+                return true;
+            }
+        }
+        return false;
+    }
+
+    // wildcard pattern ops over (kind, a, b) — reference
+    // cmb_event_pattern_find/count/cancel (cmb_event.c:537-580) with
+    // CMB_ANY_* wildcards.  kind == 0xFFFF / a == 0xFFFF / b-match disabled
+    // via match_b=false act as wildcards.
+    CMB_FORCEINLINE int32_t pattern_count(uint16_t kind, uint16_t a, bool match_b,
+                                          uint64_t b) const {
+        int32_t cnt = 0;
+        for (int32_t i = 0; i < n; ++i) {
+            if ((kind == 0xFFFF || e[i].kind == kind) &&
+                (a == 0xFFFF || e[i].a == a) && (!match_b || e[i].b == b))
+                ++cnt;
+        }
+        return cnt;
+    }
+
+    CMB_FORCEINLINE int32_t pattern_cancel(uint16_t kind, uint16_t a, bool match_b,
+                                           uint64_t b) {
+        int32_t cnt = 0;
+        int32_t i = 0;
+        while (i < n) {
+            if ((kind == 0xFFFF || e[i].kind == kind) &&
+                (a == 0xFFFF || e[i].a == a) && (!match_b || e[i].b == b)) {
+                remove_at(i);
+                ++cnt;
+                // re-examine slot i (a new entry moved in)
+            } else {
+                ++i;
+            }
+        }
+        return cnt;
+    }
+
+    CMB_FORCEINLINE uint32_t pattern_find(uint16_t kind, uint16_t a, bool match_b,
+                                          uint64_t b) const {
+        for (int32_t i = 0; i < n; ++i) {
+            if ((kind == 0xFFFF || e[i].kind == kind) &&
+                (a == 0xFFFF || e[i].a == a) && (!match_b || e[i].b == b))
+                return e[i].handle;
+        }
+        return 0;  // 0 = no handle
+    }
+};
+
+}  // namespace cmb
