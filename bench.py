@@ -1,0 +1,142 @@
+#!/usr/bin/env python3
+"""cimba_amd flagship benchmark: M/M/1 multi-replication events/sec.
+
+Measures the BASELINE.json headline metric — simulated events/sec (whole
+node) on the M/M/1 multi-replication workload (reference
+benchmark/MM1_multi.c: arrival+service processes, unlimited queue,
+exponential interarrival 1/0.9 and service 1.0, avg system time ~ 10) —
+with the GPU-resident trial-per-wavefront engine.
+
+Contract: `python bench.py --gpus N --steps K --warmup W`; for N>1 the
+driver launches one rank per GPU via torch.distributed.run.  One step = a
+fixed batch of replications per GPU (weak scaling).  Rank 0 prints ONE
+JSON line.
+
+vs_baseline divides by 1.0e9 events/s — the reference's published
+whole-node aggregate (~25M ev/s/core x 32 cores ~ 0.8-1 G;
+we use the stated "~1 G" figure).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+BASELINE_EVENTS_PER_SEC = 1.0e9  # reference 3970X whole-node aggregate
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--trials", type=int, default=8192,
+                    help="replications per GPU per step")
+    ap.add_argument("--objects", type=int, default=100000,
+                    help="objects per replication")
+    ap.add_argument("--seed", type=lambda s: int(s, 0), default=0x34F05C64D7AD598F)
+    ap.add_argument("--host", action="store_true",
+                    help="debug: run the CPU host engine instead of the GPU")
+    args = ap.parse_args()
+
+    import cimba_amd as ca
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = world if world > 1 else args.gpus
+
+    dist = None
+    torch = None
+    if world > 1:
+        import torch  # noqa: F811
+        import torch.distributed as dist  # noqa: F811
+        backend = "nccl" if (not args.host and torch.cuda.is_available()) else "gloo"
+        dist.init_process_group(backend=backend)
+        if backend == "nccl":
+            torch.cuda.set_device(local_rank)
+
+    use_gpu = not args.host
+    if use_gpu and ca.gpu_device_count() <= local_rank:
+        print(json.dumps({"error": "no HIP device visible", "rank": rank}),
+              file=sys.stderr)
+        sys.exit(2)
+
+    def barrier_sync():
+        if use_gpu:
+            ca.gpu_sync()
+        if dist is not None:
+            dist.barrier()
+        if use_gpu:
+            ca.gpu_sync()
+
+    def one_step(step_idx):
+        seed = ca.fmix64((args.seed ^ (rank << 32)) + step_idx + 1)
+        if use_gpu:
+            r = ca.mm1_gpu(ntrials=args.trials, num_objects=args.objects,
+                           seed=seed, device=local_rank)
+        else:
+            r = ca.mm1_host(ntrials=args.trials, num_objects=args.objects,
+                            seed=seed, threads=0)
+        if r["trials_ok"] != args.trials:
+            raise RuntimeError(
+                f"rank {rank}: {args.trials - r['trials_ok']} trials aborted "
+                f"(status {r['first_bad_status']})")
+        return r["total_events"]
+
+    # warmup (untimed)
+    for i in range(args.warmup):
+        one_step(-1 - i)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    events = 0
+    for k in range(args.steps):
+        events += one_step(k)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max elapsed over ranks; sum events over ranks
+    if dist is not None:
+        import torch
+        te = torch.tensor([elapsed], dtype=torch.float64)
+        tv = torch.tensor([float(events)], dtype=torch.float64)
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        dist.all_reduce(tv, op=dist.ReduceOp.SUM)
+        elapsed = te.item()
+        events = int(tv.item())
+
+    if rank == 0:
+        value = events / elapsed
+        out = {
+            "metric": "sim_events_per_sec",
+            "value": value,
+            "unit": "events/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": value / BASELINE_EVENTS_PER_SEC,
+            "dtype": "fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "MM1_multi",
+                "trials_per_gpu_per_step": args.trials,
+                "objects_per_trial": args.objects,
+                "arrival_rate": 0.9,
+                "service_rate": 1.0,
+                "parallelism": f"trial-parallel dp{n_gpus}",
+                "engine": "trial-per-wavefront, LDS-resident",
+                "device": "gpu" if use_gpu else "cpu-host-debug",
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

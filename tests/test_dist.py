@@ -1,0 +1,76 @@
+"""Multi-process (gloo, world_size=2) tests of the distributed statistics
+reduction — the CPU-testable half of the RCCL/xGMI fan-out (SURVEY.md
+§5.8: cross-trial merge becomes a rank reduce)."""
+import multiprocessing as mp
+import os
+
+import numpy as np
+import pytest
+
+
+def _dist_worker(rank, world, port, q):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+        "RANK": str(rank),
+        "WORLD_SIZE": str(world),
+    })
+    import torch.distributed as dist
+
+    dist.init_process_group(backend="gloo", rank=rank, world_size=world)
+    import cimba_amd as ca
+    from cimba_amd.parallel import (allreduce_datasummary,
+                                    allreduce_wtdsummary, shard_range)
+
+    # each rank summarizes its shard of a known dataset
+    data = np.arange(1000, dtype=np.float64) * 0.5 + 3.0
+    lo, hi = shard_range(len(data), rank, world)
+    ds = ca.DataSummary()
+    for v in data[lo:hi]:
+        ds.add(v)
+    merged = allreduce_datasummary(ds)
+
+    ws = ca.WtdSummary()
+    for v in data[lo:hi]:
+        ws.add(v, 0.25 + (v % 1.0))
+    wmerged = allreduce_wtdsummary(ws)
+
+    if rank == 0:
+        q.put({
+            "n": merged.count(),
+            "mean": merged.mean(),
+            "var": merged.variance(),
+            "wmean": wmerged.mean(),
+            "wsum": wmerged.sumw(),
+        })
+    dist.destroy_process_group()
+
+
+def test_allreduce_summaries_gloo():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29511
+    procs = [ctx.Process(target=_dist_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = q.get(timeout=180)
+    for p in procs:
+        p.join(timeout=60)
+    data = np.arange(1000, dtype=np.float64) * 0.5 + 3.0
+    w = 0.25 + (data % 1.0)
+    assert res["n"] == 1000
+    assert abs(res["mean"] - data.mean()) < 1e-12
+    assert abs(res["var"] - data.var(ddof=1)) < 1e-9
+    assert abs(res["wmean"] - (data * w).sum() / w.sum()) < 1e-12
+    assert abs(res["wsum"] - w.sum()) < 1e-12
+
+
+def test_shard_range():
+    from cimba_amd.parallel import shard_range
+
+    covered = []
+    for r in range(3):
+        lo, hi = shard_range(10, r, 3)
+        covered.extend(range(lo, hi))
+    assert covered == list(range(10))
