@@ -29,7 +29,7 @@ using EngMM1 = Engine<MM1>;
 // with room for multiple resident workgroups.
 static_assert(sizeof(EngMM1) * WPB < 60 * 1024, "MM1 engine too big for LDS plan");
 
-__global__ __launch_bounds__(WPB * 64) void mm1_kernel(
+__global__ __launch_bounds__(WPB * 64) __attribute__((flatten)) void mm1_kernel(
     MM1::Params P, uint64_t master_seed, uint32_t ntrials, double until,
     uint64_t max_events, MM1::Result* __restrict__ out) {
     __shared__ EngMM1 eng[WPB];

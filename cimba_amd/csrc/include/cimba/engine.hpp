@@ -206,11 +206,11 @@ constexpr int32_t CMB_UNLIMITED = 0x7FFFFFFF;
 // Default model hooks: models inherit and override what they use.
 struct ModelBase {
     template <class E_>
-    CMB_HD static bool demand(E_&, int /*pidx*/, uint8_t /*kind*/, uint32_t /*ctx*/) {
+    CMB_FORCEINLINE static bool demand(E_&, int /*pidx*/, uint8_t /*kind*/, uint32_t /*ctx*/) {
         return false;
     }
     template <class E_>
-    CMB_HD static void on_event(E_&, const EvEntry&) {}
+    CMB_FORCEINLINE static void on_event(E_&, const EvEntry&) {}
 };
 
 template <int N>
@@ -264,7 +264,7 @@ struct Engine {
 
     // ---- lifecycle --------------------------------------------------------
 
-    CMB_HD void init(const Params* p, uint64_t trial_seed, uint32_t tidx,
+    CMB_FORCEINLINE void init(const Params* p, uint64_t trial_seed, uint32_t tidx,
                      double start_time = 0.0) {
         now = start_time;
         ev_dispatched = 0;
@@ -383,7 +383,7 @@ struct Engine {
 
     // ---- process control ---------------------------------------------------
 
-    CMB_HD void proc_init(int pidx, uint8_t func, int priority) {
+    CMB_FORCEINLINE void proc_init(int pidx, uint8_t func, int priority) {
         ProcT& p = procs[pidx];
         cmb_assert_debug(p.state == PS_UNINIT);
         p.state = PS_READY;
@@ -394,7 +394,7 @@ struct Engine {
 
     // non-blocking start: schedules a start event (reference
     // cmb_process_start, cmb_process.c:247-254)
-    CMB_HD void proc_start(int pidx, double delay = 0.0) {
+    CMB_FORCEINLINE void proc_start(int pidx, double delay = 0.0) {
         ProcT& p = procs[pidx];
         cmb_assert_debug(p.state == PS_READY);
         schedule(EV_PROC_START, (uint16_t)pidx, 0, 0, now + delay, p.priority);
@@ -402,7 +402,7 @@ struct Engine {
 
     // non-blocking interrupt: wake the target's current wait with `sg`
     // (reference cmb_process_interrupt)
-    CMB_HD void proc_interrupt(int pidx, sig_t sg) {
+    CMB_FORCEINLINE void proc_interrupt(int pidx, sig_t sg) {
         ProcT& p = procs[pidx];
         if (p.state != PS_RUNNING || p.await_kind == AW_NONE) return;
         schedule(EV_RESUME, (uint16_t)pidx, p.epoch, (uint64_t)sg, now, p.priority);
@@ -410,7 +410,7 @@ struct Engine {
 
     // kill a process: drop resources, cancel awaitables, wake waiters
     // (reference kill path, cmb_process.c:979-1004)
-    CMB_HD void proc_stop(int pidx) {
+    CMB_FORCEINLINE void proc_stop(int pidx) {
         ProcT& p = procs[pidx];
         if (p.state == PS_FINISHED || p.state == PS_UNINIT) return;
         for (int t = 0; t < Cfg::TIMERS; ++t) timer_cancel(p, t);
@@ -422,7 +422,7 @@ struct Engine {
         finish_common(p, SIG_STOPPED);
     }
 
-    CMB_HD void proc_priority_set(int pidx, int priority) {
+    CMB_FORCEINLINE void proc_priority_set(int pidx, int priority) {
         procs[pidx].priority = (int16_t)priority;
     }
 
@@ -450,7 +450,7 @@ struct Engine {
     // release everything a killed/finished process still holds (reference
     // cmi_holdable drop polymorphism, src/cmi_holdable.h:53-78); bounded
     // scans over the small toolkit arrays replace the intrusive list.
-    CMB_HD void drop_held(int pidx) {
+    CMB_FORCEINLINE void drop_held(int pidx) {
         for (int r = 0; r < NR; ++r) {
             if (resources[r].holder == (int16_t)pidx) resource_release(r);
         }
@@ -532,7 +532,7 @@ struct Engine {
         ++n_event_waiters;
     }
 
-    CMB_HD void wake_event_waiters(uint32_t handle, sig_t sg) {
+    CMB_FORCEINLINE void wake_event_waiters(uint32_t handle, sig_t sg) {
         for (int i = 0; i < Cfg::MAX_PROC; ++i) {
             ProcT& p = procs[i];
             if (p.await_kind == AW_EVENT && p.await_key == handle) {
@@ -587,7 +587,7 @@ struct Engine {
         return best;
     }
 
-    CMB_HD bool eval_demand(const ProcT& p) {
+    CMB_FORCEINLINE bool eval_demand(const ProcT& p) {
         const uint32_t ctx = p.demand_ctx;
         switch (p.demand_kind) {
             case DEM_QSPACE: return queues[ctx].len < queues[ctx].limit;
@@ -609,7 +609,7 @@ struct Engine {
     // evaluate the front waiter's demand; if satisfied, schedule a grant
     // event (reference cmb_resourceguard.c:240-260 semantics: the grant is
     // a hint — the woken process revalidates in its acquire loop)
-    CMB_HD bool guard_signal(int gid) {
+    CMB_FORCEINLINE bool guard_signal(int gid) {
         const int w = guard_front(gid);
         bool granted = false;
         if (w >= 0 && eval_demand(procs[w])) {
@@ -625,7 +625,7 @@ struct Engine {
 
     // condition signal: wake EVERY satisfied waiter (reference
     // include/cmb_condition.h:17-24)
-    CMB_HD uint64_t condition_signal(int ci) {
+    CMB_FORCEINLINE uint64_t condition_signal(int ci) {
         const int gid = conds[ci].gid;
         uint64_t cnt = 0;
         for (int16_t i = guards[gid].head; i >= 0; i = procs[i].gnext) {
@@ -648,7 +648,7 @@ struct Engine {
         }
     }
 
-    CMB_HD bool q_try_put(int qi, ProcT& p, uint64_t val) {
+    CMB_FORCEINLINE bool q_try_put(int qi, ProcT& p, uint64_t val) {
         ObjQueue<Cfg::QCAP>& q = queues[qi];
         const bool may = p.g_granted || guards[q.g_rear].empty();
         p.g_granted = 0;
@@ -661,7 +661,7 @@ struct Engine {
         return true;
     }
 
-    CMB_HD bool q_try_get(int qi, ProcT& p, uint64_t* out) {
+    CMB_FORCEINLINE bool q_try_get(int qi, ProcT& p, uint64_t* out) {
         ObjQueue<Cfg::QCAP>& q = queues[qi];
         const bool may = p.g_granted || guards[q.g_front].empty();
         p.g_granted = 0;
@@ -676,7 +676,7 @@ struct Engine {
 
     CMB_FORCEINLINE int64_t q_length(int qi) const { return queues[qi].len; }
 
-    CMB_HD bool pq_try_put(int qi, ProcT& p, uint64_t val, int priority) {
+    CMB_FORCEINLINE bool pq_try_put(int qi, ProcT& p, uint64_t val, int priority) {
         PrioQueue<Cfg::PQCAP>& q = pqueues[qi];
         const bool may = p.g_granted || guards[q.g_rear].empty();
         p.g_granted = 0;
@@ -698,7 +698,7 @@ struct Engine {
         return true;
     }
 
-    CMB_HD bool pq_try_get(int qi, ProcT& p, uint64_t* out) {
+    CMB_FORCEINLINE bool pq_try_get(int qi, ProcT& p, uint64_t* out) {
         PrioQueue<Cfg::PQCAP>& q = pqueues[qi];
         const bool may = p.g_granted || guards[q.g_front].empty();
         p.g_granted = 0;
@@ -723,7 +723,7 @@ struct Engine {
         return true;
     }
 
-    CMB_HD bool res_try_acquire(int ri, ProcT& p) {
+    CMB_FORCEINLINE bool res_try_acquire(int ri, ProcT& p) {
         Resource& r = resources[ri];
         const bool may = p.g_granted || guards[r.gid].empty();
         p.g_granted = 0;
@@ -736,7 +736,7 @@ struct Engine {
         return true;
     }
 
-    CMB_HD void resource_release(int ri) {
+    CMB_FORCEINLINE void resource_release(int ri) {
         Resource& r = resources[ri];
         cmb_assert_debug(r.holder >= 0);
         if (r.recording) {
@@ -747,7 +747,7 @@ struct Engine {
         guard_signal(r.gid);
     }
 
-    CMB_HD int32_t pool_try_take(int pi, ProcT& p, int32_t want) {
+    CMB_FORCEINLINE int32_t pool_try_take(int pi, ProcT& p, int32_t want) {
         Pool& pl = pools[pi];
         const bool may = p.g_granted || guards[pl.gid].empty();
         p.g_granted = 0;
@@ -764,7 +764,7 @@ struct Engine {
         return take;
     }
 
-    CMB_HD void pool_release(int pi, int32_t amount) {
+    CMB_FORCEINLINE void pool_release(int pi, int32_t amount) {
         Pool& pl = pools[pi];
         cmb_assert_debug(pl.in_use >= amount);
         if (pl.recording) {
@@ -775,7 +775,7 @@ struct Engine {
         guard_signal(pl.gid);
     }
 
-    CMB_HD bool buf_try_get(int bi, ProcT& p, int64_t amount) {
+    CMB_FORCEINLINE bool buf_try_get(int bi, ProcT& p, int64_t amount) {
         Buffer& b = buffers[bi];
         const bool may = p.g_granted || guards[b.g_get].empty();
         p.g_granted = 0;
@@ -789,7 +789,7 @@ struct Engine {
         return true;
     }
 
-    CMB_HD bool buf_try_put(int bi, ProcT& p, int64_t amount) {
+    CMB_FORCEINLINE bool buf_try_put(int bi, ProcT& p, int64_t amount) {
         Buffer& b = buffers[bi];
         const bool may = p.g_granted || guards[b.g_put].empty();
         p.g_granted = 0;
@@ -811,7 +811,7 @@ struct Engine {
         Model::step(*this, pidx);
     }
 
-    CMB_HD bool dispatch_one() {
+    CMB_FORCEINLINE bool dispatch_one() {
         if (status != ST_OK || evq.empty()) return false;
         const EvEntry ev = evq.pop();
         now = ev.t;
@@ -876,7 +876,7 @@ struct Engine {
 
     // run until the event queue drains (reference cmb_event_queue_execute,
     // cmb_event.c:402) or a limit is hit
-    CMB_HD void run(double until, uint64_t max_events) {
+    CMB_FORCEINLINE void run(double until, uint64_t max_events) {
         while (status == ST_OK && !evq.empty()) {
             if (evq.top().t > until) {
                 now = until;

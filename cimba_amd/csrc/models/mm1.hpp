@@ -65,7 +65,7 @@ struct MM1 : cmb::ModelBase {
 
     // reference MM1_multi.c:56-69 arrivalfunc
     template <class E_>
-    CMB_HD static void arrival(E_& E, typename E_::ProcT* self) {
+    CMB_FORCEINLINE static void arrival(E_& E, typename E_::ProcT* self) {
         const Params& P = *E.params;
         ArrFrame& f = E.frames[0].arr;
         CMB_BEGIN();
@@ -79,7 +79,7 @@ struct MM1 : cmb::ModelBase {
 
     // reference MM1_multi.c:71-88 servicefunc; accumulators are in acc_of(E)
     template <class E_>
-    CMB_HD static void service(E_& E, typename E_::ProcT* self) {
+    CMB_FORCEINLINE static void service(E_& E, typename E_::ProcT* self) {
         const Params& P = *E.params;
         SrvFrame& f = E.frames[1].srv;
         CMB_BEGIN();
@@ -94,12 +94,12 @@ struct MM1 : cmb::ModelBase {
     }
 
     template <class E_>
-    CMB_HD static Acc& acc_of(E_& E) {
+    CMB_FORCEINLINE static Acc& acc_of(E_& E) {
         return E.frames[0].acc;
     }
 
     template <class E_>
-    CMB_HD static void step(E_& E, int pidx) {
+    CMB_FORCEINLINE static void step(E_& E, int pidx) {
         auto* self = &E.procs[pidx];
         if (self->func == F_ARRIVAL)
             arrival(E, self);
@@ -108,7 +108,7 @@ struct MM1 : cmb::ModelBase {
     }
 
     template <class E_>
-    CMB_HD static void setup(E_& E) {
+    CMB_FORCEINLINE static void setup(E_& E) {
         E.queues[0].limit = cmb::CMB_UNLIMITED;
         acc_of(E).cnt = 0;
         acc_of(E).sum = 0.0;
@@ -119,7 +119,7 @@ struct MM1 : cmb::ModelBase {
     }
 
     template <class E_>
-    CMB_HD static void finish(E_& E, Result& r) {
+    CMB_FORCEINLINE static void finish(E_& E, Result& r) {
         r.obj_cnt = acc_of(E).cnt;
         r.sum_wait = acc_of(E).sum;
         r.events = E.ev_dispatched;
