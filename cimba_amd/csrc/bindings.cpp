@@ -10,6 +10,7 @@
 #include "cimba/runner.hpp"
 #include "cimba/stats.hpp"
 #include "../models/mm1.hpp"
+#include "../models/scenarios.hpp"
 
 #include <string>
 #include <vector>
@@ -34,6 +35,38 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
                       double until, uint64_t max_events, Mm1GpuOut* out);
 int cimba_gpu_device_count(int* n);
 int cimba_gpu_sync(void);
+int cimba_scenario_gpu_run(int which, void* result_out);
+}
+
+using cmb_models::Scenario;
+
+static py::dict scenario_result_to_dict(const Scenario::Result& r) {
+    py::list trace;
+    for (int i = 0; i < r.n; ++i)
+        trace.append(py::make_tuple(r.ev[i].t, r.ev[i].code));
+    py::dict d;
+    d["trace"] = trace;
+    d["status"] = r.status;
+    d["events"] = r.events;
+    return d;
+}
+
+static py::dict scenario_host(int which) {
+    Scenario::Params p{which};
+    auto eng = std::make_unique<Engine<Scenario>>();
+    eng->init(&p, 123, 0);
+    Scenario::setup(*eng);
+    eng->run(1.0e308, 100000);
+    Scenario::Result r;
+    Scenario::finish(*eng, r);
+    return scenario_result_to_dict(r);
+}
+
+static py::dict scenario_gpu(int which) {
+    Scenario::Result r;
+    int rc = cimba_scenario_gpu_run(which, &r);
+    if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
+    return scenario_result_to_dict(r);
 }
 
 static py::dict mm1_host(uint64_t ntrials, uint64_t num_objects, double arr_rate,
@@ -208,6 +241,8 @@ PYBIND11_MODULE(_C, m) {
     m.def("mm1_gpu", &mm1_gpu, py::arg("ntrials"), py::arg("num_objects"),
           py::arg("arr_rate") = 0.9, py::arg("srv_rate") = 1.0,
           py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0);
+    m.def("scenario_host", &scenario_host, py::arg("which"));
+    m.def("scenario_gpu", &scenario_gpu, py::arg("which"));
     m.def("gpu_device_count", &gpu_device_count);
     m.def("gpu_sync", []() { return cimba_gpu_sync(); });
 

@@ -14,12 +14,14 @@
 #include <hip/hip_runtime.h>
 
 #include "../models/mm1.hpp"
+#include "../models/scenarios.hpp"
 #include "../include/cimba/runner.hpp"
 
 namespace {
 
 using cmb::Engine;
 using cmb_models::MM1;
+using cmb_models::Scenario;
 
 constexpr int WPB = 4;  // waves (= trials) per workgroup
 
@@ -44,6 +46,18 @@ __global__ __launch_bounds__(WPB * 64) __attribute__((flatten)) void mm1_kernel(
         E.run(until, max_events);
         MM1::finish(E, out[trial]);
     }
+}
+
+// single-trial semantic-parity kernel: runs one Scenario trial on-device;
+// the trace must match the host engine exactly (tests/test_gpu.py)
+__global__ __launch_bounds__(64) __attribute__((flatten)) void scenario_kernel(
+    Scenario::Params P, Scenario::Result* __restrict__ out) {
+    __shared__ Engine<Scenario> eng;
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    eng.init(&P, 123, 0);
+    Scenario::setup(eng);
+    eng.run(1.0e308, 100000);
+    Scenario::finish(eng, *out);
 }
 
 #define HIP_TRY(x)                                    \
@@ -113,6 +127,18 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
         else if (out->first_bad_status == 0)
             out->first_bad_status = res[i].status;
     }
+    return 0;
+}
+
+int cimba_scenario_gpu_run(int which, void* result_out) {
+    Scenario::Params P{which};
+    Scenario::Result* d_out = nullptr;
+    HIP_TRY(hipMalloc(&d_out, sizeof(Scenario::Result)));
+    hipLaunchKernelGGL(scenario_kernel, dim3(1), dim3(64), 0, 0, P, d_out);
+    HIP_TRY(hipGetLastError());
+    HIP_TRY(hipMemcpy(result_out, d_out, sizeof(Scenario::Result),
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipFree(d_out));
     return 0;
 }
 

@@ -205,6 +205,7 @@ constexpr int32_t CMB_UNLIMITED = 0x7FFFFFFF;
 
 // Default model hooks: models inherit and override what they use.
 struct ModelBase {
+    struct Globals {};  // model-wide per-trial state (override as needed)
     template <class E_>
     CMB_FORCEINLINE static bool demand(E_&, int /*pidx*/, uint8_t /*kind*/, uint32_t /*ctx*/) {
         return false;
@@ -252,6 +253,7 @@ struct Engine {
     Rng rng;
 
     HashHeap<Cfg::MAX_EV> evq;
+    typename Model::Globals globals;  // model-wide per-trial state
     ProcT procs[Cfg::MAX_PROC];
     Frame frames[Cfg::MAX_PROC];
     Guard guards[NGUARD];
@@ -424,6 +426,61 @@ struct Engine {
 
     CMB_FORCEINLINE void proc_priority_set(int pidx, int priority) {
         procs[pidx].priority = (int16_t)priority;
+    }
+
+    // allocate a process slot, recycling finished ones — the device-native
+    // replacement for reference cmb_process_create()'s heap allocation
+    // (dynamic entities draw from a bounded slot pool, cf. cmi_mempool)
+    CMB_FORCEINLINE int proc_alloc() {
+        for (int i = 0; i < Cfg::MAX_PROC; ++i) {
+            ProcT& p = procs[i];
+            if (p.state == PS_UNINIT || p.state == PS_FINISHED) {
+                p.pc = 0;
+                p.state = PS_UNINIT;
+                p.await_kind = AW_NONE;
+                p.g_granted = 0;
+                p.await_key = 0;
+                p.sig = SIG_SUCCESS;
+                p.gid = -1;
+                p.gnext = -1;
+                p.waiters_head = -1;
+                p.pnext = -1;
+                for (int t = 0; t < Cfg::TIMERS; ++t) p.timers[t] = 0;
+                return i;
+            }
+        }
+        return -1;
+    }
+
+    // preemptive takeover (reference cmb_resource_preempt): returns true if
+    // the caller now holds the resource; a lower-priority holder is kicked
+    // with SIG_PREEMPTED delivered to its current blocking call
+    CMB_FORCEINLINE bool res_try_preempt(int ri, ProcT& p) {
+        Resource& r = resources[ri];
+        if (r.holder < 0) {
+            p.g_granted = 1;  // bypass the no-queue-jump check
+            return res_try_acquire(ri, p);
+        }
+        ProcT& h = procs[r.holder];
+        if (h.priority >= p.priority) return false;
+        const int old = r.holder;
+        if (r.recording) {
+            r.busy.add(1.0, now - r.t_last);
+            r.t_last = now;
+        }
+        r.holder = (int16_t)pidx_of(&p);
+        proc_interrupt(old, SIG_PREEMPTED);
+        return true;
+    }
+
+    // register a condition as an observer of another guard (reference
+    // cmb_resourceguard_register, include/cmb_resourceguard.h:168-179):
+    // any signal on that guard re-evaluates the condition's waiters
+    CMB_FORCEINLINE void condition_observe(int ci, int gid) {
+        guards[gid].observer = (int16_t)ci;
+    }
+    CMB_FORCEINLINE void condition_unobserve(int gid) {
+        guards[gid].observer = -1;
     }
 
     // called by CMB_END / early exit
@@ -988,6 +1045,12 @@ struct Engine {
                  cmb::DEM_RES, (uint32_t)(ri))
 
 #define CMB_RES_RELEASE(ri) E.resource_release((ri))
+
+// preemptive acquire: takes the resource from any lower-priority holder
+// (reference cmb_resource_preempt); falls back to a priority-ordered wait
+#define CMB_RES_PREEMPT(ri)                                            \
+    CMB_GUARDED_(E.res_try_preempt((ri), *self), E.resources[ri].gid,  \
+                 cmb::DEM_RES, (uint32_t)(ri))
 
 // pool acquire: greedy partial acquisition (reference
 // include/cmb_resourcepool.h:15-19); `remvar` is a frame lvalue tracking

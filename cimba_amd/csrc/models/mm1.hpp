@@ -48,17 +48,15 @@ struct MM1 : cmb::ModelBase {
     struct SrvFrame {
         uint64_t obj;  // bit-cast arrival time
     };
-    struct Acc {
-        uint64_t cnt;
-        double sum;
-    };
     // per-process persistent locals; frames[0] = arrival, frames[1] = service
-    // (the Acc block is only used in frames[0], by the service process —
-    // a struct, not a union, so nothing overlaps)
-    struct Frame {
+    union Frame {
         ArrFrame arr;
         SrvFrame srv;
-        Acc acc;
+    };
+    // model-wide per-trial accumulators
+    struct Globals {
+        uint64_t cnt;
+        double sum;
     };
 
     enum Func : uint8_t { F_ARRIVAL = 0, F_SERVICE = 1 };
@@ -94,8 +92,8 @@ struct MM1 : cmb::ModelBase {
     }
 
     template <class E_>
-    CMB_FORCEINLINE static Acc& acc_of(E_& E) {
-        return E.frames[0].acc;
+    CMB_FORCEINLINE static Globals& acc_of(E_& E) {
+        return E.globals;
     }
 
     template <class E_>

@@ -1,0 +1,61 @@
+"""Golden semantic traces for the process/guard/toolkit layer.
+
+Each expectation below is derived BY HAND from the engine contract
+(reference semantics): event ordering is (time asc,
+priority desc, FIFO), guards grant front-waiter-only with stale-grant
+pass-on, blocking calls return signals per include/cmb_process.h:60-100.
+Codes: pidx*1000 + tag (tag: 1 start, 2 wake, 3 done, 10 acquire,
+11 release, 20+v got-value, 100+|sig| signal, 900 user event).
+"""
+import pytest
+
+import cimba_amd as ca
+
+EXPECTED = {
+    # hold ordering: higher event priority first at equal times
+    1: [(0.0, 1001), (0.0, 1), (1.5, 1002), (1.5, 2), (3.0, 1003), (3.0, 3)],
+    # interrupt delivers user signal 42 into a hold
+    2: [(0.0, 1), (1.0, 1003), (1.0, 442)],
+    # resource grants by (priority desc, FIFO): p2 (pri 9) beats p1
+    3: [(0.0, 10), (5.0, 11), (5.0, 2010), (6.0, 2011), (6.0, 1010), (7.0, 1011)],
+    # resource wait timeout -> SIG_TIMEOUT (105), guard entry removed
+    4: [(0.0, 10), (2.5, 1105), (5.0, 11)],
+    # preempt: taker acquires, holder's hold returns SIG_PREEMPTED (101)
+    5: [(0.0, 10), (1.0, 1010), (1.0, 101), (3.0, 1011)],
+    # pool greedy partial acquisition: 3 now + 3 when released
+    6: [(0.0, 10), (3.0, 11), (3.0, 1010), (4.0, 1011)],
+    # buffer: consumer unblocks when level first reaches 5 (t=2)
+    7: [(1.0, 11), (2.0, 11), (2.0, 1020), (3.0, 11)],
+    # condition: each signal wakes exactly the satisfied waiters
+    8: [(1.0, 2), (2.0, 1002)],
+    # stop: killed holder's resource passes on; waiter gets SIG_STOPPED (103)
+    9: [(0.0, 10), (1.0, 2003), (1.0, 3010), (1.0, 1103), (2.0, 3011)],
+    # priority queue: (pri desc, FIFO) -> values 2, 3, then 1
+    10: [(1.0, 1022), (1.0, 1023), (1.0, 1021)],
+    # wait_event: waiter woken when the event executes
+    11: [(5.0, 7900), (5.0, 100)],
+    # event cancel wakes waiters with SIG_CANCELLED (104)
+    12: [(2.0, 104)],
+    # stale grant passes on: granted waiter times out at the same
+    # timestamp, the grant must not be lost -> next waiter gets the object
+    13: [(1.0, 11), (1.0, 1105), (1.0, 2097)],
+}
+
+
+@pytest.mark.parametrize("which", sorted(EXPECTED))
+def test_scenario_host(which):
+    r = ca._C.scenario_host(which)
+    assert r["status"] == 0
+    assert r["trace"] == EXPECTED[which], (
+        f"scenario {which}: {r['trace']} != {EXPECTED[which]}")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("which", sorted(EXPECTED))
+def test_scenario_gpu_matches_host(which):
+    g = ca._C.scenario_gpu(which)
+    assert g["status"] == 0
+    assert g["trace"] == EXPECTED[which], (
+        f"GPU scenario {which}: {g['trace']} != {EXPECTED[which]}")
+    h = ca._C.scenario_host(which)
+    assert g["events"] == h["events"]
