@@ -37,6 +37,10 @@ def main():
     ap.add_argument("--seed", type=lambda s: int(s, 0), default=0x34F05C64D7AD598F)
     ap.add_argument("--host", action="store_true",
                     help="debug: run the CPU host engine instead of the GPU")
+    ap.add_argument("--model", choices=["mm1", "mg1", "jobshop"],
+                    default="mm1",
+                    help="mm1 = headline benchmark; mg1/jobshop = "
+                         "BASELINE configs 3-4")
     args = ap.parse_args()
 
     import cimba_amd as ca
@@ -72,12 +76,23 @@ def main():
 
     def one_step(step_idx):
         seed = ca.fmix64((args.seed ^ (rank << 32)) + step_idx + 1)
-        if use_gpu:
-            r = ca.mm1_gpu(ntrials=args.trials, num_objects=args.objects,
-                           seed=seed, device=local_rank)
+        if args.model == "mm1":
+            fn = ca.mm1_gpu if use_gpu else ca.mm1_host
+            kw = dict(ntrials=args.trials, num_objects=args.objects, seed=seed)
+        elif args.model == "mg1":
+            fn = ca.mg1_gpu if use_gpu else ca.mg1_host
+            kw = dict(ntrials=args.trials, num_objects=args.objects,
+                      arr_rate=0.8, srv_mean=1.0, srv_scv=0.25, dist=3,
+                      seed=seed)
         else:
-            r = ca.mm1_host(ntrials=args.trials, num_objects=args.objects,
-                            seed=seed, threads=0)
+            fn = ca.jobshop_gpu if use_gpu else ca.jobshop_host
+            kw = dict(ntrials=args.trials, entities=args.objects, njobs=24,
+                      seed=seed)
+        if use_gpu:
+            kw["device"] = local_rank
+        else:
+            kw["threads"] = 0
+        r = fn(**kw)
         if r["trials_ok"] != args.trials:
             raise RuntimeError(
                 f"rank {rank}: {args.trials - r['trials_ok']} trials aborted "
@@ -122,7 +137,8 @@ def main():
             "dtype": "fp64",
             "data": "synthetic",
             "config": {
-                "model": "MM1_multi",
+                "model": {"mm1": "MM1_multi", "mg1": "MG1_resource",
+                          "jobshop": "JobShop_pools"}[args.model],
                 "trials_per_gpu_per_step": args.trials,
                 "objects_per_trial": args.objects,
                 "arrival_rate": 0.9,

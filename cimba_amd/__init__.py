@@ -21,12 +21,16 @@ except ImportError:  # build in-tree on first use (hipcc cross-compiles on CPU)
     _build.build()
     from . import _C  # noqa: F401
 
-from ._C import (  # noqa: F401,E402
+from ._C import (  # noqa: E402  # noqa: F401,E402
     DataSummary,
     WtdSummary,
     fmix64,
     gpu_device_count,
     gpu_sync,
+    jobshop_gpu,
+    jobshop_host,
+    mg1_gpu,
+    mg1_host,
     mm1_gpu,
     mm1_host,
     rng_sample,

@@ -10,6 +10,8 @@
 #include "cimba/runner.hpp"
 #include "cimba/stats.hpp"
 #include "../models/mm1.hpp"
+#include "../models/mg1.hpp"
+#include "../models/jobshop.hpp"
 #include "../models/scenarios.hpp"
 
 #include <string>
@@ -36,6 +38,145 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
 int cimba_gpu_device_count(int* n);
 int cimba_gpu_sync(void);
 int cimba_scenario_gpu_run(int which, void* result_out);
+int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
+                      int device, double* elapsed_ms, void* results_out);
+int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
+                          int device, double* elapsed_ms, void* results_out);
+}
+
+using cmb_models::JobShop;
+using cmb_models::MG1;
+
+static py::dict mg1_aggregate(const std::vector<MG1::Result>& res,
+                              double elapsed_ms) {
+    uint64_t ev = 0, objs = 0, ok = 0;
+    double sys = 0.0, qw = 0.0;
+    int32_t bad = 0;
+    py::list per_trial;
+    for (auto& r : res) {
+        ev += r.events;
+        objs += r.obj_cnt;
+        sys += r.sum_system;
+        qw += r.sum_queue;
+        if (r.status == 0)
+            ++ok;
+        else if (!bad)
+            bad = r.status;
+        per_trial.append(r.obj_cnt ? r.sum_system / (double)r.obj_cnt : 0.0);
+    }
+    py::dict d;
+    d["total_events"] = ev;
+    d["total_objects"] = objs;
+    d["trials_ok"] = ok;
+    d["first_bad_status"] = bad;
+    d["avg_system_time"] = objs ? sys / (double)objs : 0.0;
+    d["avg_queue_time"] = objs ? qw / (double)objs : 0.0;
+    d["per_trial_avg"] = per_trial;
+    if (elapsed_ms >= 0) {
+        d["elapsed_ms"] = elapsed_ms;
+        d["events_per_sec"] =
+            elapsed_ms > 0 ? (double)ev / (elapsed_ms * 1e-3) : 0.0;
+    }
+    return d;
+}
+
+static py::dict mg1_host(uint64_t ntrials, uint64_t num_objects,
+                         double arr_rate, double srv_mean, double srv_scv,
+                         int dist, uint64_t seed, int threads) {
+    MG1::Params p{1.0 / arr_rate, srv_mean, srv_scv, num_objects, dist, 0};
+    std::vector<MG1::Result> res(ntrials);
+    {
+        py::gil_scoped_release nogil;
+        run_host<MG1>(p, seed, ntrials, threads, res.data());
+    }
+    return mg1_aggregate(res, -1.0);
+}
+
+static py::dict mg1_gpu(uint64_t ntrials, uint64_t num_objects,
+                        double arr_rate, double srv_mean, double srv_scv,
+                        int dist, uint64_t seed, int device) {
+    MG1::Params p{1.0 / arr_rate, srv_mean, srv_scv, num_objects, dist, 0};
+    std::vector<MG1::Result> res(ntrials);
+    double ms = 0.0;
+    int rc;
+    {
+        py::gil_scoped_release nogil;
+        rc = cimba_mg1_gpu_run(ntrials, &p, seed, device, &ms, res.data());
+    }
+    if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
+    return mg1_aggregate(res, ms);
+}
+
+static py::dict jobshop_aggregate(const std::vector<JobShop::Result>& res,
+                                  double elapsed_ms) {
+    uint64_t ev = 0, done = 0, ok = 0;
+    double makespan = 0.0;
+    double busy[JobShop::NST] = {0, 0, 0};
+    int32_t bad = 0;
+    for (auto& r : res) {
+        ev += r.events;
+        done += r.completed;
+        makespan += r.makespan;
+        for (int s = 0; s < JobShop::NST; ++s) busy[s] += r.busy_time[s];
+        if (r.status == 0)
+            ++ok;
+        else if (!bad)
+            bad = r.status;
+    }
+    py::dict d;
+    d["total_events"] = ev;
+    d["total_completed"] = done;
+    d["trials_ok"] = ok;
+    d["first_bad_status"] = bad;
+    d["mean_makespan"] = res.empty() ? 0.0 : makespan / (double)res.size();
+    py::list ub;
+    for (int s = 0; s < JobShop::NST; ++s)
+        ub.append(makespan > 0 ? busy[s] / makespan : 0.0);  // mean units busy
+    d["mean_units_busy"] = ub;
+    if (elapsed_ms >= 0) {
+        d["elapsed_ms"] = elapsed_ms;
+        d["events_per_sec"] =
+            elapsed_ms > 0 ? (double)ev / (elapsed_ms * 1e-3) : 0.0;
+    }
+    return d;
+}
+
+static JobShop::Params make_jobshop_params(uint64_t entities, int njobs,
+                                           double think_mean) {
+    JobShop::Params p;
+    p.total_entities = entities;
+    p.think_mean = think_mean;
+    p.srv_mean[0] = 1.0;
+    p.srv_mean[1] = 0.7;
+    p.srv_mean[2] = 1.3;
+    p.njobs = njobs;
+    p.pad_ = 0;
+    return p;
+}
+
+static py::dict jobshop_host(uint64_t ntrials, uint64_t entities, int njobs,
+                             double think_mean, uint64_t seed, int threads) {
+    JobShop::Params p = make_jobshop_params(entities, njobs, think_mean);
+    std::vector<JobShop::Result> res(ntrials);
+    {
+        py::gil_scoped_release nogil;
+        run_host<JobShop>(p, seed, ntrials, threads, res.data());
+    }
+    return jobshop_aggregate(res, -1.0);
+}
+
+static py::dict jobshop_gpu(uint64_t ntrials, uint64_t entities, int njobs,
+                            double think_mean, uint64_t seed, int device) {
+    JobShop::Params p = make_jobshop_params(entities, njobs, think_mean);
+    std::vector<JobShop::Result> res(ntrials);
+    double ms = 0.0;
+    int rc;
+    {
+        py::gil_scoped_release nogil;
+        rc = cimba_jobshop_gpu_run(ntrials, &p, seed, device, &ms, res.data());
+    }
+    if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
+    return jobshop_aggregate(res, ms);
 }
 
 using cmb_models::Scenario;
@@ -241,6 +382,22 @@ PYBIND11_MODULE(_C, m) {
     m.def("mm1_gpu", &mm1_gpu, py::arg("ntrials"), py::arg("num_objects"),
           py::arg("arr_rate") = 0.9, py::arg("srv_rate") = 1.0,
           py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0);
+    m.def("mg1_host", &mg1_host, py::arg("ntrials"), py::arg("num_objects"),
+          py::arg("arr_rate") = 0.8, py::arg("srv_mean") = 1.0,
+          py::arg("srv_scv") = 1.0, py::arg("dist") = 1,
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("threads") = 0);
+    m.def("mg1_gpu", &mg1_gpu, py::arg("ntrials"), py::arg("num_objects"),
+          py::arg("arr_rate") = 0.8, py::arg("srv_mean") = 1.0,
+          py::arg("srv_scv") = 1.0, py::arg("dist") = 1,
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0);
+    m.def("jobshop_host", &jobshop_host, py::arg("ntrials"),
+          py::arg("entities") = 10000, py::arg("njobs") = 24,
+          py::arg("think_mean") = 0.5, py::arg("seed") = 0x34f05c64d7ad598fULL,
+          py::arg("threads") = 0);
+    m.def("jobshop_gpu", &jobshop_gpu, py::arg("ntrials"),
+          py::arg("entities") = 10000, py::arg("njobs") = 24,
+          py::arg("think_mean") = 0.5, py::arg("seed") = 0x34f05c64d7ad598fULL,
+          py::arg("device") = 0);
     m.def("scenario_host", &scenario_host, py::arg("which"));
     m.def("scenario_gpu", &scenario_gpu, py::arg("which"));
     m.def("gpu_device_count", &gpu_device_count);

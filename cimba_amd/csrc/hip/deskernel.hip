@@ -14,38 +14,79 @@
 #include <hip/hip_runtime.h>
 
 #include "../models/mm1.hpp"
+#include "../models/mg1.hpp"
+#include "../models/jobshop.hpp"
 #include "../models/scenarios.hpp"
 #include "../include/cimba/runner.hpp"
 
 namespace {
 
 using cmb::Engine;
+using cmb_models::JobShop;
+using cmb_models::MG1;
 using cmb_models::MM1;
 using cmb_models::Scenario;
 
-constexpr int WPB = 4;  // waves (= trials) per workgroup
-
-using EngMM1 = Engine<MM1>;
-
-// LDS budget check: WPB engines must fit the 160 KiB LDS of a gfx950 CU
-// with room for multiple resident workgroups.
-static_assert(sizeof(EngMM1) * WPB < 60 * 1024, "MM1 engine too big for LDS plan");
-
-__global__ __launch_bounds__(WPB * 64) __attribute__((flatten)) void mm1_kernel(
-    MM1::Params P, uint64_t master_seed, uint32_t ntrials, double until,
-    uint64_t max_events, MM1::Result* __restrict__ out) {
-    __shared__ EngMM1 eng[WPB];
+// generic trial-per-wavefront kernel; WPB = waves (= trials) per workgroup
+template <class Model, int WPB>
+__global__ __launch_bounds__(WPB * 64) __attribute__((flatten)) void trial_kernel(
+    typename Model::Params P, uint64_t master_seed, uint32_t ntrials,
+    double until, uint64_t max_events, typename Model::Result* __restrict__ out) {
+    __shared__ Engine<Model> eng[WPB];
     const int w = (int)(threadIdx.x >> 6);
     if ((threadIdx.x & 63) != 0) return;  // lane 0 of each wave drives
-    EngMM1& E = eng[w];
+    Engine<Model>& E = eng[w];
     const uint32_t stride = gridDim.x * WPB;
     for (uint32_t trial = blockIdx.x * WPB + (uint32_t)w; trial < ntrials;
          trial += stride) {
         E.init(&P, cmb::trial_seed(master_seed, trial), trial);
-        MM1::setup(E);
+        Model::setup(E);
         E.run(until, max_events);
-        MM1::finish(E, out[trial]);
+        Model::finish(E, out[trial]);
     }
+}
+
+// LDS budget checks: WPB engines per workgroup, multiple workgroups per CU
+static_assert(sizeof(Engine<MM1>) * 4 < 60 * 1024, "MM1 engine LDS plan");
+static_assert(sizeof(Engine<MG1>) * 4 < 64 * 1024, "MG1 engine LDS plan");
+static_assert(sizeof(Engine<JobShop>) * 4 < 64 * 1024, "JobShop engine LDS plan");
+
+#define HIP_TRY(x)                                    \
+    do {                                              \
+        hipError_t err_ = (x);                        \
+        if (err_ != hipSuccess) return (int)err_;     \
+    } while (0)
+
+// host-side launcher: upload params, launch, copy per-trial results back
+template <class Model, int WPB>
+int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
+                   uint64_t seed, double until, uint64_t max_events,
+                   double* elapsed_ms, typename Model::Result* host_out) {
+    using Result = typename Model::Result;
+    Result* d_out = nullptr;
+    HIP_TRY(hipMalloc(&d_out, sizeof(Result) * ntrials));
+    const uint32_t want_blocks = (uint32_t)((ntrials + WPB - 1) / WPB);
+    const uint32_t grid = want_blocks < 16384u ? want_blocks : 16384u;
+
+    hipEvent_t t0, t1;
+    HIP_TRY(hipEventCreate(&t0));
+    HIP_TRY(hipEventCreate(&t1));
+    HIP_TRY(hipEventRecord(t0));
+    hipLaunchKernelGGL((trial_kernel<Model, WPB>), dim3(grid), dim3(WPB * 64),
+                       0, 0, P, seed, (uint32_t)ntrials, until, max_events,
+                       d_out);
+    HIP_TRY(hipGetLastError());
+    HIP_TRY(hipEventRecord(t1));
+    HIP_TRY(hipEventSynchronize(t1));
+    float ms = 0.f;
+    HIP_TRY(hipEventElapsedTime(&ms, t0, t1));
+    *elapsed_ms = (double)ms;
+    HIP_TRY(hipMemcpy(host_out, d_out, sizeof(Result) * ntrials,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipFree(d_out));
+    HIP_TRY(hipEventDestroy(t0));
+    HIP_TRY(hipEventDestroy(t1));
+    return 0;
 }
 
 // single-trial semantic-parity kernel: runs one Scenario trial on-device;
@@ -59,12 +100,6 @@ __global__ __launch_bounds__(64) __attribute__((flatten)) void scenario_kernel(
     eng.run(1.0e308, 100000);
     Scenario::finish(eng, *out);
 }
-
-#define HIP_TRY(x)                                    \
-    do {                                              \
-        hipError_t err_ = (x);                        \
-        if (err_ != hipSuccess) return (int)err_;     \
-    } while (0)
 
 }  // namespace
 
@@ -80,39 +115,15 @@ struct Mm1GpuOut {
     int32_t pad_;
 };
 
-// Launch ntrials M/M/1 replications on `device`; aggregates results on the
-// host.  Returns 0 or a hipError_t.
 int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
                       uint64_t num_objects, uint64_t seed, int device,
                       double until, uint64_t max_events, Mm1GpuOut* out) {
     HIP_TRY(hipSetDevice(device));
     MM1::Params P{arr_mean, srv_mean, num_objects};
-    MM1::Result* d_out = nullptr;
-    HIP_TRY(hipMalloc(&d_out, sizeof(MM1::Result) * ntrials));
-
-    const uint32_t want_blocks = (uint32_t)((ntrials + WPB - 1) / WPB);
-    const uint32_t grid = want_blocks < 16384u ? want_blocks : 16384u;
-
-    hipEvent_t t0, t1;
-    HIP_TRY(hipEventCreate(&t0));
-    HIP_TRY(hipEventCreate(&t1));
-    HIP_TRY(hipEventRecord(t0));
-    hipLaunchKernelGGL(mm1_kernel, dim3(grid), dim3(WPB * 64), 0, 0, P, seed,
-                       (uint32_t)ntrials, until, max_events, d_out);
-    HIP_TRY(hipGetLastError());
-    HIP_TRY(hipEventRecord(t1));
-    HIP_TRY(hipEventSynchronize(t1));
-    float ms = 0.f;
-    HIP_TRY(hipEventElapsedTime(&ms, t0, t1));
-
     std::vector<MM1::Result> res(ntrials);
-    HIP_TRY(hipMemcpy(res.data(), d_out, sizeof(MM1::Result) * ntrials,
-                      hipMemcpyDeviceToHost));
-    HIP_TRY(hipFree(d_out));
-    HIP_TRY(hipEventDestroy(t0));
-    HIP_TRY(hipEventDestroy(t1));
-
-    out->elapsed_ms = (double)ms;
+    int rc = run_trials_gpu<MM1, 4>(P, ntrials, seed, until, max_events,
+                                    &out->elapsed_ms, res.data());
+    if (rc) return rc;
     out->total_events = 0;
     out->total_objs = 0;
     out->total_wait = 0.0;
@@ -128,6 +139,24 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
             out->first_bad_status = res[i].status;
     }
     return 0;
+}
+
+// MG1: results array provided by caller (per-trial)
+int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
+                      int device, double* elapsed_ms, void* results_out) {
+    HIP_TRY(hipSetDevice(device));
+    return run_trials_gpu<MG1, 4>(*(const MG1::Params*)params, ntrials, seed,
+                                  1.0e308, UINT64_C(0xFFFFFFFFFFFFFFFF),
+                                  elapsed_ms, (MG1::Result*)results_out);
+}
+
+int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
+                          int device, double* elapsed_ms, void* results_out) {
+    HIP_TRY(hipSetDevice(device));
+    return run_trials_gpu<JobShop, 4>(*(const JobShop::Params*)params, ntrials,
+                                      seed, 1.0e308,
+                                      UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
+                                      (JobShop::Result*)results_out);
 }
 
 int cimba_scenario_gpu_run(int which, void* result_out) {

@@ -83,6 +83,7 @@ enum DemandKind : uint8_t {
     DEM_BUF_SP = 6,  // ctx = buf idx | amount<<8: space >= amount
     DEM_PQOBJ = 7,   // ctx = pq idx: object available
     DEM_PQSP = 8,    // ctx = pq idx: space available
+    DEM_POOL_GE = 9, // ctx = pool idx | amount<<8: free >= amount
     DEM_USER = 32,   // >= DEM_USER: Model::demand(E, pidx, kind, ctx)
 };
 
@@ -658,6 +659,9 @@ struct Engine {
                        (int64_t)(ctx >> 8);
             case DEM_PQOBJ: return pqueues[ctx].len > 0;
             case DEM_PQSP: return pqueues[ctx].len < pqueues[ctx].limit;
+            case DEM_POOL_GE:
+                return pools[ctx & 0xFF].capacity - pools[ctx & 0xFF].in_use >=
+                       (int32_t)(ctx >> 8);
             default:
                 return Model::demand(*this, (int)(&p - procs), p.demand_kind, ctx);
         }
@@ -819,6 +823,21 @@ struct Engine {
             pl.in_use += take;
         }
         return take;
+    }
+
+    // all-or-nothing take (deadlock-free alternative to the greedy partial
+    // loop for multi-unit requests; companion of CMB_POOL_ACQUIRE_ALL)
+    CMB_FORCEINLINE bool pool_try_take_all(int pi, ProcT& p, int32_t want) {
+        Pool& pl = pools[pi];
+        const bool may = p.g_granted || guards[pl.gid].empty();
+        p.g_granted = 0;
+        if (!may || pl.capacity - pl.in_use < want) return false;
+        if (pl.recording) {
+            pl.use_stats.add((double)pl.in_use, now - pl.t_last);
+            pl.t_last = now;
+        }
+        pl.in_use += want;
+        return true;
     }
 
     CMB_FORCEINLINE void pool_release(int pi, int32_t amount) {
@@ -1071,6 +1090,14 @@ struct Engine {
     } while (0)
 
 #define CMB_POOL_RELEASE(pi, amount) E.pool_release((pi), (amount))
+
+// all-or-nothing pool acquire: waits until `amount` units are free and
+// takes them atomically (deadlock-free for multi-unit requests, unlike the
+// reference's greedy loop — see docs/PARITY.md)
+#define CMB_POOL_ACQUIRE_ALL(pi, amount)                                   \
+    CMB_GUARDED_(E.pool_try_take_all((pi), *self, (amount)),               \
+                 E.pools[pi].gid, cmb::DEM_POOL_GE,                        \
+                 (uint32_t)(pi) | ((uint32_t)(amount) << 8))
 
 // buffer get/put of `amount` units (reference cmb_buffer_get/put)
 #define CMB_BUF_GET(bi, amount)                                           \

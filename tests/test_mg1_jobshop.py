@@ -1,0 +1,76 @@
+"""M/G/1 (BASELINE config 3) and job-shop (config 4) model tests.
+
+M/G/1 is validated against the Pollaczek-Khinchine formula
+E[T] = lambda*E[S^2]/(2(1-rho)) + E[S] for several service distributions
+(incl. ziggurat-normal service times), exercising the resource + guard
+path under contention.
+"""
+import pytest
+
+import cimba_amd as ca
+
+
+def pk_system_time(lam, srv_mean, srv_scv):
+    es2 = srv_scv * srv_mean ** 2 + srv_mean ** 2
+    rho = lam * srv_mean
+    return lam * es2 / (2.0 * (1.0 - rho)) + srv_mean
+
+
+@pytest.mark.parametrize("dist,scv", [
+    (0, 1.0),    # exponential (M/M/1 point)
+    (1, 0.25),   # gamma, low variability
+    (1, 2.0),    # gamma, high variability
+    (2, 0.5),    # lognormal
+    (3, 0.04),   # ziggurat-normal service, cv=0.2
+])
+def test_mg1_pollaczek_khinchine(dist, scv):
+    lam, m = 0.8, 1.0
+    r = ca.mg1_host(ntrials=8, num_objects=40_000, arr_rate=lam, srv_mean=m,
+                    srv_scv=scv, dist=dist, seed=42, threads=4)
+    assert r["trials_ok"] == 8
+    theory = pk_system_time(lam, m, scv)
+    assert abs(r["avg_system_time"] - theory) / theory < 0.10, (
+        dist, scv, r["avg_system_time"], theory)
+
+
+def test_mg1_deterministic():
+    a = ca.mg1_host(ntrials=4, num_objects=5000, seed=3, threads=1)
+    b = ca.mg1_host(ntrials=4, num_objects=5000, seed=3, threads=4)
+    assert a["per_trial_avg"] == b["per_trial_avg"]
+    assert a["total_events"] == b["total_events"]
+
+
+def test_jobshop_completes():
+    j = ca.jobshop_host(ntrials=4, entities=2000, njobs=12, think_mean=0.5,
+                        seed=9, threads=4)
+    assert j["trials_ok"] == 4
+    assert j["total_completed"] >= 4 * 2000
+    # utilization sanity: mean units busy below station capacities 3/2/4
+    ub = j["mean_units_busy"]
+    assert 0.2 < ub[0] < 3.0 and 0.2 < ub[1] < 2.0 and 0.2 < ub[2] < 4.0
+
+
+def test_jobshop_deterministic():
+    a = ca.jobshop_host(ntrials=2, entities=500, njobs=8, seed=5, threads=1)
+    b = ca.jobshop_host(ntrials=2, entities=500, njobs=8, seed=5, threads=2)
+    assert a["total_events"] == b["total_events"]
+    assert a["mean_makespan"] == b["mean_makespan"]
+
+
+@pytest.mark.gpu
+def test_mg1_gpu_matches_host():
+    r = ca.mg1_gpu(ntrials=64, num_objects=20_000, arr_rate=0.8, srv_mean=1.0,
+                   srv_scv=0.25, dist=1, seed=7, device=0)
+    h = ca.mg1_host(ntrials=64, num_objects=20_000, arr_rate=0.8,
+                    srv_mean=1.0, srv_scv=0.25, dist=1, seed=7, threads=0)
+    assert r["trials_ok"] == 64
+    assert abs(r["avg_system_time"] - h["avg_system_time"]) < 0.3
+
+
+@pytest.mark.gpu
+def test_jobshop_gpu_matches_host():
+    g = ca.jobshop_gpu(ntrials=32, entities=2000, njobs=12, seed=5, device=0)
+    h = ca.jobshop_host(ntrials=32, entities=2000, njobs=12, seed=5, threads=0)
+    assert g["trials_ok"] == 32
+    assert g["total_completed"] == h["total_completed"]
+    assert abs(g["mean_makespan"] - h["mean_makespan"]) / h["mean_makespan"] < 0.02
