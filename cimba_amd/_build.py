@@ -17,6 +17,7 @@ SOURCES = [
     os.path.join(CSRC, "host", "support.cpp"),
     os.path.join(CSRC, "bindings.cpp"),
     os.path.join(CSRC, "hip", "deskernel.hip"),
+    os.path.join(CSRC, "hip", "rng_kernel.hip"),
 ]
 
 

@@ -14,6 +14,7 @@
 #include "../models/jobshop.hpp"
 #include "../models/scenarios.hpp"
 
+#include <map>
 #include <string>
 #include <vector>
 
@@ -42,6 +43,63 @@ int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                       int device, double* elapsed_ms, void* results_out);
 int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                           int device, double* elapsed_ms, void* results_out);
+int cimba_sample_gpu(int dist, double p0, uint64_t n, uint64_t seed,
+                     int device, double* host_out, double* elapsed_ms);
+int cimba_sample_moments_gpu(int dist, double p0, uint64_t n, uint64_t seed,
+                             int device, double* out7, double* elapsed_ms);
+}
+
+static const std::map<std::string, int> GPU_DISTS = {
+    {"u01", 0}, {"std_normal", 1}, {"std_exponential", 2}, {"std_gamma", 3},
+    {"poisson", 4},
+};
+
+static py::dict rng_sample_gpu(const std::string& dist, double p0, uint64_t n,
+                               uint64_t seed, int device) {
+    auto it = GPU_DISTS.find(dist);
+    if (it == GPU_DISTS.end())
+        throw std::invalid_argument("unknown gpu distribution: " + dist);
+    py::array_t<double> out((py::ssize_t)n);
+    double ms = 0.0;
+    int rc;
+    {
+        py::gil_scoped_release nogil;
+        rc = cimba_sample_gpu(it->second, p0, n, seed, device,
+                              out.mutable_data(), &ms);
+    }
+    if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
+    py::dict d;
+    d["samples"] = out;
+    d["elapsed_ms"] = ms;
+    d["gsamples_per_sec"] = ms > 0 ? (double)n / (ms * 1e6) : 0.0;
+    return d;
+}
+
+static py::dict rng_moments_gpu(const std::string& dist, double p0, uint64_t n,
+                                uint64_t seed, int device) {
+    auto it = GPU_DISTS.find(dist);
+    if (it == GPU_DISTS.end())
+        throw std::invalid_argument("unknown gpu distribution: " + dist);
+    double o[7];
+    double ms = 0.0;
+    int rc;
+    {
+        py::gil_scoped_release nogil;
+        rc = cimba_sample_moments_gpu(it->second, p0, n, seed, device, o, &ms);
+    }
+    if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
+    py::dict d;
+    d["n"] = o[0];
+    const double mean = o[1] / o[0];
+    d["mean"] = mean;
+    d["var"] = o[2] / o[0] - mean * mean;
+    d["s3"] = o[3];
+    d["s4"] = o[4];
+    d["min"] = o[5];
+    d["max"] = o[6];
+    d["elapsed_ms"] = ms;
+    d["gsamples_per_sec"] = ms > 0 ? (double)n / (ms * 1e6) : 0.0;
+    return d;
 }
 
 using cmb_models::JobShop;
@@ -405,6 +463,12 @@ PYBIND11_MODULE(_C, m) {
 
     m.def("rng_sample", &rng_sample, py::arg("dist"), py::arg("params"),
           py::arg("n"), py::arg("seed") = 1ULL);
+    m.def("rng_sample_gpu", &rng_sample_gpu, py::arg("dist"),
+          py::arg("p0") = 0.0, py::arg("n") = 1 << 20, py::arg("seed") = 1ULL,
+          py::arg("device") = 0);
+    m.def("rng_moments_gpu", &rng_moments_gpu, py::arg("dist"),
+          py::arg("p0") = 0.0, py::arg("n") = 1 << 26, py::arg("seed") = 1ULL,
+          py::arg("device") = 0);
     m.def("fmix64", &py_fmix64);
     m.def("sfc64_raw", &py_sfc64_raw, py::arg("seed"), py::arg("skip") = 0);
     m.def("engine_sizeof_mm1", []() { return sizeof(Engine<MM1>); });
