@@ -7,6 +7,8 @@
 #include <pybind11/numpy.h>
 #include <pybind11/stl.h>
 
+#include "cimba/dataset.hpp"
+#include "cimba/logger.hpp"
 #include "cimba/runner.hpp"
 #include "cimba/stats.hpp"
 #include "../models/mm1.hpp"
@@ -268,6 +270,28 @@ static py::dict scenario_gpu(int which) {
     return scenario_result_to_dict(r);
 }
 
+// run scenarios through the full executive (hooks + abandon recovery)
+static py::dict scenario_run_host(int which, uint64_t ntrials, int threads) {
+    Scenario::Params p{which};
+    std::vector<Scenario::Result> res(ntrials);
+    int thread_inits = 0, thread_exits = 0, cleanups = 0;
+    RunHooks hooks;
+    hooks.thread_init = [&](int) { ++thread_inits; };
+    hooks.thread_exit = [&](int) { ++thread_exits; };
+    hooks.trial_cleanup = [&](uint64_t) { ++cleanups; };
+    RunReport rep =
+        run_host<Scenario>(p, 77, ntrials, threads, res.data(), {}, &hooks);
+    py::dict d;
+    d["trials"] = rep.trials;
+    d["failed"] = rep.failed;
+    d["abandoned"] = rep.abandoned;
+    d["thread_inits"] = thread_inits;
+    d["thread_exits"] = thread_exits;
+    d["cleanups"] = cleanups;
+    d["first_status"] = ntrials ? res[0].status : 0;
+    return d;
+}
+
 static py::dict mm1_host(uint64_t ntrials, uint64_t num_objects, double arr_rate,
                          double srv_rate, uint64_t seed, int threads) {
     MM1::Params p{1.0 / arr_rate, 1.0 / srv_rate, num_objects};
@@ -458,6 +482,46 @@ PYBIND11_MODULE(_C, m) {
           py::arg("device") = 0);
     m.def("scenario_host", &scenario_host, py::arg("which"));
     m.def("scenario_gpu", &scenario_gpu, py::arg("which"));
+    m.def("scenario_run_host", &scenario_run_host, py::arg("which"),
+          py::arg("ntrials") = 4, py::arg("threads") = 2);
+
+    // logger controls (reference cmb_logger_flags_on/off)
+    m.def("logger_flags_on", &logger_flags_on);
+    m.def("logger_flags_off", &logger_flags_off);
+    m.def("logger_flags", &logger_flags);
+    m.attr("LOG_FATAL") = (uint32_t)LOG_FATAL;
+    m.attr("LOG_ERROR") = (uint32_t)LOG_ERROR;
+    m.attr("LOG_WARNING") = (uint32_t)LOG_WARNING;
+    m.attr("LOG_INFO") = (uint32_t)LOG_INFO;
+
+    py::class_<Dataset>(m, "Dataset")
+        .def(py::init<>())
+        .def("add", &Dataset::add)
+        .def("size", &Dataset::size)
+        .def("merge", &Dataset::merge)
+        .def("sort", &Dataset::sort)
+        .def("median", &Dataset::median)
+        .def("quantile", &Dataset::quantile)
+        .def("fivenum", [](Dataset& d) {
+            double o[5];
+            d.fivenum(o);
+            return py::make_tuple(o[0], o[1], o[2], o[3], o[4]);
+        })
+        .def("summarize", &Dataset::summarize)
+        .def("histogram", &Dataset::histogram)
+        .def("acf", &Dataset::acf)
+        .def("pacf", &Dataset::pacf)
+        .def("values", [](const Dataset& d) {
+            return py::array_t<double>((py::ssize_t)d.size(),
+                                       d.values().data());
+        });
+
+    py::class_<Timeseries>(m, "Timeseries")
+        .def(py::init<>())
+        .def("add", &Timeseries::add)
+        .def("size", &Timeseries::size)
+        .def("summarize", &Timeseries::summarize)
+        .def("median", &Timeseries::median);
     m.def("gpu_device_count", &gpu_device_count);
     m.def("gpu_sync", []() { return cimba_gpu_sync(); });
 

@@ -1,15 +1,21 @@
 // Host experiment executive: the CPU-plumbing counterpart of reference
 // src/cimba.c cimba_run: worker threads claim trials from
 // a shared atomic index, run the model trial function, write results back
-// into the experiment array.  The GPU path (hip/deskernel.hip) replaces
-// worker threads with wavefronts; this host path exists for BASELINE.json
-// config 1 (CPU plumbing) and for CPU-only unit testing of the exact same
-// engine code.
+// into the experiment array.  Per-thread hooks (reference
+// cimba_thread_hooks_set, cimba.c:150-157: e.g. per-thread GPU streams),
+// trial-abandon recovery (reference __builtin_setjmp/longjmp,
+// cimba.c:74-76,289-329 — here a C++ exception over a POD engine) and the
+// failed-trial count (cimba.c:399) are all supported.  The GPU path
+// (hip/deskernel.hip) replaces worker threads with wavefronts; this host
+// path exists for BASELINE.json config 1 (CPU plumbing) and for CPU-only
+// unit testing of the exact same engine code.
 #pragma once
 
 #include "engine.hpp"
+#include "logger.hpp"
 
 #include <atomic>
+#include <functional>
 #include <memory>
 #include <thread>
 #include <vector>
@@ -23,14 +29,28 @@ CMB_FORCEINLINE uint64_t trial_seed(uint64_t master, uint64_t idx) {
 }
 
 struct RunLimits {
-    double until;
-    uint64_t max_events;
+    double until = 1.0e308;
+    uint64_t max_events = UINT64_C(0xFFFFFFFFFFFFFFFF);
+};
+
+// reference cimba_thread_hooks_set / cimba_trial_cleanup_set
+struct RunHooks {
+    std::function<void(int)> thread_init;          // worker index
+    std::function<void(int)> thread_exit;
+    std::function<void(uint64_t)> trial_cleanup;   // abandoned trial index
+};
+
+struct RunReport {
+    uint64_t trials = 0;
+    uint64_t failed = 0;     // aborted (status != 0) or abandoned trials
+    uint64_t abandoned = 0;  // subset of failed: TrialAbandon thrown
 };
 
 template <class Model>
-void run_host(const typename Model::Params& params, uint64_t master_seed,
-              uint64_t ntrials, int nthreads, typename Model::Result* out,
-              RunLimits limits = {1.0e308, UINT64_C(0xFFFFFFFFFFFFFFFF)}) {
+RunReport run_host(const typename Model::Params& params, uint64_t master_seed,
+                   uint64_t ntrials, int nthreads,
+                   typename Model::Result* out, RunLimits limits = {},
+                   const RunHooks* hooks = nullptr) {
     if (nthreads <= 0) {
         nthreads = (int)std::thread::hardware_concurrency();
         if (nthreads <= 0) nthreads = 1;
@@ -38,28 +58,48 @@ void run_host(const typename Model::Params& params, uint64_t master_seed,
     if ((uint64_t)nthreads > ntrials) nthreads = (int)(ntrials ? ntrials : 1);
 
     std::atomic<uint64_t> next{0};  // reference work-claim: cimba.c:280
-    auto worker = [&]() {
+    std::atomic<uint64_t> failed{0};
+    std::atomic<uint64_t> abandoned{0};
+
+    auto worker = [&](int widx) {
+        if (hooks && hooks->thread_init) hooks->thread_init(widx);
         // one engine per worker, reused across trials (reference: per-thread
         // event queue reset between trials, cimba.c:332)
         auto eng = std::make_unique<Engine<Model>>();
         for (;;) {
             const uint64_t t = next.fetch_add(1, std::memory_order_relaxed);
             if (t >= ntrials) break;
-            eng->init(&params, trial_seed(master_seed, t), (uint32_t)t);
-            Model::setup(*eng);
-            eng->run(limits.until, limits.max_events);
+            const uint64_t seed = trial_seed(master_seed, t);
+            logger_ctx().trial = (uint32_t)t;
+            logger_ctx().seed = seed;
+            logger_ctx().sim_time = 0.0;
+            eng->init(&params, seed, (uint32_t)t);
+            try {
+                Model::setup(*eng);
+                eng->run(limits.until, limits.max_events);
+            } catch (const TrialAbandon& ab) {
+                // recovery: the engine is POD — re-init on the next trial is
+                // the whole cleanup (reference memregistry LIFO teardown)
+                eng->fail(ST_USER_ABORT);
+                abandoned.fetch_add(1, std::memory_order_relaxed);
+                if (hooks && hooks->trial_cleanup) hooks->trial_cleanup(t);
+            }
             Model::finish(*eng, out[t]);
+            if (eng->status != ST_OK)
+                failed.fetch_add(1, std::memory_order_relaxed);
         }
+        if (hooks && hooks->thread_exit) hooks->thread_exit(widx);
     };
 
     if (nthreads == 1) {
-        worker();
-        return;
+        worker(0);
+    } else {
+        std::vector<std::thread> threads;
+        threads.reserve(nthreads);
+        for (int i = 0; i < nthreads; ++i) threads.emplace_back(worker, i);
+        for (auto& th : threads) th.join();
     }
-    std::vector<std::thread> threads;
-    threads.reserve(nthreads);
-    for (int i = 0; i < nthreads; ++i) threads.emplace_back(worker);
-    for (auto& th : threads) th.join();
+    return RunReport{ntrials, failed.load(), abandoned.load()};
 }
 
 }  // namespace cmb

@@ -9,6 +9,9 @@
 #pragma once
 
 #include "../include/cimba/engine.hpp"
+#if !defined(__HIP_DEVICE_COMPILE__)
+#include "../include/cimba/logger.hpp"
+#endif
 
 namespace cmb_models {
 
@@ -42,6 +45,7 @@ struct Scenario : cmb::ModelBase {
         W_WAIT_EVENT = 11,
         W_WAIT_EVENT_CANCEL = 12,
         W_STALE_GRANT = 13,
+        W_ABANDON = 14,
     };
 
     struct Params {
@@ -124,6 +128,8 @@ struct Scenario : cmb::ModelBase {
         F_Q_GETTER_TMO,     // timeout-armed queue get (stale-grant scenario)
         F_Q_GETTER,         // plain queue get, trace value
         F_Q_PUTTER,         // hold d, put value a
+        F_ABANDONER,        // hold 1, then abandon the trial (host logger
+                            // error path; device: Engine::fail)
     };
 
     template <class E_>
@@ -309,6 +315,16 @@ struct Scenario : cmb::ModelBase {
             trace(E, me, T_REL);
             CMB_END();
         }
+        case F_ABANDONER: {
+            CMB_BEGIN();
+            CMB_HOLD(1.0);
+#if defined(__HIP_DEVICE_COMPILE__)
+            E.fail(cmb::ST_USER_ABORT);  // device trial-abort flag
+#else
+            cmb::logger_error("scenario abandon at t=%g", E.now);
+#endif
+            CMB_END();
+        }
         }
     }
 
@@ -416,6 +432,10 @@ struct Scenario : cmb::ModelBase {
             sp(E, 0, F_Q_PUTTER, 0, /*value*/ 77, 0, 1.0);
             sp(E, 1, F_Q_GETTER_TMO, 0, /*tmo*/ 1.0, 0, 0.0);
             sp(E, 2, F_Q_GETTER, 0, 0, 0, 0.5);
+            break;
+        case W_ABANDON:
+            sp(E, 0, F_HOLDER, 0, 0, 0, 1.5);   // a bystander process
+            sp(E, 1, F_ABANDONER, 0, 0, 0, 0.0);
             break;
         default:
             E.fail(cmb::ST_USER_ABORT);
