@@ -95,6 +95,21 @@ def build(verbose=True, force=False):
     if verbose:
         print("[cimba_amd build]", " ".join(cmd), flush=True)
     subprocess.run(cmd, check=True)
+
+    # the public cmb_* C library (include/cimba.h)
+    capi = os.path.join(CSRC, "host", "capi.cpp")
+    capi_obj = os.path.join(objdir, "capi.cpp.o")
+    root_inc = os.path.join(os.path.dirname(ROOT), "include")
+    cmd = [hipcc, "-c", capi, "-o", capi_obj] + cflags + [f"-I{root_inc}"]
+    if verbose:
+        print("[cimba_amd build]", " ".join(cmd), flush=True)
+    subprocess.run(cmd, check=True)
+    libcimba = os.path.join(ROOT, "libcimba.so")
+    cmd = [hipcc, "-shared", "-fPIC", "-o", libcimba, capi_obj,
+           os.path.join(objdir, "support.cpp.o")]
+    if verbose:
+        print("[cimba_amd build]", " ".join(cmd), flush=True)
+    subprocess.run(cmd, check=True)
     return so
 
 

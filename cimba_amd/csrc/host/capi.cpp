@@ -1,0 +1,824 @@
+// Implementation of the public cmb_* C API (include/cimba.h) over the
+// MI355X-native engine: a runtime-configured model ("CModel") whose
+// process bodies, event actions and demand predicates are C function
+// pointers.  Host-side counterpart of the reference's whole public
+// surface (reference src/cimba.c, cmb_event.c, cmb_process.c,
+// cmb_resource*.c, cmb_buffer.c, cmb_objectqueue.c, cmb_priorityqueue.c,
+// cmb_condition.c, cmb_random.c, cmb_datasummary.c, cmb_logger.c).
+// engine headers first: cimba.h defines CMB_* macros (incl. CMB_UNLIMITED)
+// that would otherwise collide with the engine's identifiers
+#include "cimba/engine.hpp"
+#include "cimba/logger.hpp"
+#include "cimba/runner.hpp"
+
+#include "../../../include/cimba.h"
+
+#include <atomic>
+#include <cstring>
+#include <ctime>
+#include <thread>
+#include <vector>
+
+using namespace cmb;
+
+namespace {
+
+constexpr int C_MAX_PROC = 128;
+constexpr int C_UEV = 512;  // pending user events
+constexpr int C_NAME = 32;  // reference CMB_PROCESS_NAMEBUF_SZ
+
+struct CModel : ModelBase {
+    struct Cfg {
+        static constexpr int MAX_PROC = C_MAX_PROC;
+        static constexpr int MAX_EV = 1024;
+        static constexpr int TIMERS = 2;
+        static constexpr int NUM_QUEUES = 8;
+        static constexpr int QCAP = 1024;
+        static constexpr int NUM_RES = 8;
+        static constexpr int NUM_POOLS = 8;
+        static constexpr int NUM_BUFS = 8;
+        static constexpr int NUM_PQ = 4;
+        static constexpr int PQCAP = 256;
+        static constexpr int NUM_COND = 8;
+    };
+    struct Params {
+        cmb_sim* sim;
+    };
+    struct Frame {};
+    struct Result {};
+    struct UEv {
+        cmb_event_func* fn;
+        void* subj;
+        void* obj;
+        int32_t next_free;  // -1 = in use
+    };
+    struct Globals {
+        cmb_process_func* fn[C_MAX_PROC];
+        void* ctx[C_MAX_PROC];
+        char name[C_MAX_PROC][C_NAME];
+        cmb_demand_func* dem_fn[C_MAX_PROC];
+        void* dem_ctx[C_MAX_PROC];
+        UEv uev[C_UEV];
+        int32_t uev_free;  // freelist head
+        int32_t nproc, nq, npq, nres, npool, nbuf, ncond;
+    };
+
+    template <class E_>
+    static void step(E_& E, int pidx);
+    template <class E_>
+    static bool demand(E_& E, int pidx, uint8_t kind, uint32_t ctx);
+    template <class E_>
+    static void on_event(E_& E, const EvEntry& ev);
+    template <class E_>
+    static void setup(E_&) {}
+    template <class E_>
+    static void finish(E_&, Result&) {}
+};
+
+using CEngine = Engine<CModel>;
+
+}  // namespace
+
+struct cmb_sim {
+    CEngine* E;
+    CModel::Params params;
+    uint64_t seed;
+};
+
+namespace {
+
+// ---- handle encoding: index+1 as pointer ----
+template <class T>
+T* enc(int idx) {
+    return (T*)(uintptr_t)(idx + 1);
+}
+template <class T>
+int dec(const T* h) {
+    return (int)((uintptr_t)h - 1);
+}
+
+void init_globals(CEngine& E) {
+    CModel::Globals& g = E.globals;
+    std::memset(&g, 0, sizeof(g));
+    for (int i = 0; i < C_UEV; ++i) g.uev[i].next_free = i + 1;
+    g.uev[C_UEV - 1].next_free = -1;
+    g.uev_free = 0;
+}
+
+int uev_alloc(CEngine& E, cmb_event_func* fn, void* subj, void* obj) {
+    CModel::Globals& g = E.globals;
+    const int s = g.uev_free;
+    if (s < 0) {
+        E.fail(ST_HEAP_FULL);
+        return -1;
+    }
+    g.uev_free = g.uev[s].next_free;
+    g.uev[s] = {fn, subj, obj, -1};
+    return s;
+}
+
+void uev_free_slot(CEngine& E, int s) {
+    CModel::Globals& g = E.globals;
+    g.uev[s].next_free = g.uev_free;
+    g.uev_free = s;
+}
+
+template <class E_>
+void CModel::step(E_& E, int pidx) {
+    cmb_sim* sim = E.params->sim;
+    Globals& g = E.globals;
+    if (g.fn[pidx]) g.fn[pidx](sim, enc<cmb_process>(pidx), g.ctx[pidx]);
+}
+
+template <class E_>
+bool CModel::demand(E_& E, int pidx, uint8_t kind, uint32_t /*ctx*/) {
+    if (kind != DEM_USER) return false;
+    Globals& g = E.globals;
+    if (!g.dem_fn[pidx]) return false;
+    return g.dem_fn[pidx](E.params->sim, g.dem_ctx[pidx]);
+}
+
+template <class E_>
+void CModel::on_event(E_& E, const EvEntry& ev) {
+    if (ev.kind != EV_USER) return;
+    const int s = (int)ev.b;
+    UEv u = E.globals.uev[s];
+    uev_free_slot(E, s);
+    if (u.fn) u.fn(E.params->sim, u.subj, u.obj);
+}
+
+// ---- executive globals (reference cimba.c statics) ----
+std::atomic<uint64_t> g_next{0};
+std::atomic<uint64_t> g_total{0};
+int g_default_threads = 0;
+void (*g_thread_init)(int) = nullptr;
+void (*g_thread_exit)(int) = nullptr;
+void (*g_trial_cleanup)(uint64_t) = nullptr;
+
+}  // namespace
+
+extern "C" {
+
+/* ---- executive ---- */
+
+uint64_t cimba_run(void* experiment, uint64_t n, size_t size,
+                   cimba_trial_func* trial_fn, uint64_t master_seed,
+                   int nthreads) {
+    if (nthreads <= 0) nthreads = g_default_threads;
+    if (nthreads <= 0) nthreads = (int)std::thread::hardware_concurrency();
+    if (nthreads <= 0) nthreads = 1;
+    if ((uint64_t)nthreads > n) nthreads = (int)(n ? n : 1);
+
+    g_next.store(0);
+    g_total.store(n);
+    std::atomic<uint64_t> failed{0};
+
+    auto worker = [&](int widx) {
+        if (g_thread_init) g_thread_init(widx);
+        auto eng = std::make_unique<CEngine>();
+        cmb_sim sim;
+        sim.E = eng.get();
+        sim.params.sim = &sim;
+        for (;;) {
+            const uint64_t t = g_next.fetch_add(1);
+            if (t >= n) break;
+            const uint64_t seed = trial_seed(master_seed, t);
+            sim.seed = seed;
+            logger_ctx().trial = (uint32_t)t;
+            logger_ctx().seed = seed;
+            logger_ctx().sim_time = 0.0;
+            eng->init(&sim.params, seed, (uint32_t)t);
+            init_globals(*eng);
+            try {
+                trial_fn(&sim, (char*)experiment + t * size);
+            } catch (const TrialAbandon&) {
+                eng->fail(ST_USER_ABORT);
+                if (g_trial_cleanup) g_trial_cleanup(t);
+            }
+            if (eng->status != ST_OK) failed.fetch_add(1);
+        }
+        if (g_thread_exit) g_thread_exit(widx);
+    };
+
+    if (nthreads == 1) {
+        worker(0);
+    } else {
+        std::vector<std::thread> th;
+        for (int i = 0; i < nthreads; ++i) th.emplace_back(worker, i);
+        for (auto& t : th) t.join();
+    }
+    return failed.load();
+}
+
+void cimba_threads_use(int nthreads) { g_default_threads = nthreads; }
+uint64_t cimba_trials_remaining(void) {
+    const uint64_t next = g_next.load(), total = g_total.load();
+    return next >= total ? 0 : total - next;
+}
+void cimba_trial_abandon(cmb_sim*) { throw TrialAbandon{1}; }
+void cimba_thread_hooks_set(void (*init_fn)(int), void (*exit_fn)(int)) {
+    g_thread_init = init_fn;
+    g_thread_exit = exit_fn;
+}
+void cimba_trial_cleanup_set(void (*cleanup_fn)(uint64_t)) {
+    g_trial_cleanup = cleanup_fn;
+}
+uint32_t cmb_sim_trial_index(const cmb_sim* s) { return s->E->trial_index; }
+uint64_t cmb_sim_trial_seed(const cmb_sim* s) { return s->seed; }
+uint64_t cmb_sim_events_dispatched(const cmb_sim* s) {
+    return s->E->ev_dispatched;
+}
+
+/* ---- clock & events ---- */
+
+double cmb_time(const cmb_sim* s) { return s->E->now; }
+
+uint64_t cmb_event_schedule(cmb_sim* s, cmb_event_func* action, void* subject,
+                            void* object, double time, int priority) {
+    const int slot = uev_alloc(*s->E, action, subject, object);
+    if (slot < 0) return 0;
+    return s->E->schedule(EV_USER, 0, 0, (uint64_t)slot, time, priority);
+}
+
+bool cmb_event_cancel(cmb_sim* s, uint64_t handle) {
+    EvEntry out;
+    if (!s->E->evq.cancel((uint32_t)handle, &out)) return false;
+    if (out.kind == EV_USER) uev_free_slot(*s->E, (int)out.b);
+    if (s->E->n_event_waiters)
+        s->E->wake_event_waiters((uint32_t)handle, SIG_CANCELLED);
+    return true;
+}
+
+bool cmb_event_reschedule(cmb_sim* s, uint64_t handle, double time,
+                          int priority) {
+    return s->E->event_reschedule((uint32_t)handle, time, priority);
+}
+
+uint64_t cmb_event_pattern_count(cmb_sim* s, cmb_event_func* action,
+                                 void* subject, void* object) {
+    uint64_t cnt = 0;
+    auto& q = s->E->evq;
+    for (int32_t i = 0; i < q.n; ++i) {
+        if (q.e[i].kind != EV_USER) continue;
+        const auto& u = s->E->globals.uev[(int)q.e[i].b];
+        if ((!action || u.fn == action) && (!subject || u.subj == subject) &&
+            (!object || u.obj == object))
+            ++cnt;
+    }
+    return cnt;
+}
+
+uint64_t cmb_event_pattern_cancel(cmb_sim* s, cmb_event_func* action,
+                                  void* subject, void* object) {
+    uint64_t cnt = 0;
+    auto& q = s->E->evq;
+    int32_t i = 0;
+    while (i < q.n) {
+        bool match = false;
+        if (q.e[i].kind == EV_USER) {
+            const auto& u = s->E->globals.uev[(int)q.e[i].b];
+            match = (!action || u.fn == action) &&
+                    (!subject || u.subj == subject) &&
+                    (!object || u.obj == object);
+        }
+        if (match) {
+            uev_free_slot(*s->E, (int)q.e[i].b);
+            q.remove_at(i);
+            ++cnt;
+        } else {
+            ++i;
+        }
+    }
+    return cnt;
+}
+
+void cmb_event_queue_execute(cmb_sim* s) {
+    s->E->run(1.0e308, UINT64_C(0xFFFFFFFFFFFFFFFF));
+}
+void cmb_event_queue_execute_until(cmb_sim* s, double until) {
+    s->E->run(until, UINT64_C(0xFFFFFFFFFFFFFFFF));
+}
+
+/* ---- processes ---- */
+
+cmb_process* cmb_process_spawn(cmb_sim* s, const char* name,
+                               cmb_process_func* fn, void* ctx,
+                               int priority) {
+    CEngine& E = *s->E;
+    const int idx = E.proc_alloc();
+    if (idx < 0) {
+        E.fail(ST_BAD_STATE);
+        return nullptr;
+    }
+    E.proc_init(idx, 0, priority);
+    E.globals.fn[idx] = fn;
+    E.globals.ctx[idx] = ctx;
+    std::strncpy(E.globals.name[idx], name ? name : "", C_NAME - 1);
+    E.globals.name[idx][C_NAME - 1] = 0;
+    return enc<cmb_process>(idx);
+}
+
+void cmb_process_start(cmb_sim* s, cmb_process* p) {
+    s->E->proc_start(dec(p));
+}
+void cmb_process_start_at(cmb_sim* s, cmb_process* p, double delay) {
+    s->E->proc_start(dec(p), delay);
+}
+void cmb_process_interrupt(cmb_sim* s, cmb_process* p, int64_t sig) {
+    s->E->proc_interrupt(dec(p), sig);
+}
+void cmb_process_stop(cmb_sim* s, cmb_process* p) { s->E->proc_stop(dec(p)); }
+void cmb_process_resume(cmb_sim* s, cmb_process* p) {
+    s->E->proc_interrupt(dec(p), SIG_SUCCESS);
+}
+void cmb_process_priority_set(cmb_sim* s, cmb_process* p, int priority) {
+    s->E->proc_priority_set(dec(p), priority);
+}
+int64_t cmb_process_priority(const cmb_sim* s, const cmb_process* p) {
+    return s->E->procs[dec(p)].priority;
+}
+const char* cmb_process_name(const cmb_sim* s, const cmb_process* p) {
+    return s->E->globals.name[dec(p)];
+}
+int cmb_process_state(const cmb_sim* s, const cmb_process* p) {
+    return (int)s->E->procs[dec(p)].state;
+}
+int64_t cmb_process_signal(const cmb_sim* s, const cmb_process* p) {
+    return s->E->procs[dec(p)].sig;
+}
+void* cmb_process_context(const cmb_sim* s, const cmb_process* p) {
+    return s->E->globals.ctx[dec(p)];
+}
+
+int cmb_proc_pc_(const cmb_sim* s, const cmb_process* p) {
+    return s->E->procs[dec(p)].pc;
+}
+void cmb_proc_set_pc_(cmb_sim* s, cmb_process* p, int pc) {
+    s->E->procs[dec(p)].pc = (int16_t)pc;
+}
+void cmb_proc_resumed_(cmb_sim* s, cmb_process* p) {
+    s->E->await_cleanup(s->E->procs[dec(p)]);
+}
+void cmb_hold_setup_(cmb_sim* s, cmb_process* p, double duration) {
+    s->E->hold_setup(s->E->procs[dec(p)], duration);
+}
+int cmb_wait_process_setup_(cmb_sim* s, cmb_process* p, cmb_process* tgt) {
+    return s->E->wait_proc_setup(s->E->procs[dec(p)], dec(tgt)) ? 1 : 0;
+}
+void cmb_wait_event_setup_(cmb_sim* s, cmb_process* p, uint64_t handle) {
+    s->E->wait_event_setup(s->E->procs[dec(p)], (uint32_t)handle);
+}
+void cmb_proc_finish_(cmb_sim* s, cmb_process* p) {
+    s->E->proc_finish(s->E->procs[dec(p)]);
+}
+void cmb_timer_arm_(cmb_sim* s, cmb_process* p, double delay, int64_t sig) {
+    s->E->timer_add(s->E->procs[dec(p)], 0, delay, sig);
+}
+void cmb_timer_disarm_(cmb_sim* s, cmb_process* p) {
+    s->E->timer_cancel(s->E->procs[dec(p)], 0);
+}
+int cmb_sim_ok_(const cmb_sim* s) { return s->E->status == ST_OK; }
+
+/* ---- toolkit ---- */
+
+cmb_objectqueue* cmb_objectqueue_create(cmb_sim* s) {
+    CEngine& E = *s->E;
+    if (E.globals.nq >= CModel::Cfg::NUM_QUEUES) {
+        E.fail(ST_BAD_STATE);
+        return nullptr;
+    }
+    return enc<cmb_objectqueue>(E.globals.nq++);
+}
+void cmb_objectqueue_initialize(cmb_sim* s, cmb_objectqueue* q,
+                                const char* /*name*/, int32_t capacity) {
+    s->E->queues[dec(q)].limit =
+        capacity == CMB_UNLIMITED ? CMB_UNLIMITED : capacity;
+}
+uint64_t cmb_objectqueue_length(const cmb_sim* s, const cmb_objectqueue* q) {
+    return (uint64_t)s->E->queues[dec(q)].len;
+}
+void cmb_objectqueue_recording_start(cmb_sim* s, cmb_objectqueue* q) {
+    auto& Q = s->E->queues[dec(q)];
+    Q.recording = 1;
+    Q.len_stats.reset();
+    Q.t_last = s->E->now;
+}
+void cmb_objectqueue_recording_stop(cmb_sim* s, cmb_objectqueue* q) {
+    s->E->queues[dec(q)].recording = 0;
+}
+void cmb_objectqueue_stats(cmb_sim* s, const cmb_objectqueue* q,
+                           double out4[4]) {
+    auto Q = s->E->queues[dec(q)];  // copy
+    Q.len_stats.add((double)Q.len, s->E->now - Q.t_last);
+    out4[0] = Q.len_stats.mean;
+    out4[1] = Q.len_stats.stddev();
+    out4[2] = Q.len_stats.mn;
+    out4[3] = Q.len_stats.mx;
+}
+
+bool cmb_queue_try_put_(cmb_sim* s, cmb_objectqueue* q, cmb_process* p,
+                        void* object) {
+    return s->E->q_try_put(dec(q), s->E->procs[dec(p)], (uint64_t)object);
+}
+bool cmb_queue_try_get_(cmb_sim* s, cmb_objectqueue* q, cmb_process* p,
+                        void** object) {
+    uint64_t v = 0;
+    if (!s->E->q_try_get(dec(q), s->E->procs[dec(p)], &v)) return false;
+    *object = (void*)v;
+    return true;
+}
+void cmb_queue_wait_space_(cmb_sim* s, cmb_objectqueue* q, cmb_process* p) {
+    CEngine& E = *s->E;
+    E.guard_wait(E.procs[dec(p)], E.queues[dec(q)].g_rear, DEM_QSPACE,
+                 (uint32_t)dec(q));
+}
+void cmb_queue_wait_object_(cmb_sim* s, cmb_objectqueue* q, cmb_process* p) {
+    CEngine& E = *s->E;
+    E.guard_wait(E.procs[dec(p)], E.queues[dec(q)].g_front, DEM_QOBJ,
+                 (uint32_t)dec(q));
+}
+
+cmb_priorityqueue* cmb_priorityqueue_create(cmb_sim* s) {
+    CEngine& E = *s->E;
+    if (E.globals.npq >= CModel::Cfg::NUM_PQ) {
+        E.fail(ST_BAD_STATE);
+        return nullptr;
+    }
+    return enc<cmb_priorityqueue>(E.globals.npq++);
+}
+void cmb_priorityqueue_initialize(cmb_sim* s, cmb_priorityqueue* q,
+                                  const char* /*name*/, int32_t capacity) {
+    s->E->pqueues[dec(q)].limit =
+        capacity == CMB_UNLIMITED ? CModel::Cfg::PQCAP : capacity;
+}
+uint64_t cmb_priorityqueue_length(const cmb_sim* s,
+                                  const cmb_priorityqueue* q) {
+    return (uint64_t)s->E->pqueues[dec(q)].len;
+}
+bool cmb_pqueue_try_put_(cmb_sim* s, cmb_priorityqueue* q, cmb_process* p,
+                         void* object, int priority) {
+    return s->E->pq_try_put(dec(q), s->E->procs[dec(p)], (uint64_t)object,
+                            priority);
+}
+bool cmb_pqueue_try_get_(cmb_sim* s, cmb_priorityqueue* q, cmb_process* p,
+                         void** object) {
+    uint64_t v = 0;
+    if (!s->E->pq_try_get(dec(q), s->E->procs[dec(p)], &v)) return false;
+    *object = (void*)v;
+    return true;
+}
+void cmb_pqueue_wait_space_(cmb_sim* s, cmb_priorityqueue* q,
+                            cmb_process* p) {
+    CEngine& E = *s->E;
+    E.guard_wait(E.procs[dec(p)], E.pqueues[dec(q)].g_rear, DEM_PQSP,
+                 (uint32_t)dec(q));
+}
+void cmb_pqueue_wait_object_(cmb_sim* s, cmb_priorityqueue* q,
+                             cmb_process* p) {
+    CEngine& E = *s->E;
+    E.guard_wait(E.procs[dec(p)], E.pqueues[dec(q)].g_front, DEM_PQOBJ,
+                 (uint32_t)dec(q));
+}
+
+cmb_resource* cmb_resource_create(cmb_sim* s) {
+    CEngine& E = *s->E;
+    if (E.globals.nres >= CModel::Cfg::NUM_RES) {
+        E.fail(ST_BAD_STATE);
+        return nullptr;
+    }
+    return enc<cmb_resource>(E.globals.nres++);
+}
+void cmb_resource_initialize(cmb_sim*, cmb_resource*, const char*) {}
+void cmb_resource_release(cmb_sim* s, cmb_resource* r, cmb_process*) {
+    s->E->resource_release(dec(r));
+}
+bool cmb_resource_in_use(const cmb_sim* s, const cmb_resource* r) {
+    return s->E->resources[dec(r)].holder >= 0;
+}
+cmb_process* cmb_resource_holder(const cmb_sim* s, const cmb_resource* r) {
+    const int h = s->E->resources[dec(r)].holder;
+    return h < 0 ? nullptr : enc<cmb_process>(h);
+}
+void cmb_resource_recording_start(cmb_sim* s, cmb_resource* r) {
+    auto& R = s->E->resources[dec(r)];
+    R.recording = 1;
+    R.busy.reset();
+    R.t_last = s->E->now;
+}
+void cmb_resource_stats(cmb_sim* s, const cmb_resource* r, double out4[4]) {
+    auto R = s->E->resources[dec(r)];  // copy
+    R.busy.add(R.holder >= 0 ? 1.0 : 0.0, s->E->now - R.t_last);
+    out4[0] = R.busy.mean;
+    out4[1] = R.busy.stddev();
+    out4[2] = R.busy.mn;
+    out4[3] = R.busy.mx;
+}
+bool cmb_resource_try_acquire_(cmb_sim* s, cmb_resource* r, cmb_process* p) {
+    return s->E->res_try_acquire(dec(r), s->E->procs[dec(p)]);
+}
+bool cmb_resource_try_preempt_(cmb_sim* s, cmb_resource* r, cmb_process* p) {
+    return s->E->res_try_preempt(dec(r), s->E->procs[dec(p)]);
+}
+void cmb_resource_wait_(cmb_sim* s, cmb_resource* r, cmb_process* p) {
+    CEngine& E = *s->E;
+    E.guard_wait(E.procs[dec(p)], E.resources[dec(r)].gid, DEM_RES,
+                 (uint32_t)dec(r));
+}
+
+cmb_resourcepool* cmb_resourcepool_create(cmb_sim* s) {
+    CEngine& E = *s->E;
+    if (E.globals.npool >= CModel::Cfg::NUM_POOLS) {
+        E.fail(ST_BAD_STATE);
+        return nullptr;
+    }
+    return enc<cmb_resourcepool>(E.globals.npool++);
+}
+void cmb_resourcepool_initialize(cmb_sim* s, cmb_resourcepool* r,
+                                 const char* /*name*/, int32_t capacity) {
+    s->E->pools[dec(r)].capacity = capacity;
+}
+void cmb_resourcepool_release(cmb_sim* s, cmb_resourcepool* r,
+                              int32_t amount) {
+    s->E->pool_release(dec(r), amount);
+}
+int32_t cmb_resourcepool_capacity(const cmb_sim* s,
+                                  const cmb_resourcepool* r) {
+    return s->E->pools[dec(r)].capacity;
+}
+int32_t cmb_resourcepool_in_use(const cmb_sim* s, const cmb_resourcepool* r) {
+    return s->E->pools[dec(r)].in_use;
+}
+int32_t cmb_resourcepool_available(const cmb_sim* s,
+                                   const cmb_resourcepool* r) {
+    return s->E->pools[dec(r)].capacity - s->E->pools[dec(r)].in_use;
+}
+int32_t cmb_pool_try_take_(cmb_sim* s, cmb_resourcepool* r, cmb_process* p,
+                           int32_t want) {
+    return s->E->pool_try_take(dec(r), s->E->procs[dec(p)], want);
+}
+bool cmb_pool_try_take_all_(cmb_sim* s, cmb_resourcepool* r, cmb_process* p,
+                            int32_t want) {
+    return s->E->pool_try_take_all(dec(r), s->E->procs[dec(p)], want);
+}
+void cmb_pool_wait_(cmb_sim* s, cmb_resourcepool* r, cmb_process* p) {
+    CEngine& E = *s->E;
+    E.guard_wait(E.procs[dec(p)], E.pools[dec(r)].gid, DEM_POOL,
+                 (uint32_t)dec(r));
+}
+void cmb_pool_wait_ge_(cmb_sim* s, cmb_resourcepool* r, cmb_process* p,
+                       int32_t amount) {
+    CEngine& E = *s->E;
+    E.guard_wait(E.procs[dec(p)], E.pools[dec(r)].gid, DEM_POOL_GE,
+                 (uint32_t)dec(r) | ((uint32_t)amount << 8));
+}
+
+cmb_buffer* cmb_buffer_create(cmb_sim* s) {
+    CEngine& E = *s->E;
+    if (E.globals.nbuf >= CModel::Cfg::NUM_BUFS) {
+        E.fail(ST_BAD_STATE);
+        return nullptr;
+    }
+    return enc<cmb_buffer>(E.globals.nbuf++);
+}
+void cmb_buffer_initialize(cmb_sim* s, cmb_buffer* b, const char* /*name*/,
+                           int64_t capacity, int64_t initial_level) {
+    auto& B = s->E->buffers[dec(b)];
+    B.capacity = capacity == CMB_UNLIMITED ? INT64_MAX / 2 : capacity;
+    B.level = initial_level;
+}
+int64_t cmb_buffer_level(const cmb_sim* s, const cmb_buffer* b) {
+    return s->E->buffers[dec(b)].level;
+}
+int64_t cmb_buffer_capacity(const cmb_sim* s, const cmb_buffer* b) {
+    return s->E->buffers[dec(b)].capacity;
+}
+bool cmb_buffer_try_get_(cmb_sim* s, cmb_buffer* b, cmb_process* p,
+                         int64_t amount) {
+    return s->E->buf_try_get(dec(b), s->E->procs[dec(p)], amount);
+}
+bool cmb_buffer_try_put_(cmb_sim* s, cmb_buffer* b, cmb_process* p,
+                         int64_t amount) {
+    return s->E->buf_try_put(dec(b), s->E->procs[dec(p)], amount);
+}
+void cmb_buffer_wait_level_(cmb_sim* s, cmb_buffer* b, cmb_process* p,
+                            int64_t amount) {
+    CEngine& E = *s->E;
+    E.guard_wait(E.procs[dec(p)], E.buffers[dec(b)].g_get, DEM_BUF_GE,
+                 (uint32_t)dec(b) | ((uint32_t)amount << 8));
+}
+void cmb_buffer_wait_space_(cmb_sim* s, cmb_buffer* b, cmb_process* p,
+                            int64_t amount) {
+    CEngine& E = *s->E;
+    E.guard_wait(E.procs[dec(p)], E.buffers[dec(b)].g_put, DEM_BUF_SP,
+                 (uint32_t)dec(b) | ((uint32_t)amount << 8));
+}
+
+cmb_condition* cmb_condition_create(cmb_sim* s) {
+    CEngine& E = *s->E;
+    if (E.globals.ncond >= CModel::Cfg::NUM_COND) {
+        E.fail(ST_BAD_STATE);
+        return nullptr;
+    }
+    return enc<cmb_condition>(E.globals.ncond++);
+}
+void cmb_condition_initialize(cmb_sim*, cmb_condition*, const char*) {}
+uint64_t cmb_condition_signal(cmb_sim* s, cmb_condition* c) {
+    return s->E->condition_signal(dec(c));
+}
+void cmb_condition_wait_setup_(cmb_sim* s, cmb_condition* c, cmb_process* p,
+                               cmb_demand_func* demand, void* ctx) {
+    CEngine& E = *s->E;
+    const int pidx = dec(p);
+    E.globals.dem_fn[pidx] = demand;
+    E.globals.dem_ctx[pidx] = ctx;
+    E.guard_wait(E.procs[pidx], E.conds[dec(c)].gid, DEM_USER, 0);
+}
+
+/* ---- RNG ---- */
+
+uint64_t cmb_random_sfc64(cmb_sim* s) { return s->E->rng.next(); }
+uint64_t cmb_random_fmix64(uint64_t x) { return fmix64(x); }
+uint64_t cmb_random_curseed(const cmb_sim* s) { return s->seed; }
+double cmb_random_uniform(cmb_sim* s, double lo, double hi) {
+    return s->E->rng.uniform(lo, hi);
+}
+bool cmb_random_flip(cmb_sim* s, double p) { return s->E->rng.flip(p); }
+int64_t cmb_random_bernoulli(cmb_sim* s, double p) {
+    return s->E->rng.bernoulli(p);
+}
+double cmb_random_std_normal(cmb_sim* s) { return s->E->rng.std_normal(); }
+double cmb_random_normal(cmb_sim* s, double mu, double sg) {
+    return s->E->rng.normal(mu, sg);
+}
+double cmb_random_std_exponential(cmb_sim* s) {
+    return s->E->rng.std_exponential();
+}
+double cmb_random_exponential(cmb_sim* s, double mean) {
+    return s->E->rng.exponential(mean);
+}
+double cmb_random_lognormal(cmb_sim* s, double mu, double sg) {
+    return s->E->rng.lognormal(mu, sg);
+}
+double cmb_random_logistic(cmb_sim* s, double a, double b) {
+    return s->E->rng.logistic(a, b);
+}
+double cmb_random_cauchy(cmb_sim* s, double a, double b) {
+    return s->E->rng.cauchy(a, b);
+}
+double cmb_random_rayleigh(cmb_sim* s, double sg) {
+    return s->E->rng.rayleigh(sg);
+}
+double cmb_random_weibull(cmb_sim* s, double k, double l) {
+    return s->E->rng.weibull(k, l);
+}
+double cmb_random_pareto(cmb_sim* s, double a, double b) {
+    return s->E->rng.pareto(a, b);
+}
+double cmb_random_triangular(cmb_sim* s, double lo, double mo, double hi) {
+    return s->E->rng.triangular(lo, mo, hi);
+}
+double cmb_random_pert(cmb_sim* s, double lo, double mo, double hi) {
+    return s->E->rng.pert(lo, mo, hi);
+}
+double cmb_random_std_gamma(cmb_sim* s, double a) {
+    return s->E->rng.std_gamma(a);
+}
+double cmb_random_gamma(cmb_sim* s, double k, double t) {
+    return s->E->rng.gamma(k, t);
+}
+double cmb_random_erlang(cmb_sim* s, int64_t k, double m) {
+    return s->E->rng.erlang(k, m);
+}
+double cmb_random_hypoexponential(cmb_sim* s, double a, double b) {
+    return s->E->rng.hypoexponential(a, b);
+}
+double cmb_random_hyperexponential(cmb_sim* s, double p, double a, double b) {
+    return s->E->rng.hyperexponential(p, a, b);
+}
+double cmb_random_std_beta(cmb_sim* s, double a, double b) {
+    return s->E->rng.std_beta(a, b);
+}
+double cmb_random_beta(cmb_sim* s, double a, double b, double lo, double hi) {
+    return s->E->rng.beta(a, b, lo, hi);
+}
+double cmb_random_chisquared(cmb_sim* s, double k) {
+    return s->E->rng.chisquared(k);
+}
+double cmb_random_std_t_dist(cmb_sim* s, double df) {
+    return s->E->rng.std_t_dist(df);
+}
+double cmb_random_t_dist(cmb_sim* s, double df, double a, double b) {
+    return s->E->rng.t_dist(df, a, b);
+}
+double cmb_random_f_dist(cmb_sim* s, double a, double b) {
+    return s->E->rng.f_dist(a, b);
+}
+int64_t cmb_random_geometric(cmb_sim* s, double p) {
+    return s->E->rng.geometric(p);
+}
+int64_t cmb_random_poisson(cmb_sim* s, double m) { return s->E->rng.poisson(m); }
+int64_t cmb_random_binomial(cmb_sim* s, int64_t n, double p) {
+    return s->E->rng.binomial(n, p);
+}
+int64_t cmb_random_negative_binomial(cmb_sim* s, double r, double p) {
+    return s->E->rng.negative_binomial(r, p);
+}
+int64_t cmb_random_pascal(cmb_sim* s, int64_t r, double p) {
+    return s->E->rng.pascal(r, p);
+}
+int64_t cmb_random_discrete_uniform(cmb_sim* s, int64_t lo, int64_t hi) {
+    return s->E->rng.discrete_uniform(lo, hi);
+}
+int64_t cmb_random_dice(cmb_sim* s, int64_t n) { return s->E->rng.dice(n); }
+int64_t cmb_random_discrete_nonuniform(cmb_sim* s, const double* w,
+                                       int64_t n) {
+    return s->E->rng.discrete_nonuniform(w, n);
+}
+int64_t cmb_random_loaded_dice(cmb_sim* s, const double* w, int64_t n) {
+    return s->E->rng.loaded_dice(w, n);
+}
+uint64_t cmb_random_hwseed(void) {
+    // host entropy (reference RDSEED asm; /dev/urandom is the portable
+    // equivalent — see docs/PARITY.md)
+    uint64_t x = 0;
+    FILE* f = fopen("/dev/urandom", "rb");
+    if (f) {
+        if (fread(&x, sizeof(x), 1, f) != 1) x = 0;
+        fclose(f);
+    }
+    if (!x) x = (uint64_t)time(nullptr) ^ 0x9E3779B97F4A7C15ULL;
+    return x;
+}
+
+/* ---- summaries ---- */
+
+void cmb_datasummary_initialize(cmb_datasummary* s) {
+    reinterpret_cast<DataSummary*>(s)->reset();
+}
+void cmb_datasummary_add(cmb_datasummary* s, double x) {
+    reinterpret_cast<DataSummary*>(s)->add(x);
+}
+void cmb_datasummary_merge(cmb_datasummary* s, const cmb_datasummary* o) {
+    reinterpret_cast<DataSummary*>(s)->merge(
+        *reinterpret_cast<const DataSummary*>(o));
+}
+double cmb_datasummary_count(const cmb_datasummary* s) { return s->n; }
+double cmb_datasummary_mean(const cmb_datasummary* s) { return s->mean; }
+double cmb_datasummary_variance(const cmb_datasummary* s) {
+    return reinterpret_cast<const DataSummary*>(s)->variance();
+}
+double cmb_datasummary_stddev(const cmb_datasummary* s) {
+    return reinterpret_cast<const DataSummary*>(s)->stddev();
+}
+double cmb_datasummary_skewness(const cmb_datasummary* s) {
+    return reinterpret_cast<const DataSummary*>(s)->skewness();
+}
+double cmb_datasummary_kurtosis(const cmb_datasummary* s) {
+    return reinterpret_cast<const DataSummary*>(s)->kurtosis();
+}
+double cmb_datasummary_minimum(const cmb_datasummary* s) { return s->mn; }
+double cmb_datasummary_maximum(const cmb_datasummary* s) { return s->mx; }
+
+void cmb_wtdsummary_initialize(cmb_wtdsummary* s) {
+    reinterpret_cast<WtdSummary*>(s)->reset();
+}
+void cmb_wtdsummary_add(cmb_wtdsummary* s, double x, double w) {
+    reinterpret_cast<WtdSummary*>(s)->add(x, w);
+}
+void cmb_wtdsummary_merge(cmb_wtdsummary* s, const cmb_wtdsummary* o) {
+    reinterpret_cast<WtdSummary*>(s)->merge(
+        *reinterpret_cast<const WtdSummary*>(o));
+}
+double cmb_wtdsummary_mean(const cmb_wtdsummary* s) { return s->mean; }
+double cmb_wtdsummary_variance(const cmb_wtdsummary* s) {
+    return reinterpret_cast<const WtdSummary*>(s)->variance();
+}
+
+/* ---- logger ---- */
+
+void cmb_logger_flags_on(uint32_t flags) { logger_flags_on(flags); }
+void cmb_logger_flags_off(uint32_t flags) { logger_flags_off(flags); }
+void cmb_logger_info(cmb_sim* s, const char* fmt, ...) {
+    logger_ctx().sim_time = s->E->now;
+    va_list ap;
+    va_start(ap, fmt);
+    logger_vlog(LOG_INFO, "info", fmt, ap);
+    va_end(ap);
+}
+void cmb_logger_warning(cmb_sim* s, const char* fmt, ...) {
+    logger_ctx().sim_time = s->E->now;
+    va_list ap;
+    va_start(ap, fmt);
+    logger_vlog(LOG_WARNING, "warning", fmt, ap);
+    va_end(ap);
+}
+void cmb_logger_error(cmb_sim* s, const char* fmt, ...) {
+    logger_ctx().sim_time = s->E->now;
+    va_list ap;
+    va_start(ap, fmt);
+    logger_vlog(LOG_ERROR, "error", fmt, ap);
+    va_end(ap);
+    throw TrialAbandon{1};
+}
+
+}  // extern "C"

@@ -1,0 +1,57 @@
+"""Compile-and-run test of the public cmb_* C API (include/cimba.h):
+builds tutorial/mm1_capi.c with gcc against libcimba.so and checks the
+M/M/1 result — the counterpart of the reference's install-verification job
+(reference .github/workflows/ci.yml install job + header self-containment
+check)."""
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.fixture(scope="module")
+def libcimba():
+    import cimba_amd  # triggers build (includes libcimba.so)
+
+    lib = os.path.join(ROOT, "cimba_amd", "libcimba.so")
+    if not os.path.exists(lib):
+        from cimba_amd import _build
+
+        _build.build(force=True)
+    assert os.path.exists(lib)
+    return lib
+
+
+def test_capi_mm1_compiles_and_runs(libcimba, tmp_path):
+    exe = str(tmp_path / "mm1_capi")
+    compile_cmd = [
+        "gcc", "-std=c11", "-O2", "-Wall", "-Werror",
+        "-I", os.path.join(ROOT, "include"),
+        os.path.join(ROOT, "tutorial", "mm1_capi.c"),
+        "-L", os.path.join(ROOT, "cimba_amd"), "-lcimba",
+        f"-Wl,-rpath,{os.path.join(ROOT, 'cimba_amd')}",
+        "-lm", "-o", exe,
+    ]
+    r = subprocess.run(compile_cmd, capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    out = subprocess.run([exe], capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, (out.stdout, out.stderr)
+    assert "Average system time" in out.stdout
+    # determinism: same master seed => identical output
+    out2 = subprocess.run([exe], capture_output=True, text=True, timeout=300)
+    assert out2.stdout == out.stdout
+
+
+def test_capi_header_is_c_clean(tmp_path):
+    # header must compile as plain C99 without the library
+    src = tmp_path / "hdr.c"
+    src.write_text('#include "cimba.h"\nint main(void){return 0;}\n')
+    r = subprocess.run(
+        ["gcc", "-std=c99", "-Wall", "-Werror", "-I",
+         os.path.join(ROOT, "include"), "-c", str(src), "-o",
+         str(tmp_path / "hdr.o")],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
