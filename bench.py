@@ -37,7 +37,7 @@ def main():
     ap.add_argument("--seed", type=lambda s: int(s, 0), default=0x34F05C64D7AD598F)
     ap.add_argument("--host", action="store_true",
                     help="debug: run the CPU host engine instead of the GPU")
-    ap.add_argument("--model", choices=["mm1", "mg1", "jobshop"],
+    ap.add_argument("--model", choices=["mm1", "mg1", "jobshop", "awacs"],
                     default="mm1",
                     help="mm1 = headline benchmark; mg1/jobshop = "
                          "BASELINE configs 3-4")
@@ -84,10 +84,14 @@ def main():
             kw = dict(ntrials=args.trials, num_objects=args.objects,
                       arr_rate=0.8, srv_mean=1.0, srv_scv=0.25, dist=3,
                       seed=seed)
-        else:
+        elif args.model == "jobshop":
             fn = ca.jobshop_gpu if use_gpu else ca.jobshop_host
             kw = dict(ntrials=args.trials, entities=args.objects, njobs=24,
                       seed=seed)
+        else:
+            fn = ca._C.awacs_gpu if use_gpu else ca._C.awacs_host
+            kw = dict(ntrials=args.trials, duration=args.objects / 25.0,
+                      ntargets=1000, seed=seed)
         if use_gpu:
             kw["device"] = local_rank
         else:
@@ -138,7 +142,7 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": {"mm1": "MM1_multi", "mg1": "MG1_resource",
-                          "jobshop": "JobShop_pools"}[args.model],
+                          "jobshop": "JobShop_pools", "awacs": "AWACS_radar"}[args.model],
                 "trials_per_gpu_per_step": args.trials,
                 "objects_per_trial": args.objects,
                 "arrival_rate": 0.9,

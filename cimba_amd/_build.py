@@ -18,6 +18,7 @@ SOURCES = [
     os.path.join(CSRC, "bindings.cpp"),
     os.path.join(CSRC, "hip", "deskernel.hip"),
     os.path.join(CSRC, "hip", "rng_kernel.hip"),
+    os.path.join(CSRC, "hip", "awacs_kernel.hip"),
 ]
 
 

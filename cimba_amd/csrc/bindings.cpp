@@ -14,6 +14,7 @@
 #include "../models/mm1.hpp"
 #include "../models/mg1.hpp"
 #include "../models/jobshop.hpp"
+#include "../models/awacs.hpp"
 #include "../models/scenarios.hpp"
 
 #include <map>
@@ -49,6 +50,137 @@ int cimba_sample_gpu(int dist, double p0, uint64_t n, uint64_t seed,
                      int device, double* host_out, double* elapsed_ms);
 int cimba_sample_moments_gpu(int dist, double p0, uint64_t n, uint64_t seed,
                              int device, double* out7, double* elapsed_ms);
+int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
+                        int device, double* elapsed_ms, void* results_out);
+int cimba_awacs_power_test(const void* params, uint64_t seed, int device,
+                           float* out_powers, int* nt_out);
+}
+
+using cmb_models::AWACS;
+
+static AWACS::Params make_awacs_params(double duration, double dwell,
+                                       double maneuver_mean, int ntargets,
+                                       double area, double speed,
+                                       double snr_ref) {
+    AWACS::Params p;
+    p.duration = duration;
+    p.dwell = dwell;
+    p.maneuver_mean = maneuver_mean;
+    p.ntargets = ntargets;
+    p.pad_ = 0;
+    p.area = area;
+    p.speed = speed;
+    p.snr_ref = snr_ref;
+    return p;
+}
+
+static py::dict awacs_aggregate(const std::vector<AWACS::Result>& res,
+                                double elapsed_ms) {
+    uint64_t ev = 0, det = 0, dwl = 0, man = 0, ok = 0;
+    double pw = 0.0;
+    int32_t bad = 0;
+    for (auto& r : res) {
+        ev += r.events;
+        det += r.detections;
+        dwl += r.dwells;
+        man += r.maneuvers;
+        pw += r.sum_power;
+        if (r.status == 0)
+            ++ok;
+        else if (!bad)
+            bad = r.status;
+    }
+    py::dict d;
+    d["total_events"] = ev;
+    d["total_detections"] = det;
+    d["total_dwells"] = dwl;
+    d["total_maneuvers"] = man;
+    d["sum_power"] = pw;
+    d["trials_ok"] = ok;
+    d["first_bad_status"] = bad;
+    if (elapsed_ms >= 0) {
+        d["elapsed_ms"] = elapsed_ms;
+        d["events_per_sec"] =
+            elapsed_ms > 0 ? (double)ev / (elapsed_ms * 1e-3) : 0.0;
+        d["target_dwells_per_sec"] = elapsed_ms > 0
+            ? (double)dwl * 1000.0 / (elapsed_ms * 1e-3)  // nt folded below
+            : 0.0;
+    }
+    return d;
+}
+
+static py::dict awacs_host(uint64_t ntrials, double duration, double dwell,
+                           double maneuver_mean, int ntargets, uint64_t seed,
+                           int threads) {
+    AWACS::Params p = make_awacs_params(duration, dwell, maneuver_mean,
+                                        ntargets, 50000.0, 250.0, 2.0e15);
+    std::vector<AWACS::Result> res(ntrials);
+    {
+        py::gil_scoped_release nogil;
+        run_host<AWACS>(p, seed, ntrials, threads, res.data());
+    }
+    return awacs_aggregate(res, -1.0);
+}
+
+static py::dict awacs_gpu(uint64_t ntrials, double duration, double dwell,
+                          double maneuver_mean, int ntargets, uint64_t seed,
+                          int device) {
+    AWACS::Params p = make_awacs_params(duration, dwell, maneuver_mean,
+                                        ntargets, 50000.0, 250.0, 2.0e15);
+    std::vector<AWACS::Result> res(ntrials);
+    double ms = 0.0;
+    int rc;
+    {
+        py::gil_scoped_release nogil;
+        rc = cimba_awacs_gpu_run(ntrials, &p, seed, device, &ms, res.data());
+    }
+    if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
+    return awacs_aggregate(res, ms);
+}
+
+// numerics: device MFMA beamforming powers vs host f32 scalar and fp64
+// reference for the same (seeded) target set
+static py::dict awacs_power_check(int ntargets, uint64_t seed, int device) {
+    AWACS::Params p = make_awacs_params(10.0, 0.04, 5.0, ntargets, 50000.0,
+                                        250.0, 2.0e15);
+    std::vector<float> dev_pow(AWACS::MAX_T, 0.f);
+    int nt = 0;
+    int rc = cimba_awacs_power_test(&p, seed, device, dev_pow.data(), &nt);
+    if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
+
+    // host f32 path + fp64 reference on the identical state
+    auto eng = std::make_unique<Engine<AWACS>>();
+    eng->init(&p, seed, 0);
+    AWACS::setup(*eng);
+    const AWACS::Globals& g = eng->globals;
+    py::array_t<double> host32((py::ssize_t)nt), host64((py::ssize_t)nt),
+        dev((py::ssize_t)nt);
+    for (int t = 0; t < nt; ++t) {
+        host32.mutable_data()[t] = (double)AWACS::target_power(g, t);
+        // fp64 reference of the same formula
+        const double x = g.x[t], y = g.y[t];
+        const double r2 = x * x + y * y + 1.0;
+        const double s = sin(atan2(y, x));
+        double best = 0.0;
+        for (int b = 0; b < AWACS::BEAMS; ++b) {
+            double re = 0.0, im = 0.0;
+            for (int e = 0; e < AWACS::ELEM; ++e) {
+                const double ph = 3.14159265358979 * e * s;
+                re += cos(ph) * g.wr[e][b] + sin(ph) * g.wi[e][b];
+                im += sin(ph) * g.wr[e][b] - cos(ph) * g.wi[e][b];
+            }
+            const double pw = re * re + im * im;
+            best = pw > best ? pw : best;
+        }
+        host64.mutable_data()[t] = best * g.rcs[t] / (r2 * r2);
+        dev.mutable_data()[t] = (double)dev_pow[t];
+    }
+    py::dict d;
+    d["device_mfma"] = dev;
+    d["host_f32"] = host32;
+    d["host_f64"] = host64;
+    d["nt"] = nt;
+    return d;
 }
 
 static const std::map<std::string, int> GPU_DISTS = {
@@ -480,6 +612,16 @@ PYBIND11_MODULE(_C, m) {
           py::arg("entities") = 10000, py::arg("njobs") = 24,
           py::arg("think_mean") = 0.5, py::arg("seed") = 0x34f05c64d7ad598fULL,
           py::arg("device") = 0);
+    m.def("awacs_host", &awacs_host, py::arg("ntrials"),
+          py::arg("duration") = 60.0, py::arg("dwell") = 0.04,
+          py::arg("maneuver_mean") = 5.0, py::arg("ntargets") = 1000,
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("threads") = 0);
+    m.def("awacs_gpu", &awacs_gpu, py::arg("ntrials"),
+          py::arg("duration") = 60.0, py::arg("dwell") = 0.04,
+          py::arg("maneuver_mean") = 5.0, py::arg("ntargets") = 1000,
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0);
+    m.def("awacs_power_check", &awacs_power_check, py::arg("ntargets") = 1000,
+          py::arg("seed") = 42ULL, py::arg("device") = 0);
     m.def("scenario_host", &scenario_host, py::arg("which"));
     m.def("scenario_gpu", &scenario_gpu, py::arg("which"));
     m.def("scenario_run_host", &scenario_run_host, py::arg("which"),
