@@ -54,6 +54,12 @@ int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                         int device, double* elapsed_ms, void* results_out);
 int cimba_awacs_power_test(const void* params, uint64_t seed, int device,
                            float* out_powers, int* nt_out);
+int cimba_awacs_power_test_devinit(const void* params, uint64_t seed,
+                                   int device, float* out_powers,
+                                   int* nt_out);
+int cimba_awacs_first_dwell_dbg(const void* params, uint64_t master_seed,
+                                int device, float* out_powers);
+int cimba_xlane_repro(int iters, int device, int* out64);
 }
 
 using cmb_models::AWACS;
@@ -146,6 +152,11 @@ static py::dict awacs_power_check(int ntargets, uint64_t seed, int device) {
     std::vector<float> dev_pow(AWACS::MAX_T, 0.f);
     int nt = 0;
     int rc = cimba_awacs_power_test(&p, seed, device, dev_pow.data(), &nt);
+    std::vector<float> dev_pow2(AWACS::MAX_T, 0.f);
+    int nt2 = 0;
+    if (rc == 0)
+        rc = cimba_awacs_power_test_devinit(&p, seed, device, dev_pow2.data(),
+                                            &nt2);
     if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
 
     // host f32 path + fp64 reference on the identical state
@@ -175,8 +186,11 @@ static py::dict awacs_power_check(int ntargets, uint64_t seed, int device) {
         host64.mutable_data()[t] = best * g.rcs[t] / (r2 * r2);
         dev.mutable_data()[t] = (double)dev_pow[t];
     }
+    py::array_t<double> dev2((py::ssize_t)nt);
+    for (int t = 0; t < nt; ++t) dev2.mutable_data()[t] = (double)dev_pow2[t];
     py::dict d;
     d["device_mfma"] = dev;
+    d["device_mfma_devinit"] = dev2;
     d["host_f32"] = host32;
     d["host_f64"] = host64;
     d["nt"] = nt;
@@ -620,8 +634,60 @@ PYBIND11_MODULE(_C, m) {
           py::arg("duration") = 60.0, py::arg("dwell") = 0.04,
           py::arg("maneuver_mean") = 5.0, py::arg("ntargets") = 1000,
           py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0);
+    m.def("xlane_repro", [](int iters, int device) {
+        std::vector<int> o(64);
+        int rc = cimba_xlane_repro(iters, device, o.data());
+        if (rc) throw std::runtime_error("hip " + std::to_string(rc));
+        return o;
+    }, py::arg("iters") = 4, py::arg("device") = 0);
     m.def("awacs_power_check", &awacs_power_check, py::arg("ntargets") = 1000,
           py::arg("seed") = 42ULL, py::arg("device") = 0);
+    m.def("awacs_first_dwell_dbg", [](int ntargets, uint64_t master_seed,
+                                      int device) {
+        AWACS::Params p = make_awacs_params(10.0, 0.04, 5.0, ntargets,
+                                            50000.0, 250.0, 2.0e15);
+        std::vector<float> dev_pow(AWACS::MAX_T, 0.f);
+        int rc = cimba_awacs_first_dwell_dbg(&p, master_seed, device,
+                                             dev_pow.data());
+        if (rc) throw std::runtime_error("hip " + std::to_string(rc));
+        // host reference: trial 0 of the same master seed, first-dwell state
+        auto eng = std::make_unique<Engine<AWACS>>();
+        eng->init(&p, trial_seed(master_seed, 0), 0);
+        AWACS::setup(*eng);
+        py::array_t<double> dev((py::ssize_t)ntargets), h((py::ssize_t)ntargets);
+        for (int t = 0; t < ntargets; ++t) {
+            dev.mutable_data()[t] = (double)dev_pow[t];
+            h.mutable_data()[t] = (double)AWACS::target_power(eng->globals, t);
+        }
+        py::dict d;
+        d["dev"] = dev;
+        d["host"] = h;
+        d["dbg_snr_ref"] = (double)dev_pow[AWACS::MAX_T - 1];
+        d["dbg_trial"] = (double)dev_pow[AWACS::MAX_T - 2];
+        d["dbg_dwells"] = (double)dev_pow[AWACS::MAX_T - 3];
+        d["dbg_draw0"] = (double)dev_pow[AWACS::MAX_T - 4];
+        d["dbg_u01_0"] = (double)dev_pow[AWACS::MAX_T - 5];
+        d["dbg_pow_lane0_pre"] = (double)dev_pow[AWACS::MAX_T - 6];
+        d["dbg_det_lane0_pre"] = (double)dev_pow[AWACS::MAX_T - 7];
+        d["dbg_pow_post"] = (double)dev_pow[AWACS::MAX_T - 8];
+        d["dbg_det_post"] = (double)dev_pow[AWACS::MAX_T - 9];
+        py::list xs, sazs, wrs, accs;
+        for (int l = 0; l < 20; ++l) {
+            xs.append((double)dev_pow[400 + l]);
+            sazs.append((double)dev_pow[500 + l]);
+            wrs.append((double)dev_pow[600 + l]);
+            accs.append((double)dev_pow[700 + l]);
+        }
+        d["x20"] = xs; d["saz20"] = sazs; d["wr20"] = wrs; d["acc20"] = accs;
+
+
+        // host-side comparison for target 0
+        d["host_draw0"] = AWACS::detect_draw(0, 0, 0,
+                                             (float)h.data()[0], 2.0e15);
+        d["host_u01_0"] = AWACS::draw_u01(0, 0, 0);
+        return d;
+    }, py::arg("ntargets") = 64, py::arg("master_seed") = 11ULL,
+       py::arg("device") = 0);
     m.def("scenario_host", &scenario_host, py::arg("which"));
     m.def("scenario_gpu", &scenario_gpu, py::arg("which"));
     m.def("scenario_run_host", &scenario_run_host, py::arg("which"),

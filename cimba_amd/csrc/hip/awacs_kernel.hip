@@ -74,8 +74,9 @@ __device__ void beamform_tile(EngA& E, int base, int lane,
                         g.x[t] * g.x[t] + g.y[t] * g.y[t] + 1.0f;
                     const float power = p * g.rcs[t] / (r2 * r2);
                     if (pow_out) {
-                        pow_out[t] = power;  // numerics-test mode: no draws
-                    } else {
+                        pow_out[t] = power;
+                    }
+                    if (E.params) {
                         if (AWACS::detect_draw(E.trial_index, g.dwells,
                                                (uint32_t)t, power,
                                                E.params->snr_ref)) {
@@ -90,14 +91,23 @@ __device__ void beamform_tile(EngA& E, int base, int lane,
     }
 }
 
-__device__ void dwell_physics_wave(EngA& E, int lane, float* pow_out) {
+// dt/nt/area are wave-uniform: computed on lane 0 and broadcast via
+// readfirstlane so no lane reads the engine's scalar block directly
+__device__ void dwell_physics_wave(EngA& E, int lane, float dt, int nt,
+                                   float area, float* pow_out) {
     AWACS::Globals& g = E.globals;
-    const float dt = (float)(E.now - g.last_t);
-    for (int t = lane; t < g.nt; t += 64) AWACS::advance_target(E, t, dt);
+    for (int t = lane; t < nt; t += 64) {
+        g.x[t] += g.vx[t] * dt;
+        g.y[t] += g.vy[t] * dt;
+        if (g.x[t] > area) g.x[t] -= 2.0f * area;
+        if (g.x[t] < -area) g.x[t] += 2.0f * area;
+        if (g.y[t] > area) g.y[t] -= 2.0f * area;
+        if (g.y[t] < -area) g.y[t] += 2.0f * area;
+    }
     // (no barrier needed: one wave, lockstep)
     unsigned long long det_local = 0;
     double pow_local = 0.0;
-    for (int base = 0; base < g.nt; base += 64)
+    for (int base = 0; base < nt; base += 64)
         beamform_tile(E, base, lane, &det_local, &pow_local, pow_out);
     // reduce the per-lane accumulators (only col==0 lanes are nonzero)
     for (int w = 32; w >= 1; w >>= 1) {
@@ -112,10 +122,44 @@ __device__ void dwell_physics_wave(EngA& E, int lane, float* pow_out) {
     }
 }
 
+// Lane-0 engine phases, __noinline__ so the trial loop's control flow
+// stays trivial: the fully-inlined engine (dispatch_one + model step) in
+// the SAME loop body produced a CFG whose reconvergence the compiler got
+// wrong (lanes 1..63 left the loop after the first dwell -> 1-lane MFMA).
+__device__ __noinline__ int engine_phase(EngA& E, int lane) {
+    if (lane != 0) return 0;
+    E.globals.phys_request = 0;
+    while (E.status == cmb::ST_OK && !E.evq.empty() &&
+           !E.globals.phys_request)
+        E.dispatch_one();
+    return E.globals.phys_request |
+           ((int)(uint32_t)E.globals.dwells << 1);
+}
+
+__device__ __noinline__ void engine_init_phase(EngA& E, int lane,
+                                               const AWACS::Params* dP,
+                                               uint64_t master_seed,
+                                               uint32_t trial) {
+    if (lane != 0) return;
+    E.init(dP, cmb::trial_seed(master_seed, trial), trial);
+    AWACS::setup(E);
+}
+
+__device__ __noinline__ void engine_resume_phase(EngA& E, int lane) {
+    if (lane != 0) return;
+    E.resume_proc(0, cmb::SIG_SUCCESS);
+}
+
+__device__ __noinline__ void engine_finish_phase(EngA& E, int lane,
+                                                 AWACS::Result* out) {
+    if (lane != 0) return;
+    AWACS::finish(E, *out);
+}
+
 __global__ __launch_bounds__(256) void awacs_kernel(
     const AWACS::Params* __restrict__ dP, uint64_t master_seed,
     uint32_t ntrials, AWACS::Result* __restrict__ out,
-    EngA* __restrict__ engines) {
+    EngA* __restrict__ engines, float* __restrict__ dbg, int scalar_phys) {
     const int lane = (int)(threadIdx.x & 63);
     const uint32_t wslot =
         blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
@@ -126,27 +170,45 @@ __global__ __launch_bounds__(256) void awacs_kernel(
 
     EngA& E = engines[wslot];
     for (uint32_t trial = wslot; trial < ntrials; trial += nwaves) {
-        if (lane == 0) {
-            E.init(dP, cmb::trial_seed(master_seed, trial), trial);
-            AWACS::setup(E);
-        }
+        engine_init_phase(E, lane, dP, master_seed, trial);
         for (;;) {
+            const int phase_word =
+                __builtin_amdgcn_readfirstlane(engine_phase(E, lane));
+            // publish lane 0's engine/event-loop stores to the other
+            // lanes: drain them (vmcnt), then agent-scope acquire so every
+            // lane's L1 drops its stale lines (guide §6 G16: vector L1 is
+            // not refreshed by another lane's stores; a workgroup-scope
+            // acquire would NOT invalidate L1).  Also covers the physics
+            // lanes' own cross-lane x/y reads from the previous dwell.
+            WAVE_FENCE();
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+            // uniform broadcast of lane 0's loop state via s_readfirstlane
+            // (scalar, exec-proof).  A __shfl of a lane-0-predicated load
+            // miscompiled here (lanes 1..63 read 0 and left the loop,
+            // leaving the MFMA with one active lane), and a volatile poll
+            // can spin on a stale L1 line - see docs/PARITY.md notes.
+            const int req = phase_word & 1;
+            const uint32_t dwl = (uint32_t)phase_word >> 1;
+            if (!req) break;
+            int mynt = 0;
+            float mydt = 0.0f;
             if (lane == 0) {
-                E.globals.phys_request = 0;
-                while (E.status == cmb::ST_OK && !E.evq.empty() &&
-                       !E.globals.phys_request)
-                    E.dispatch_one();
+                mynt = E.globals.nt;
+                mydt = (float)(E.now - E.globals.last_t);
+            }
+            const int nt = __builtin_amdgcn_readfirstlane(mynt);
+            const float dt = __int_as_float(
+                __builtin_amdgcn_readfirstlane(__float_as_int(mydt)));
+            float* dbg_now = (dbg && trial == 0 && dwl == 0) ? dbg : nullptr;
+            if (scalar_phys) {
+                if (lane == 0) AWACS::physics_all(E);
+            } else {
+                dwell_physics_wave(E, lane, dt, nt, (float)dP->area, dbg_now);
             }
             WAVE_FENCE();
-            // broadcast lane 0's view of the request
-            const int req =
-                __shfl((lane == 0) ? E.globals.phys_request : 0, 0);
-            if (!req) break;
-            dwell_physics_wave(E, lane, nullptr);
-            WAVE_FENCE();
-            if (lane == 0) E.resume_proc(0, cmb::SIG_SUCCESS);
+            engine_resume_phase(E, lane);
         }
-        if (lane == 0) AWACS::finish(E, out[trial]);
+        engine_finish_phase(E, lane, &out[trial]);
     }
 }
 
@@ -162,15 +224,79 @@ __global__ __launch_bounds__(64) void awacs_power_kernel(
         beamform_tile(E, base, lane, &det, &pw, pow_out);
 }
 
+// variant: engine initialized ON DEVICE by lane 0 (isolates the
+// lane-0-setup -> wave-visibility path of the full kernel)
+__global__ __launch_bounds__(64) void awacs_power_kernel_devinit(
+    const AWACS::Params* __restrict__ dP, uint64_t seed,
+    EngA* __restrict__ eng, float* __restrict__ pow_out) {
+    const int lane = (int)(threadIdx.x & 63);
+    EngA& E = *eng;
+    if (lane == 0) {
+        E.init(dP, seed, 0);
+        AWACS::setup(E);
+        E.params = nullptr;  // match the host-built variant (no draws)
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    unsigned long long det = 0;
+    double pw = 0.0;
+    for (int base = 0; base < E.globals.nt; base += 64)
+        beamform_tile(E, base, lane, &det, &pw, pow_out);
+}
+
 #define HIP_TRY(x)                                    \
     do {                                              \
         hipError_t err_ = (x);                        \
         if (err_ != hipSuccess) return (int)err_;     \
     } while (0)
 
+__global__ __launch_bounds__(64) void xlane_repro_kernel(int* buf,
+                                                          int iters) {
+    const int lane = (int)(threadIdx.x & 63);
+    for (int i = 0; i < iters; ++i) {
+        if (lane == 0) buf[0] = i + 1;
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        const int v = buf[0];
+        buf[2 + lane] += v;
+    }
+}
+
+// repro of the AWACS loop shape: masked store + shfl-broadcast ternary
+// load + data-dependent break; each lane should run `iters` iterations
+__global__ __launch_bounds__(64) void xlane_repro2_kernel(int* buf,
+                                                          int iters) {
+    const int lane = (int)(threadIdx.x & 63);
+    for (int it = 0;; ++it) {
+        if (lane == 0) buf[0] = (it < iters) ? 1 : 0;
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        const int req = __shfl((lane == 0) ? buf[0] : 0, 0);
+        if (!req) break;
+        buf[2 + lane] += 1;
+    }
+}
+
 }  // namespace
 
 extern "C" {
+
+// cross-lane store->load visibility microtest: each lane should accumulate
+// 1+2+...+iters
+int cimba_xlane_repro(int iters, int device, int* out64) {
+    HIP_TRY(hipSetDevice(device));
+    int* d = nullptr;
+    HIP_TRY(hipMalloc(&d, sizeof(int) * 66));
+    HIP_TRY(hipMemset(d, 0, sizeof(int) * 66));
+    hipLaunchKernelGGL(iters >= 0 ? xlane_repro_kernel : xlane_repro2_kernel,
+                       dim3(1), dim3(64), 0, 0, d,
+                       iters >= 0 ? iters : -iters);
+    HIP_TRY(hipGetLastError());
+    int h[66];
+    HIP_TRY(hipMemcpy(h, d, sizeof(int) * 66, hipMemcpyDeviceToHost));
+    HIP_TRY(hipFree(d));
+    for (int l = 0; l < 64; ++l) out64[l] = h[2 + l];
+    return 0;
+}
 
 int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                         int device, double* elapsed_ms, void* results_out) {
@@ -192,8 +318,10 @@ int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
     HIP_TRY(hipEventCreate(&t0));
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
+    const char* sp = getenv("CIMBA_AWACS_SCALAR");
     hipLaunchKernelGGL(awacs_kernel, dim3(blocks), dim3(256), 0, 0, d_P,
-                       seed, (uint32_t)ntrials, d_out, d_eng);
+                       seed, (uint32_t)ntrials, d_out, d_eng,
+                       (float*)nullptr, sp ? atoi(sp) : 0);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -207,6 +335,58 @@ int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
     HIP_TRY(hipFree(d_eng));
     HIP_TRY(hipEventDestroy(t0));
     HIP_TRY(hipEventDestroy(t1));
+    return 0;
+}
+
+// full-kernel first-dwell powers for trial 0 (debug instrumentation)
+int cimba_awacs_first_dwell_dbg(const void* params, uint64_t master_seed,
+                                int device, float* out_powers) {
+    HIP_TRY(hipSetDevice(device));
+    const AWACS::Params& P = *(const AWACS::Params*)params;
+    AWACS::Params* d_P = nullptr;
+    AWACS::Result* d_out = nullptr;
+    EngA* d_eng = nullptr;
+    float* d_dbg = nullptr;
+    HIP_TRY(hipMalloc(&d_P, sizeof(P)));
+    HIP_TRY(hipMemcpy(d_P, &P, sizeof(P), hipMemcpyHostToDevice));
+    HIP_TRY(hipMalloc(&d_out, sizeof(AWACS::Result)));
+    HIP_TRY(hipMalloc(&d_eng, sizeof(EngA) * 4));
+    HIP_TRY(hipMalloc(&d_dbg, sizeof(float) * AWACS::MAX_T));
+    HIP_TRY(hipMemset(d_dbg, 0, sizeof(float) * AWACS::MAX_T));
+    hipLaunchKernelGGL(awacs_kernel, dim3(1), dim3(256), 0, 0, d_P,
+                       master_seed, 1u, d_out, d_eng, d_dbg, 0);
+    HIP_TRY(hipGetLastError());
+    HIP_TRY(hipMemcpy(out_powers, d_dbg, sizeof(float) * AWACS::MAX_T,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipFree(d_P));
+    HIP_TRY(hipFree(d_out));
+    HIP_TRY(hipFree(d_eng));
+    HIP_TRY(hipFree(d_dbg));
+    return 0;
+}
+
+int cimba_awacs_power_test_devinit(const void* params, uint64_t seed,
+                                   int device, float* out_powers,
+                                   int* nt_out) {
+    HIP_TRY(hipSetDevice(device));
+    const AWACS::Params& P = *(const AWACS::Params*)params;
+    AWACS::Params* d_P = nullptr;
+    EngA* d_eng = nullptr;
+    float* d_pow = nullptr;
+    HIP_TRY(hipMalloc(&d_P, sizeof(P)));
+    HIP_TRY(hipMemcpy(d_P, &P, sizeof(P), hipMemcpyHostToDevice));
+    HIP_TRY(hipMalloc(&d_eng, sizeof(EngA)));
+    HIP_TRY(hipMalloc(&d_pow, sizeof(float) * AWACS::MAX_T));
+    HIP_TRY(hipMemset(d_pow, 0, sizeof(float) * AWACS::MAX_T));
+    hipLaunchKernelGGL(awacs_power_kernel_devinit, dim3(1), dim3(64), 0, 0,
+                       d_P, seed, d_eng, d_pow);
+    HIP_TRY(hipGetLastError());
+    HIP_TRY(hipMemcpy(out_powers, d_pow, sizeof(float) * AWACS::MAX_T,
+                      hipMemcpyDeviceToHost));
+    *nt_out = P.ntargets;
+    HIP_TRY(hipFree(d_P));
+    HIP_TRY(hipFree(d_eng));
+    HIP_TRY(hipFree(d_pow));
     return 0;
 }
 
