@@ -27,9 +27,10 @@ using cmb_models::MG1;
 using cmb_models::MM1;
 using cmb_models::Scenario;
 
-// generic trial-per-wavefront kernel; WPB = waves (= trials) per workgroup
-template <class Model, int WPB>
-__global__ __launch_bounds__(WPB * 64) __attribute__((flatten)) void trial_kernel(
+// generic trial-per-wavefront kernel; WPB = waves (= trials) per workgroup;
+// MINW = requested waves/SIMD (caps the register allocator; occupancy knob)
+template <class Model, int WPB, int MINW = 1>
+__global__ __launch_bounds__(WPB * 64, MINW) __attribute__((flatten)) void trial_kernel(
     typename Model::Params P, uint64_t master_seed, uint32_t ntrials,
     double until, uint64_t max_events, typename Model::Result* __restrict__ out) {
     __shared__ Engine<Model> eng[WPB];
@@ -58,7 +59,7 @@ static_assert(sizeof(Engine<JobShop>) * 4 < 64 * 1024, "JobShop engine LDS plan"
     } while (0)
 
 // host-side launcher: upload params, launch, copy per-trial results back
-template <class Model, int WPB>
+template <class Model, int WPB, int MINW = 1>
 int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
                    uint64_t seed, double until, uint64_t max_events,
                    double* elapsed_ms, typename Model::Result* host_out) {
@@ -72,9 +73,9 @@ int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
     HIP_TRY(hipEventCreate(&t0));
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
-    hipLaunchKernelGGL((trial_kernel<Model, WPB>), dim3(grid), dim3(WPB * 64),
-                       0, 0, P, seed, (uint32_t)ntrials, until, max_events,
-                       d_out);
+    hipLaunchKernelGGL((trial_kernel<Model, WPB, MINW>), dim3(grid),
+                       dim3(WPB * 64), 0, 0, P, seed, (uint32_t)ntrials,
+                       until, max_events, d_out);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -121,8 +122,25 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
     HIP_TRY(hipSetDevice(device));
     MM1::Params P{arr_mean, srv_mean, num_objects};
     std::vector<MM1::Result> res(ntrials);
-    int rc = run_trials_gpu<MM1, 4>(P, ntrials, seed, until, max_events,
-                                    &out->elapsed_ms, res.data());
+    const char* mw = getenv("CIMBA_MM1_MINW");
+    const int minw = mw ? atoi(mw) : 6;  // measured best (profiles/)
+    int rc;
+    switch (minw) {
+        case 5:
+            rc = run_trials_gpu<MM1, 4, 5>(P, ntrials, seed, until,
+                                           max_events, &out->elapsed_ms,
+                                           res.data());
+            break;
+        case 6:
+            rc = run_trials_gpu<MM1, 4, 6>(P, ntrials, seed, until,
+                                           max_events, &out->elapsed_ms,
+                                           res.data());
+            break;
+        default:
+            rc = run_trials_gpu<MM1, 4, 4>(P, ntrials, seed, until,
+                                           max_events, &out->elapsed_ms,
+                                           res.data());
+    }
     if (rc) return rc;
     out->total_events = 0;
     out->total_objs = 0;
@@ -145,6 +163,20 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
 int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                       int device, double* elapsed_ms, void* results_out) {
     HIP_TRY(hipSetDevice(device));
+    const char* mw = getenv("CIMBA_MG1_MINW");
+    const int minw = mw ? atoi(mw) : 4;
+    if (minw >= 4)
+        return run_trials_gpu<MG1, 4, 4>(*(const MG1::Params*)params, ntrials,
+                                         seed, 1.0e308,
+                                         UINT64_C(0xFFFFFFFFFFFFFFFF),
+                                         elapsed_ms,
+                                         (MG1::Result*)results_out);
+    if (minw == 3)
+        return run_trials_gpu<MG1, 4, 3>(*(const MG1::Params*)params, ntrials,
+                                         seed, 1.0e308,
+                                         UINT64_C(0xFFFFFFFFFFFFFFFF),
+                                         elapsed_ms,
+                                         (MG1::Result*)results_out);
     return run_trials_gpu<MG1, 4>(*(const MG1::Params*)params, ntrials, seed,
                                   1.0e308, UINT64_C(0xFFFFFFFFFFFFFFFF),
                                   elapsed_ms, (MG1::Result*)results_out);
@@ -153,6 +185,13 @@ int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
 int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                           int device, double* elapsed_ms, void* results_out) {
     HIP_TRY(hipSetDevice(device));
+    const char* mw = getenv("CIMBA_JS_MINW");
+    const int minw = mw ? atoi(mw) : 4;
+    if (minw >= 4)
+        return run_trials_gpu<JobShop, 4, 4>(
+            *(const JobShop::Params*)params, ntrials, seed, 1.0e308,
+            UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
+            (JobShop::Result*)results_out);
     return run_trials_gpu<JobShop, 4>(*(const JobShop::Params*)params, ntrials,
                                       seed, 1.0e308,
                                       UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
