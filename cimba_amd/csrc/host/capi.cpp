@@ -538,8 +538,16 @@ void cmb_resourcepool_initialize(cmb_sim* s, cmb_resourcepool* r,
     s->E->pools[dec(r)].capacity = capacity;
 }
 void cmb_resourcepool_release(cmb_sim* s, cmb_resourcepool* r,
-                              int32_t amount) {
-    s->E->pool_release(dec(r), amount);
+                              cmb_process* holder, int32_t amount) {
+    s->E->pool_release_for(dec(r), dec(holder), amount);
+}
+int32_t cmb_resourcepool_holding(const cmb_sim* s, const cmb_resourcepool* r,
+                                 const cmb_process* p) {
+    return s->E->pool_holding(dec(r), dec(p));
+}
+bool cmb_pool_try_preempt_(cmb_sim* s, cmb_resourcepool* r, cmb_process* p,
+                           int32_t want) {
+    return s->E->pool_try_preempt(dec(r), s->E->procs[dec(p)], want);
 }
 int32_t cmb_resourcepool_capacity(const cmb_sim* s,
                                   const cmb_resourcepool* r) {

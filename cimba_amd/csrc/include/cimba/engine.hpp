@@ -261,6 +261,10 @@ struct Engine {
     ObjQueue<Cfg::QCAP> queues[ArrOf<NQ>::n];
     Resource resources[ArrOf<NR>::n];
     Pool pools[ArrOf<NP>::n];
+    // per-process pool holdings (reference tracks holders in a hash-heap
+    // per pool for preemption victim choice, include/cmb_resourcepool.h:23-26;
+    // here a dense [proc][pool] table - bounded and branch-free)
+    int32_t pool_held[ArrOf<NP>::n * Cfg::MAX_PROC];
     Buffer buffers[ArrOf<NB>::n];
     PrioQueue<Cfg::PQCAP> pqueues[ArrOf<NPQ>::n];
     Condition conds[ArrOf<NC>::n];
@@ -313,6 +317,7 @@ struct Engine {
             pools[i].capacity = 1; pools[i].in_use = 0; pools[i].gid = (int16_t)g++;
             pools[i].recording = 0; pools[i].use_stats.reset(); pools[i].t_last = now;
         }
+        for (int i = 0; i < NP * Cfg::MAX_PROC; ++i) pool_held[i] = 0;
         for (int i = 0; i < NB; ++i) {
             buffers[i].level = 0; buffers[i].capacity = CMB_UNLIMITED;
             buffers[i].g_get = (int16_t)g++; buffers[i].g_put = (int16_t)g++;
@@ -512,8 +517,10 @@ struct Engine {
         for (int r = 0; r < NR; ++r) {
             if (resources[r].holder == (int16_t)pidx) resource_release(r);
         }
-        // pools: per-holder amounts are not tracked (see docs/PARITY.md);
-        // a killed process's pool units are released by the model's cleanup.
+        for (int p = 0; p < NP; ++p) {
+            const int32_t held = pool_held[p * Cfg::MAX_PROC + pidx];
+            if (held > 0) pool_release_for(p, pidx, held);
+        }
     }
 
     // ---- the await/resume discipline --------------------------------------
@@ -821,8 +828,61 @@ struct Engine {
                 pl.t_last = now;
             }
             pl.in_use += take;
+            pool_held[pi * Cfg::MAX_PROC + pidx_of(&p)] += take;
         }
         return take;
+    }
+
+    CMB_FORCEINLINE int32_t pool_holding(int pi, int pidx) const {
+        return pool_held[pi * Cfg::MAX_PROC + pidx];
+    }
+
+    // preemptive multi-unit acquire (reference cmb_resourcepool_preempt:
+    // grabs from lowest-priority holders, include/cmb_resourcepool.h:23-26).
+    // All-or-nothing: free units + units reclaimable from strictly-lower-
+    // priority holders must cover `want`; each victim is interrupted with
+    // SIG_PREEMPTED and its holding reduced (check pool_holding() after).
+    CMB_FORCEINLINE bool pool_try_preempt(int pi, ProcT& p, int32_t want) {
+        Pool& pl = pools[pi];
+        p.g_granted = 0;
+        const int me = pidx_of(&p);
+        const int32_t free_units = pl.capacity - pl.in_use;
+        int32_t need = want - free_units;
+        if (need > 0) {
+            int32_t reclaimable = 0;
+            for (int i = 0; i < Cfg::MAX_PROC; ++i) {
+                if (i != me && procs[i].priority < p.priority)
+                    reclaimable += pool_held[pi * Cfg::MAX_PROC + i];
+            }
+            if (reclaimable < need) return false;
+            while (need > 0) {
+                // lowest-priority victim first
+                int victim = -1;
+                for (int i = 0; i < Cfg::MAX_PROC; ++i) {
+                    if (i == me || pool_held[pi * Cfg::MAX_PROC + i] <= 0)
+                        continue;
+                    if (procs[i].priority >= p.priority) continue;
+                    if (victim < 0 ||
+                        procs[i].priority < procs[victim].priority)
+                        victim = i;
+                }
+                cmb_assert_debug(victim >= 0);
+                int32_t& vh = pool_held[pi * Cfg::MAX_PROC + victim];
+                const int32_t take = vh < need ? vh : need;
+                vh -= take;
+                need -= take;
+                proc_interrupt(victim, SIG_PREEMPTED);
+            }
+        }
+        if (pl.recording) {
+            pl.use_stats.add((double)pl.in_use, now - pl.t_last);
+            pl.t_last = now;
+        }
+        // transferred units stay in_use; only the free portion is new use
+        const int32_t from_free = want < free_units ? want : free_units;
+        pl.in_use += from_free;
+        pool_held[pi * Cfg::MAX_PROC + me] += want;
+        return true;
     }
 
     // all-or-nothing take (deadlock-free alternative to the greedy partial
@@ -837,18 +897,25 @@ struct Engine {
             pl.t_last = now;
         }
         pl.in_use += want;
+        pool_held[pi * Cfg::MAX_PROC + pidx_of(&p)] += want;
         return true;
     }
 
-    CMB_FORCEINLINE void pool_release(int pi, int32_t amount) {
+    CMB_FORCEINLINE void pool_release_for(int pi, int pidx, int32_t amount) {
         Pool& pl = pools[pi];
-        cmb_assert_debug(pl.in_use >= amount);
+        int32_t& held = pool_held[pi * Cfg::MAX_PROC + pidx];
+        cmb_assert_debug(pl.in_use >= amount && held >= amount);
         if (pl.recording) {
             pl.use_stats.add((double)pl.in_use, now - pl.t_last);
             pl.t_last = now;
         }
+        held -= amount;
         pl.in_use -= amount;
         guard_signal(pl.gid);
+    }
+
+    CMB_FORCEINLINE void pool_release(int pi, ProcT& p, int32_t amount) {
+        pool_release_for(pi, pidx_of(&p), amount);
     }
 
     CMB_FORCEINLINE bool buf_try_get(int bi, ProcT& p, int64_t amount) {
@@ -1089,7 +1156,13 @@ struct Engine {
         }                                                             \
     } while (0)
 
-#define CMB_POOL_RELEASE(pi, amount) E.pool_release((pi), (amount))
+#define CMB_POOL_RELEASE(pi, amount) E.pool_release((pi), *self, (amount))
+
+// preemptive multi-unit pool acquire (reference cmb_resourcepool_preempt)
+#define CMB_POOL_PREEMPT(pi, amount)                                       \
+    CMB_GUARDED_(E.pool_try_preempt((pi), *self, (amount)),               \
+                 E.pools[pi].gid, cmb::DEM_POOL_GE,                        \
+                 (uint32_t)(pi) | ((uint32_t)(amount) << 8))
 
 // all-or-nothing pool acquire: waits until `amount` units are free and
 // takes them atomically (deadlock-free for multi-unit requests, unlike the

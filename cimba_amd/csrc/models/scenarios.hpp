@@ -46,6 +46,7 @@ struct Scenario : cmb::ModelBase {
         W_WAIT_EVENT_CANCEL = 12,
         W_STALE_GRANT = 13,
         W_ABANDON = 14,
+        W_POOL_PREEMPT = 15,
     };
 
     struct Params {
@@ -130,6 +131,8 @@ struct Scenario : cmb::ModelBase {
         F_Q_PUTTER,         // hold d, put value a
         F_ABANDONER,        // hold 1, then abandon the trial (host logger
                             // error path; device: Engine::fail)
+        F_POOL_PREEMPTOR,   // hold d, preempt a units of pool 0, hold b,
+                            // release
     };
 
     template <class E_>
@@ -198,6 +201,19 @@ struct Scenario : cmb::ModelBase {
             CMB_BEGIN();
             CMB_HOLD(f.d);
             CMB_POOL_ACQUIRE(0, (int32_t)f.a, f.rem);
+            trace(E, me, T_ACQ);
+            CMB_HOLD((double)f.b);
+            // a preemptor may have taken units mid-hold: release what we
+            // still hold (reference preemption contract)
+            if (CMB_SIG() != cmb::SIG_SUCCESS) trace(E, me, sigtag(E, self));
+            CMB_POOL_RELEASE(0, E.pool_holding(0, me));
+            trace(E, me, T_REL);
+            CMB_END();
+        }
+        case F_POOL_PREEMPTOR: {
+            CMB_BEGIN();
+            CMB_HOLD(f.d);
+            CMB_POOL_PREEMPT(0, (int32_t)f.a);
             trace(E, me, T_ACQ);
             CMB_HOLD((double)f.b);
             CMB_POOL_RELEASE(0, (int32_t)f.a);
@@ -436,6 +452,14 @@ struct Scenario : cmb::ModelBase {
         case W_ABANDON:
             sp(E, 0, F_HOLDER, 0, 0, 0, 1.5);   // a bystander process
             sp(E, 1, F_ABANDONER, 0, 0, 0, 0.0);
+            break;
+        case W_POOL_PREEMPT:
+            // cap 4: p0 (pri 0) holds 3 for 10; p1 (pri 5) preempts 2 at
+            // t=1 (1 free + 1 reclaimed); p0's hold returns SIG_PREEMPTED
+            // and it releases its remaining 2 units
+            E.pools[0].capacity = 4;
+            sp(E, 0, F_POOL_USER, 0, 3, /*hold*/ 10, 0.0);
+            sp(E, 1, F_POOL_PREEMPTOR, 5, 2, /*hold*/ 2, 1.0);
             break;
         default:
             E.fail(cmb::ST_USER_ABORT);

@@ -236,6 +236,11 @@ void cmb_condition_wait_setup_(cmb_sim* sim, cmb_condition* c, cmb_process* p,
                        cmb_pool_try_take_all_((sim), (r), (me), (amount)), \
                        cmb_pool_wait_ge_((sim), (r), (me), (amount)))
 
+#define CMB_RESOURCEPOOL_PREEMPT(sim, me, r, amount)                       \
+    CMB_BLOCKING_LOOP_(sim, me,                                            \
+                       cmb_pool_try_preempt_((sim), (r), (me), (amount)),  \
+                       cmb_pool_wait_ge_((sim), (r), (me), (amount)))
+
 #define CMB_BUFFER_GET(sim, me, b, amount)                                 \
     CMB_BLOCKING_LOOP_(sim, me,                                            \
                        cmb_buffer_try_get_((sim), (b), (me), (amount)),    \
@@ -281,7 +286,12 @@ cmb_resourcepool* cmb_resourcepool_create(cmb_sim* sim);
 void cmb_resourcepool_initialize(cmb_sim* sim, cmb_resourcepool* r,
                                  const char* name, int32_t capacity);
 void cmb_resourcepool_release(cmb_sim* sim, cmb_resourcepool* r,
-                              int32_t amount);
+                              cmb_process* holder, int32_t amount);
+int32_t cmb_resourcepool_holding(const cmb_sim* sim,
+                                 const cmb_resourcepool* r,
+                                 const cmb_process* p);
+bool cmb_pool_try_preempt_(cmb_sim* sim, cmb_resourcepool* r, cmb_process* p,
+                           int32_t want);
 int32_t cmb_resourcepool_capacity(const cmb_sim* sim,
                                   const cmb_resourcepool* r);
 int32_t cmb_resourcepool_in_use(const cmb_sim* sim, const cmb_resourcepool* r);

@@ -39,6 +39,10 @@ EXPECTED = {
     # stale grant passes on: granted waiter times out at the same
     # timestamp, the grant must not be lost -> next waiter gets the object
     13: [(1.0, 11), (1.0, 1105), (1.0, 2097)],
+    # pool preemption: higher-priority taker reclaims units from a
+    # lower-priority holder; victim's hold returns SIG_PREEMPTED (101)
+    # and it releases only its remaining holding
+    15: [(0.0, 10), (1.0, 1010), (1.0, 101), (1.0, 11), (3.0, 1011)],
 }
 
 
