@@ -160,7 +160,8 @@ static py::dict awacs_power_check(int ntargets, uint64_t seed, int device) {
     if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
 
     // host f32 path + fp64 reference on the identical state
-    auto eng = std::make_unique<Engine<AWACS>>();
+    auto st = std::make_unique<Engine<AWACS>::Storage>();
+    auto eng = std::make_unique<Engine<AWACS>>(*st);
     eng->init(&p, seed, 0);
     AWACS::setup(*eng);
     const AWACS::Globals& g = eng->globals;
@@ -400,7 +401,8 @@ static py::dict scenario_result_to_dict(const Scenario::Result& r) {
 
 static py::dict scenario_host(int which) {
     Scenario::Params p{which};
-    auto eng = std::make_unique<Engine<Scenario>>();
+    auto st = std::make_unique<Engine<Scenario>::Storage>();
+    auto eng = std::make_unique<Engine<Scenario>>(*st);
     eng->init(&p, 123, 0);
     Scenario::setup(*eng);
     eng->run(1.0e308, 100000);
@@ -651,7 +653,8 @@ PYBIND11_MODULE(_C, m) {
                                              dev_pow.data());
         if (rc) throw std::runtime_error("hip " + std::to_string(rc));
         // host reference: trial 0 of the same master seed, first-dwell state
-        auto eng = std::make_unique<Engine<AWACS>>();
+        auto st = std::make_unique<Engine<AWACS>::Storage>();
+        auto eng = std::make_unique<Engine<AWACS>>(*st);
         eng->init(&p, trial_seed(master_seed, 0), 0);
         AWACS::setup(*eng);
         py::array_t<double> dev((py::ssize_t)ntargets), h((py::ssize_t)ntargets);
@@ -743,7 +746,7 @@ PYBIND11_MODULE(_C, m) {
           py::arg("device") = 0);
     m.def("fmix64", &py_fmix64);
     m.def("sfc64_raw", &py_sfc64_raw, py::arg("seed"), py::arg("skip") = 0);
-    m.def("engine_sizeof_mm1", []() { return sizeof(Engine<MM1>); });
+    m.def("engine_sizeof_mm1", []() { return sizeof(Engine<MM1>::Storage); });
 
     py::class_<DataSummary>(m, "DataSummary")
         .def(py::init([]() {

@@ -24,6 +24,8 @@ using cmb::Engine;
 using cmb_models::AWACS;
 
 using EngA = Engine<AWACS>;
+using StA = EngA::Storage;
+using GlA = AWACS::Globals;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 constexpr float PI_F = 3.14159265358979f;
@@ -32,10 +34,10 @@ constexpr float PI_F = 3.14159265358979f;
 // Input per lane: its A-operand row is target (base + sub*16 + (lane&15)),
 // K-group lane>>4; output per lane: D rows (lane>>4)*4+r, beam col lane&15
 // (the standard 16x16x4 C/D map).  Accumulates detections into det/pow.
-__device__ void beamform_tile(EngA& E, int base, int lane,
+__device__ void beamform_tile(GlA& g, uint32_t trial, uint32_t dwl,
+                              double snr_ref, int base, int lane,
                               unsigned long long* det_local,
                               double* pow_local, float* pow_out) {
-    AWACS::Globals& g = E.globals;
     const int col = lane & 15;
     const int kgrp = lane >> 4;
     for (int sub = 0; sub < 4; ++sub) {
@@ -76,10 +78,9 @@ __device__ void beamform_tile(EngA& E, int base, int lane,
                     if (pow_out) {
                         pow_out[t] = power;
                     }
-                    if (E.params) {
-                        if (AWACS::detect_draw(E.trial_index, g.dwells,
-                                               (uint32_t)t, power,
-                                               E.params->snr_ref)) {
+                    if (snr_ref > 0.0) {
+                        if (AWACS::detect_draw(trial, dwl, (uint32_t)t,
+                                               power, snr_ref)) {
                             g.det_cnt[t] += 1u;
                             *det_local += 1ull;
                         }
@@ -91,11 +92,12 @@ __device__ void beamform_tile(EngA& E, int base, int lane,
     }
 }
 
-// dt/nt/area are wave-uniform: computed on lane 0 and broadcast via
-// readfirstlane so no lane reads the engine's scalar block directly
-__device__ void dwell_physics_wave(EngA& E, int lane, float dt, int nt,
-                                   float area, float* pow_out) {
-    AWACS::Globals& g = E.globals;
+// dt/nt/area/trial/dwl/snr are wave-uniform: computed on lane 0 and
+// broadcast via readfirstlane so no lane reads the engine context
+__device__ void dwell_physics_wave(GlA& g, uint32_t trial, uint32_t dwl,
+                                   double snr_ref, int lane, float dt,
+                                   int nt, float area, double now,
+                                   float* pow_out) {
     for (int t = lane; t < nt; t += 64) {
         g.x[t] += g.vx[t] * dt;
         g.y[t] += g.vy[t] * dt;
@@ -108,7 +110,8 @@ __device__ void dwell_physics_wave(EngA& E, int lane, float dt, int nt,
     unsigned long long det_local = 0;
     double pow_local = 0.0;
     for (int base = 0; base < nt; base += 64)
-        beamform_tile(E, base, lane, &det_local, &pow_local, pow_out);
+        beamform_tile(g, trial, dwl, snr_ref, base, lane, &det_local,
+                      &pow_local, pow_out);
     // reduce the per-lane accumulators (only col==0 lanes are nonzero)
     for (int w = 32; w >= 1; w >>= 1) {
         det_local += __shfl_xor((unsigned long long)det_local, w);
@@ -117,7 +120,7 @@ __device__ void dwell_physics_wave(EngA& E, int lane, float dt, int nt,
     if (lane == 0) {
         g.detections += det_local;
         g.sum_power += pow_local;
-        g.last_t = E.now;
+        g.last_t = now;
         g.dwells += 1u;
     }
 }
@@ -159,7 +162,7 @@ __device__ __noinline__ void engine_finish_phase(EngA& E, int lane,
 __global__ __launch_bounds__(256) void awacs_kernel(
     const AWACS::Params* __restrict__ dP, uint64_t master_seed,
     uint32_t ntrials, AWACS::Result* __restrict__ out,
-    EngA* __restrict__ engines, float* __restrict__ dbg, int scalar_phys) {
+    StA* __restrict__ stores, float* __restrict__ dbg, int scalar_phys) {
     const int lane = (int)(threadIdx.x & 63);
     const uint32_t wslot =
         blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
@@ -168,7 +171,7 @@ __global__ __launch_bounds__(256) void awacs_kernel(
 // in L1 before other lanes load them (same CU -> visible)
 #define WAVE_FENCE() asm volatile("s_waitcnt vmcnt(0)" ::: "memory")
 
-    EngA& E = engines[wslot];
+    EngA E(stores[wslot]);  // lane 0's context is the live engine
     for (uint32_t trial = wslot; trial < ntrials; trial += nwaves) {
         engine_init_phase(E, lane, dP, master_seed, trial);
         for (;;) {
@@ -192,18 +195,24 @@ __global__ __launch_bounds__(256) void awacs_kernel(
             if (!req) break;
             int mynt = 0;
             float mydt = 0.0f;
+            float mynow = 0.0f;
             if (lane == 0) {
                 mynt = E.globals.nt;
                 mydt = (float)(E.now - E.globals.last_t);
+                mynow = (float)E.now;
             }
             const int nt = __builtin_amdgcn_readfirstlane(mynt);
             const float dt = __int_as_float(
                 __builtin_amdgcn_readfirstlane(__float_as_int(mydt)));
+            const double now_b = (double)__int_as_float(
+                __builtin_amdgcn_readfirstlane(__float_as_int(mynow)));
             float* dbg_now = (dbg && trial == 0 && dwl == 0) ? dbg : nullptr;
             if (scalar_phys) {
                 if (lane == 0) AWACS::physics_all(E);
             } else {
-                dwell_physics_wave(E, lane, dt, nt, (float)dP->area, dbg_now);
+                dwell_physics_wave(stores[wslot].globals, trial, dwl,
+                                   dP->snr_ref, lane, dt, nt,
+                                   (float)dP->area, now_b, dbg_now);
             }
             WAVE_FENCE();
             engine_resume_phase(E, lane);
@@ -215,33 +224,34 @@ __global__ __launch_bounds__(256) void awacs_kernel(
 // numerics-test kernel: one dwell's beamforming powers for a preloaded
 // engine state (host compares against the scalar fp32/fp64 reference)
 __global__ __launch_bounds__(64) void awacs_power_kernel(
-    EngA* __restrict__ eng, float* __restrict__ pow_out) {
+    StA* __restrict__ st, float* __restrict__ pow_out) {
     const int lane = (int)(threadIdx.x & 63);
     unsigned long long det = 0;
     double pw = 0.0;
-    EngA& E = *eng;
-    for (int base = 0; base < E.globals.nt; base += 64)
-        beamform_tile(E, base, lane, &det, &pw, pow_out);
+    GlA& g = st->globals;
+    for (int base = 0; base < g.nt; base += 64)
+        beamform_tile(g, 0, 0, /*snr_ref=0: no draws*/ 0.0, base, lane,
+                      &det, &pw, pow_out);
 }
 
 // variant: engine initialized ON DEVICE by lane 0 (isolates the
 // lane-0-setup -> wave-visibility path of the full kernel)
 __global__ __launch_bounds__(64) void awacs_power_kernel_devinit(
     const AWACS::Params* __restrict__ dP, uint64_t seed,
-    EngA* __restrict__ eng, float* __restrict__ pow_out) {
+    StA* __restrict__ st, float* __restrict__ pow_out) {
     const int lane = (int)(threadIdx.x & 63);
-    EngA& E = *eng;
+    EngA E(*st);
     if (lane == 0) {
         E.init(dP, seed, 0);
         AWACS::setup(E);
-        E.params = nullptr;  // match the host-built variant (no draws)
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     unsigned long long det = 0;
     double pw = 0.0;
-    for (int base = 0; base < E.globals.nt; base += 64)
-        beamform_tile(E, base, lane, &det, &pw, pow_out);
+    GlA& g = st->globals;
+    for (int base = 0; base < g.nt; base += 64)
+        beamform_tile(g, 0, 0, 0.0, base, lane, &det, &pw, pow_out);
 }
 
 #define HIP_TRY(x)                                    \
@@ -308,11 +318,11 @@ int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
 
     AWACS::Params* d_P = nullptr;
     AWACS::Result* d_out = nullptr;
-    EngA* d_eng = nullptr;
+    StA* d_eng = nullptr;
     HIP_TRY(hipMalloc(&d_P, sizeof(P)));
     HIP_TRY(hipMemcpy(d_P, &P, sizeof(P), hipMemcpyHostToDevice));
     HIP_TRY(hipMalloc(&d_out, sizeof(AWACS::Result) * ntrials));
-    HIP_TRY(hipMalloc(&d_eng, sizeof(EngA) * blocks * 4));
+    HIP_TRY(hipMalloc(&d_eng, sizeof(StA) * blocks * 4));
 
     hipEvent_t t0, t1;
     HIP_TRY(hipEventCreate(&t0));
@@ -345,12 +355,12 @@ int cimba_awacs_first_dwell_dbg(const void* params, uint64_t master_seed,
     const AWACS::Params& P = *(const AWACS::Params*)params;
     AWACS::Params* d_P = nullptr;
     AWACS::Result* d_out = nullptr;
-    EngA* d_eng = nullptr;
+    StA* d_eng = nullptr;
     float* d_dbg = nullptr;
     HIP_TRY(hipMalloc(&d_P, sizeof(P)));
     HIP_TRY(hipMemcpy(d_P, &P, sizeof(P), hipMemcpyHostToDevice));
     HIP_TRY(hipMalloc(&d_out, sizeof(AWACS::Result)));
-    HIP_TRY(hipMalloc(&d_eng, sizeof(EngA) * 4));
+    HIP_TRY(hipMalloc(&d_eng, sizeof(StA) * 4));
     HIP_TRY(hipMalloc(&d_dbg, sizeof(float) * AWACS::MAX_T));
     HIP_TRY(hipMemset(d_dbg, 0, sizeof(float) * AWACS::MAX_T));
     hipLaunchKernelGGL(awacs_kernel, dim3(1), dim3(256), 0, 0, d_P,
@@ -371,11 +381,11 @@ int cimba_awacs_power_test_devinit(const void* params, uint64_t seed,
     HIP_TRY(hipSetDevice(device));
     const AWACS::Params& P = *(const AWACS::Params*)params;
     AWACS::Params* d_P = nullptr;
-    EngA* d_eng = nullptr;
+    StA* d_eng = nullptr;
     float* d_pow = nullptr;
     HIP_TRY(hipMalloc(&d_P, sizeof(P)));
     HIP_TRY(hipMemcpy(d_P, &P, sizeof(P), hipMemcpyHostToDevice));
-    HIP_TRY(hipMalloc(&d_eng, sizeof(EngA)));
+    HIP_TRY(hipMalloc(&d_eng, sizeof(StA)));
     HIP_TRY(hipMalloc(&d_pow, sizeof(float) * AWACS::MAX_T));
     HIP_TRY(hipMemset(d_pow, 0, sizeof(float) * AWACS::MAX_T));
     hipLaunchKernelGGL(awacs_power_kernel_devinit, dim3(1), dim3(64), 0, 0,
@@ -398,17 +408,17 @@ int cimba_awacs_power_test(const void* params, uint64_t seed, int device,
     const AWACS::Params& P = *(const AWACS::Params*)params;
     // build the trial state on the HOST with the host engine (identical
     // setup path), then ship it to the device and run the MFMA dwell
-    auto host_eng = std::make_unique<EngA>();
+    auto host_st = std::make_unique<StA>();
+    EngA host_eng(*host_st);
     AWACS::Params p_local = P;
-    host_eng->init(&p_local, seed, 0);
-    AWACS::setup(*host_eng);
-    *nt_out = host_eng->globals.nt;
+    host_eng.init(&p_local, seed, 0);
+    AWACS::setup(host_eng);
+    *nt_out = host_eng.globals.nt;
 
-    EngA* d_eng = nullptr;
+    StA* d_eng = nullptr;
     float* d_pow = nullptr;
-    host_eng->params = nullptr;  // power kernel reads only globals (no draws)
-    HIP_TRY(hipMalloc(&d_eng, sizeof(EngA)));
-    HIP_TRY(hipMemcpy(d_eng, host_eng.get(), sizeof(EngA),
+    HIP_TRY(hipMalloc(&d_eng, sizeof(StA)));
+    HIP_TRY(hipMemcpy(d_eng, host_st.get(), sizeof(StA),
                       hipMemcpyHostToDevice));
     HIP_TRY(hipMalloc(&d_pow, sizeof(float) * AWACS::MAX_T));
     HIP_TRY(hipMemset(d_pow, 0, sizeof(float) * AWACS::MAX_T));

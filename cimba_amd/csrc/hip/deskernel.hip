@@ -33,10 +33,12 @@ template <class Model, int WPB, int MINW = 1>
 __global__ __launch_bounds__(WPB * 64, MINW) __attribute__((flatten)) void trial_kernel(
     typename Model::Params P, uint64_t master_seed, uint32_t ntrials,
     double until, uint64_t max_events, typename Model::Result* __restrict__ out) {
-    __shared__ Engine<Model> eng[WPB];
+    // arrays in LDS; the Engine context (clock, seq, handles, status, RNG
+    // state, heap size) is a per-lane local -> register-resident
+    __shared__ typename Engine<Model>::Storage st[WPB];
     const int w = (int)(threadIdx.x >> 6);
     if ((threadIdx.x & 63) != 0) return;  // lane 0 of each wave drives
-    Engine<Model>& E = eng[w];
+    Engine<Model> E(st[w]);
     const uint32_t stride = gridDim.x * WPB;
     for (uint32_t trial = blockIdx.x * WPB + (uint32_t)w; trial < ntrials;
          trial += stride) {
@@ -47,10 +49,11 @@ __global__ __launch_bounds__(WPB * 64, MINW) __attribute__((flatten)) void trial
     }
 }
 
-// LDS budget checks: WPB engines per workgroup, multiple workgroups per CU
-static_assert(sizeof(Engine<MM1>) * 4 < 60 * 1024, "MM1 engine LDS plan");
-static_assert(sizeof(Engine<MG1>) * 4 < 64 * 1024, "MG1 engine LDS plan");
-static_assert(sizeof(Engine<JobShop>) * 4 < 64 * 1024, "JobShop engine LDS plan");
+// LDS budget checks: WPB engine stores per workgroup, several workgroups/CU
+static_assert(sizeof(Engine<MM1>::Storage) * 4 < 60 * 1024, "MM1 LDS plan");
+static_assert(sizeof(Engine<MG1>::Storage) * 4 < 64 * 1024, "MG1 LDS plan");
+static_assert(sizeof(Engine<JobShop>::Storage) * 4 < 64 * 1024,
+              "JobShop LDS plan");
 
 #define HIP_TRY(x)                                    \
     do {                                              \
@@ -94,8 +97,9 @@ int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
 // the trace must match the host engine exactly (tests/test_gpu.py)
 __global__ __launch_bounds__(64) __attribute__((flatten)) void scenario_kernel(
     Scenario::Params P, Scenario::Result* __restrict__ out) {
-    __shared__ Engine<Scenario> eng;
+    __shared__ Engine<Scenario>::Storage st;
     if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    Engine<Scenario> eng(st);
     eng.init(&P, 123, 0);
     Scenario::setup(eng);
     eng.run(1.0e308, 100000);
@@ -123,7 +127,7 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
     MM1::Params P{arr_mean, srv_mean, num_objects};
     std::vector<MM1::Result> res(ntrials);
     const char* mw = getenv("CIMBA_MM1_MINW");
-    const int minw = mw ? atoi(mw) : 6;  // measured best (profiles/)
+    const int minw = mw ? atoi(mw) : 5;  // measured best (profiles/)
     int rc;
     switch (minw) {
         case 5:

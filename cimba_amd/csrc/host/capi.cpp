@@ -84,6 +84,9 @@ struct cmb_sim {
     CModel::Params params;
     uint64_t seed;
 };
+namespace {
+using CStorage = CEngine::Storage;
+}
 
 namespace {
 
@@ -175,7 +178,8 @@ uint64_t cimba_run(void* experiment, uint64_t n, size_t size,
 
     auto worker = [&](int widx) {
         if (g_thread_init) g_thread_init(widx);
-        auto eng = std::make_unique<CEngine>();
+        auto store = std::make_unique<CStorage>();
+        auto eng = std::make_unique<CEngine>(*store);
         cmb_sim sim;
         sim.E = eng.get();
         sim.params.sim = &sim;

@@ -242,7 +242,29 @@ struct Engine {
             ? 2 * NQ + NR + NP + 2 * NB + 2 * NPQ + NC
             : 1;
 
-    // ---- state ----
+    // ---- cold storage -----------------------------------------------------
+    // The POD block that holds every array: LDS (or HBM) on the device,
+    // heap on the host.  Hot scalars live in the Engine context below.
+    struct Storage {
+        EvEntry evbuf[Cfg::MAX_EV];
+        typename Model::Globals globals;
+        ProcT procs[Cfg::MAX_PROC];
+        Frame frames[Cfg::MAX_PROC];
+        Guard guards[NGUARD];
+        ObjQueue<Cfg::QCAP> queues[ArrOf<NQ>::n];
+        Resource resources[ArrOf<NR>::n];
+        Pool pools[ArrOf<NP>::n];
+        // per-process pool holdings (reference tracks holders in a
+        // hash-heap per pool for preemption victim choice,
+        // include/cmb_resourcepool.h:23-26; here a dense [pool][proc]
+        // table - bounded and branch-free)
+        int32_t pool_held[ArrOf<NP>::n * Cfg::MAX_PROC];
+        Buffer buffers[ArrOf<NB>::n];
+        PrioQueue<Cfg::PQCAP> pqueues[ArrOf<NPQ>::n];
+        Condition conds[ArrOf<NC>::n];
+    };
+
+    // ---- hot context (register-resident on the device) --------------------
     double now;
     uint64_t ev_dispatched;   // the benchmark metric: events executed
     uint64_t seq;             // event FIFO counter
@@ -252,22 +274,26 @@ struct Engine {
     uint32_t trial_index;
     const Params* params;
     Rng rng;
+    HashHeap<Cfg::MAX_EV> evq;  // view: entries in Storage, size here
 
-    HashHeap<Cfg::MAX_EV> evq;
-    typename Model::Globals globals;  // model-wide per-trial state
-    ProcT procs[Cfg::MAX_PROC];
-    Frame frames[Cfg::MAX_PROC];
-    Guard guards[NGUARD];
-    ObjQueue<Cfg::QCAP> queues[ArrOf<NQ>::n];
-    Resource resources[ArrOf<NR>::n];
-    Pool pools[ArrOf<NP>::n];
-    // per-process pool holdings (reference tracks holders in a hash-heap
-    // per pool for preemption victim choice, include/cmb_resourcepool.h:23-26;
-    // here a dense [proc][pool] table - bounded and branch-free)
-    int32_t pool_held[ArrOf<NP>::n * Cfg::MAX_PROC];
-    Buffer buffers[ArrOf<NB>::n];
-    PrioQueue<Cfg::PQCAP> pqueues[ArrOf<NPQ>::n];
-    Condition conds[ArrOf<NC>::n];
+    // ---- references into storage (existing field syntax keeps working) ----
+    typename Model::Globals& globals;
+    ProcT (&procs)[Cfg::MAX_PROC];
+    Frame (&frames)[Cfg::MAX_PROC];
+    Guard (&guards)[NGUARD];
+    ObjQueue<Cfg::QCAP> (&queues)[ArrOf<NQ>::n];
+    Resource (&resources)[ArrOf<NR>::n];
+    Pool (&pools)[ArrOf<NP>::n];
+    int32_t (&pool_held)[ArrOf<NP>::n * Cfg::MAX_PROC];
+    Buffer (&buffers)[ArrOf<NB>::n];
+    PrioQueue<Cfg::PQCAP> (&pqueues)[ArrOf<NPQ>::n];
+    Condition (&conds)[ArrOf<NC>::n];
+
+    CMB_FORCEINLINE explicit Engine(Storage& s)
+        : evq(s.evbuf), globals(s.globals), procs(s.procs), frames(s.frames),
+          guards(s.guards), queues(s.queues), resources(s.resources),
+          pools(s.pools), pool_held(s.pool_held), buffers(s.buffers),
+          pqueues(s.pqueues), conds(s.conds) {}
 
     // ---- lifecycle --------------------------------------------------------
 

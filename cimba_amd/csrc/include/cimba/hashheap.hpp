@@ -44,10 +44,16 @@ CMB_FORCEINLINE bool ev_less(const EvEntry& x, const EvEntry& y) {
     return x.pseq < y.pseq;
 }
 
+// The heap is a VIEW: `e` references a caller-owned entry array (LDS or
+// HBM) while the size `n` lives in the view itself — on the device the
+// view is part of the register-resident Engine context, so the hot-loop
+// size checks and the count never round-trip through LDS.
 template <int CAP>
 struct HashHeap {
-    EvEntry e[CAP];
+    EvEntry (&e)[CAP];
     int32_t n;
+
+    CMB_FORCEINLINE explicit HashHeap(EvEntry (&buf)[CAP]) : e(buf), n(0) {}
 
     CMB_FORCEINLINE void reset() { n = 0; }
     CMB_FORCEINLINE bool empty() const { return n == 0; }

@@ -65,7 +65,8 @@ RunReport run_host(const typename Model::Params& params, uint64_t master_seed,
         if (hooks && hooks->thread_init) hooks->thread_init(widx);
         // one engine per worker, reused across trials (reference: per-thread
         // event queue reset between trials, cimba.c:332)
-        auto eng = std::make_unique<Engine<Model>>();
+        auto store = std::make_unique<typename Engine<Model>::Storage>();
+        auto eng = std::make_unique<Engine<Model>>(*store);
         for (;;) {
             const uint64_t t = next.fetch_add(1, std::memory_order_relaxed);
             if (t >= ntrials) break;
