@@ -30,9 +30,9 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
-    ap.add_argument("--trials", type=int, default=8192,
+    ap.add_argument("--trials", type=int, default=262144,
                     help="replications per GPU per step")
-    ap.add_argument("--objects", type=int, default=100000,
+    ap.add_argument("--objects", type=int, default=10000,
                     help="objects per replication")
     ap.add_argument("--seed", type=lambda s: int(s, 0), default=0x34F05C64D7AD598F)
     ap.add_argument("--host", action="store_true",
@@ -148,7 +148,7 @@ def main():
                 "arrival_rate": 0.9,
                 "service_rate": 1.0,
                 "parallelism": f"trial-parallel dp{n_gpus}",
-                "engine": "trial-per-wavefront, LDS-resident",
+                "engine": "trial-per-lane (64/wave) for large batches, trial-per-wavefront + LDS otherwise",
                 "device": "gpu" if use_gpu else "cpu-host-debug",
             },
         }

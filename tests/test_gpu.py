@@ -37,6 +37,19 @@ def test_gpu_mm1_no_aborts_at_scale():
     assert 9.0 < r["avg_system_time"] < 11.0
 
 
+def test_lane_kernel_matches_wave_kernel():
+    import os
+    os.environ["CIMBA_MM1_LANE"] = "1"
+    try:
+        a = ca.mm1_gpu(ntrials=512, num_objects=5000, seed=13, device=0)
+    finally:
+        os.environ["CIMBA_MM1_LANE"] = "0"
+    b = ca.mm1_gpu(ntrials=512, num_objects=5000, seed=13, device=0)
+    del os.environ["CIMBA_MM1_LANE"]
+    assert a["total_wait"] == b["total_wait"]
+    assert a["total_events"] == b["total_events"]
+
+
 def test_bench_script_single_gpu():
     out = subprocess.run(
         [sys.executable, "bench.py", "--gpus", "1", "--steps", "1",
