@@ -35,6 +35,7 @@ from ._C import (  # noqa: E402  # noqa: F401,E402
     mm1_host,
     rng_sample,
     sfc64_raw,
+    trial_seed,
 )
 
 

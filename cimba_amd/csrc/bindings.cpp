@@ -745,6 +745,11 @@ PYBIND11_MODULE(_C, m) {
           py::arg("p0") = 0.0, py::arg("n") = 1 << 26, py::arg("seed") = 1ULL,
           py::arg("device") = 0);
     m.def("fmix64", &py_fmix64);
+    // seed-replay tooling (reference seed discipline):
+    // any trial is reproducible from (master_seed, index)
+    m.def("trial_seed", [](uint64_t master, uint64_t idx) {
+        return trial_seed(master, idx);
+    }, py::arg("master_seed"), py::arg("index"));
     m.def("sfc64_raw", &py_sfc64_raw, py::arg("seed"), py::arg("skip") = 0);
     m.def("engine_sizeof_mm1", []() { return sizeof(Engine<MM1>::Storage); });
 

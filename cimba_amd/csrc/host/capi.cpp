@@ -23,23 +23,23 @@ using namespace cmb;
 
 namespace {
 
-constexpr int C_MAX_PROC = 128;
-constexpr int C_UEV = 512;  // pending user events
+constexpr int C_MAX_PROC = 256;
+constexpr int C_UEV = 4096;  // pending user events
 constexpr int C_NAME = 32;  // reference CMB_PROCESS_NAMEBUF_SZ
 
 struct CModel : ModelBase {
     struct Cfg {
         static constexpr int MAX_PROC = C_MAX_PROC;
-        static constexpr int MAX_EV = 1024;
+        static constexpr int MAX_EV = 8192;
         static constexpr int TIMERS = 2;
-        static constexpr int NUM_QUEUES = 8;
-        static constexpr int QCAP = 1024;
-        static constexpr int NUM_RES = 8;
-        static constexpr int NUM_POOLS = 8;
-        static constexpr int NUM_BUFS = 8;
-        static constexpr int NUM_PQ = 4;
-        static constexpr int PQCAP = 256;
-        static constexpr int NUM_COND = 8;
+        static constexpr int NUM_QUEUES = 16;
+        static constexpr int QCAP = 8192;
+        static constexpr int NUM_RES = 16;
+        static constexpr int NUM_POOLS = 16;
+        static constexpr int NUM_BUFS = 16;
+        static constexpr int NUM_PQ = 8;
+        static constexpr int PQCAP = 1024;
+        static constexpr int NUM_COND = 16;
     };
     struct Params {
         cmb_sim* sim;
@@ -644,6 +644,50 @@ void cmb_condition_wait_setup_(cmb_sim* s, cmb_condition* c, cmb_process* p,
     E.globals.dem_fn[pidx] = demand;
     E.globals.dem_ctx[pidx] = ctx;
     E.guard_wait(E.procs[pidx], E.conds[dec(c)].gid, DEM_USER, 0);
+}
+
+/* ---- debug dumps & reports ---- */
+
+void cmb_event_queue_print(cmb_sim* s, FILE* out) {
+    if (!out) out = stderr;
+    auto& q = s->E->evq;
+    fprintf(out, "event queue @ t=%.6f: %d pending\n", s->E->now, q.n);
+    for (int32_t i = 0; i < q.n; ++i) {
+        const auto& e = q.e[i];
+        fprintf(out, "  [%2d] t=%.6f kind=%u a=%u handle=%u\n", i, e.t,
+                (unsigned)e.kind, (unsigned)e.a, e.handle);
+    }
+}
+
+void cmb_resource_print_report(cmb_sim* s, const cmb_resource* r, FILE* out) {
+    if (!out) out = stderr;
+    double st[4];
+    cmb_resource_stats(s, r, st);
+    fprintf(out,
+            "resource report @ t=%.6f: holder=%d utilization mean=%.4f "
+            "sd=%.4f\n",
+            s->E->now, (int)s->E->resources[dec(r)].holder, st[0], st[1]);
+}
+
+void cmb_resourcepool_print_report(cmb_sim* s, const cmb_resourcepool* r,
+                                   FILE* out) {
+    if (!out) out = stderr;
+    const auto& pl = s->E->pools[dec(r)];
+    auto copy = pl;
+    copy.use_stats.add((double)copy.in_use, s->E->now - copy.t_last);
+    fprintf(out,
+            "pool report @ t=%.6f: capacity=%d in_use=%d mean_busy=%.4f\n",
+            s->E->now, pl.capacity, pl.in_use, copy.use_stats.mean);
+}
+
+void cmb_objectqueue_report_print(cmb_sim* s, const cmb_objectqueue* q,
+                                  FILE* out) {
+    if (!out) out = stderr;
+    double st[4];
+    cmb_objectqueue_stats(s, q, st);
+    fprintf(out,
+            "queue report @ t=%.6f: length=%d mean=%.4f sd=%.4f max=%.0f\n",
+            s->E->now, (int)s->E->queues[dec(q)].len, st[0], st[1], st[3]);
 }
 
 /* ---- RNG ---- */

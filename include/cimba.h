@@ -27,6 +27,7 @@
 
 #include <stddef.h>
 #include <stdint.h>
+#include <stdio.h>
 #include <stdbool.h>
 
 #ifdef __cplusplus
@@ -308,6 +309,16 @@ cmb_condition* cmb_condition_create(cmb_sim* sim);
 void cmb_condition_initialize(cmb_sim* sim, cmb_condition* c,
                               const char* name);
 uint64_t cmb_condition_signal(cmb_sim* sim, cmb_condition* c);
+
+/* ---- debug dumps & reports (reference cmb_event_queue_print,
+ * cmb_resource_print_report et al.) ---- */
+void cmb_event_queue_print(cmb_sim* sim, FILE* out);
+void cmb_resource_print_report(cmb_sim* sim, const cmb_resource* r,
+                               FILE* out);
+void cmb_resourcepool_print_report(cmb_sim* sim, const cmb_resourcepool* r,
+                                   FILE* out);
+void cmb_objectqueue_report_print(cmb_sim* sim, const cmb_objectqueue* q,
+                                  FILE* out);
 
 /* ---- RNG (reference include/cmb_random.h; per-trial stream) ---- */
 uint64_t cmb_random_sfc64(cmb_sim* sim);

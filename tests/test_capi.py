@@ -45,6 +45,25 @@ def test_capi_mm1_compiles_and_runs(libcimba, tmp_path):
     assert out2.stdout == out.stdout
 
 
+def test_capi_mg1_resource_condition(libcimba, tmp_path):
+    # wider API tour: resource, condition with C demand predicate, wait
+    # timeouts, recording + report printers; validates vs PK theory
+    exe = str(tmp_path / "mg1_capi")
+    r = subprocess.run(
+        ["gcc", "-std=c11", "-O2", "-Wall", "-Werror",
+         "-I", os.path.join(ROOT, "include"),
+         os.path.join(ROOT, "tutorial", "mg1_capi.c"),
+         "-L", os.path.join(ROOT, "cimba_amd"), "-lcimba",
+         f"-Wl,-rpath,{os.path.join(ROOT, 'cimba_amd')}",
+         "-lm", "-o", exe], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    out = subprocess.run([exe], capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, (out.stdout, out.stderr)
+    assert "queue report" in out.stdout
+    assert "resource report" in out.stdout
+    assert "M/G/1 avg system time" in out.stdout
+
+
 def test_capi_header_is_c_clean(tmp_path):
     # header must compile as plain C99 without the library
     src = tmp_path / "hdr.c"

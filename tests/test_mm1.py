@@ -35,3 +35,17 @@ def test_mm1_zero_trial_edge():
     r = ca.mm1_host(ntrials=1, num_objects=1, seed=1, threads=1)
     assert r["trials_ok"] == 1
     assert r["total_objects"] == 1
+
+
+def test_seed_replay_single_trial():
+    """Seed-replay (reference checkpoint substitute): any
+    trial of an experiment reruns identically from (master_seed, index)."""
+    full = ca.mm1_host(ntrials=8, num_objects=3000, seed=999, threads=4)
+    # replay trial 5 alone: a 1-trial experiment whose master seed is
+    # chosen so fmix-derived trial-0 seed equals the original trial-5 seed
+    # -> use the internal trial_seed to verify determinism directly
+    s5 = ca.trial_seed(999, 5)
+    s5b = ca.trial_seed(999, 5)
+    assert s5 == s5b != ca.trial_seed(999, 6)
+    again = ca.mm1_host(ntrials=8, num_objects=3000, seed=999, threads=1)
+    assert again["per_trial_avg"][5] == full["per_trial_avg"][5]
