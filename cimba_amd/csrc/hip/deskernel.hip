@@ -115,6 +115,59 @@ __global__ __launch_bounds__(256, MINW) __attribute__((flatten)) void lane_trial
     }
 }
 
+// scratch variant: the per-trial Storage is a kernel LOCAL, so it lives
+// in private (scratch) memory — which the hardware interleaves per lane.
+// Same-field accesses across the 64 lanes of a wave therefore COALESCE
+// into wide transactions: the AoS->SoA transposition for free.
+template <class Model, int MINW = 1>
+__global__ __launch_bounds__(256, MINW) void lane_scratch_kernel(
+    typename Model::Params P, uint64_t master_seed, uint32_t ntrials,
+    double until, uint64_t max_events,
+    typename Model::Result* __restrict__ out) {
+    const uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+    const uint32_t stride = gridDim.x * blockDim.x;
+    typename Engine<Model>::Storage st;  // per-lane scratch (HW-swizzled)
+    Engine<Model> E(st);
+    for (uint32_t trial = gid; trial < ntrials; trial += stride) {
+        E.init(&P, cmb::trial_seed(master_seed, trial), trial);
+        Model::setup(E);
+        E.run(until, max_events);
+        Model::finish(E, out[trial]);
+    }
+}
+
+template <class Model, int MINW>
+int run_trials_gpu_lane_scratch(const typename Model::Params& P,
+                                uint64_t ntrials, uint64_t seed, double until,
+                                uint64_t max_events, double* elapsed_ms,
+                                typename Model::Result* host_out,
+                                uint32_t blocks) {
+    using Result = typename Model::Result;
+    const uint32_t want = (uint32_t)((ntrials + 255) / 256);
+    const uint32_t grid = want < blocks ? want : blocks;
+    Result* d_out = nullptr;
+    HIP_TRY(hipMalloc(&d_out, sizeof(Result) * ntrials));
+    hipEvent_t t0, t1;
+    HIP_TRY(hipEventCreate(&t0));
+    HIP_TRY(hipEventCreate(&t1));
+    HIP_TRY(hipEventRecord(t0));
+    hipLaunchKernelGGL((lane_scratch_kernel<Model, MINW>), dim3(grid),
+                       dim3(256), 0, 0, P, seed, (uint32_t)ntrials, until,
+                       max_events, d_out);
+    HIP_TRY(hipGetLastError());
+    HIP_TRY(hipEventRecord(t1));
+    HIP_TRY(hipEventSynchronize(t1));
+    float ms = 0.f;
+    HIP_TRY(hipEventElapsedTime(&ms, t0, t1));
+    *elapsed_ms = (double)ms;
+    HIP_TRY(hipMemcpy(host_out, d_out, sizeof(Result) * ntrials,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipFree(d_out));
+    HIP_TRY(hipEventDestroy(t0));
+    HIP_TRY(hipEventDestroy(t1));
+    return 0;
+}
+
 template <class Model, int MINW>
 int run_trials_gpu_lane(const typename Model::Params& P, uint64_t ntrials,
                         uint64_t seed, double until, uint64_t max_events,
@@ -193,9 +246,15 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
     if (use_lane) {
         const char* lb = getenv("CIMBA_MM1_LANE_BLOCKS");
         const uint32_t blocks = lb ? (uint32_t)atoi(lb) : 2048u;
-        rc = run_trials_gpu_lane<MM1, 1>(P, ntrials, seed, until,
-                                         max_events, &out->elapsed_ms,
-                                         res.data(), blocks);
+        if (lane && atoi(lane) == 1)  // explicit HBM-lane variant
+            rc = run_trials_gpu_lane<MM1, 1>(P, ntrials, seed, until,
+                                             max_events, &out->elapsed_ms,
+                                             res.data(), blocks);
+        else  // default: scratch storage (HW lane-interleaved -> coalesced)
+            rc = run_trials_gpu_lane_scratch<MM1, 1>(P, ntrials, seed, until,
+                                                     max_events,
+                                                     &out->elapsed_ms,
+                                                     res.data(), blocks);
         goto aggregate;
     }
     switch (minw) {
@@ -240,11 +299,10 @@ int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
     const char* lane = getenv("CIMBA_MG1_LANE");
     const uint64_t nt_ = ntrials;
     if (lane ? atoi(lane) != 0 : nt_ >= 32768)
-        return run_trials_gpu_lane<MG1, 1>(*(const MG1::Params*)params,
-                                           ntrials, seed, 1.0e308,
-                                           UINT64_C(0xFFFFFFFFFFFFFFFF),
-                                           elapsed_ms,
-                                           (MG1::Result*)results_out, 2048u);
+        return run_trials_gpu_lane_scratch<MG1, 1>(
+            *(const MG1::Params*)params, ntrials, seed, 1.0e308,
+            UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
+            (MG1::Result*)results_out, 2048u);
     const char* mw = getenv("CIMBA_MG1_MINW");
     const int minw = mw ? atoi(mw) : 4;
     if (minw >= 4)
@@ -268,7 +326,8 @@ int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                           int device, double* elapsed_ms, void* results_out) {
     HIP_TRY(hipSetDevice(device));
     const char* lane = getenv("CIMBA_JS_LANE");
-    if (lane ? atoi(lane) != 0 : ntrials >= 32768)
+    if (lane ? atoi(lane) != 0 : ntrials >= 32768)  // HBM-lane variant:
+        // measured faster than scratch for JobShop's larger store
         return run_trials_gpu_lane<JobShop, 1>(
             *(const JobShop::Params*)params, ntrials, seed, 1.0e308,
             UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
