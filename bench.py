@@ -76,6 +76,7 @@ def main():
 
     def one_step(step_idx):
         seed = ca.fmix64((args.seed ^ (rank << 32)) + step_idx + 1)
+        ntrials = args.trials
         if args.model == "mm1":
             fn = ca.mm1_gpu if use_gpu else ca.mm1_host
             kw = dict(ntrials=args.trials, num_objects=args.objects, seed=seed)
@@ -89,17 +90,20 @@ def main():
             kw = dict(ntrials=args.trials, entities=args.objects, njobs=24,
                       seed=seed)
         else:
+            # AWACS cost ~ trials x duration/dwell x targets: scale the
+            # defaults down to a comparable per-step budget
+            ntrials = min(args.trials, 4096)
             fn = ca._C.awacs_gpu if use_gpu else ca._C.awacs_host
-            kw = dict(ntrials=args.trials, duration=args.objects / 25.0,
+            kw = dict(ntrials=ntrials, duration=min(args.objects / 250.0, 60.0),
                       ntargets=1000, seed=seed)
         if use_gpu:
             kw["device"] = local_rank
         else:
             kw["threads"] = 0
         r = fn(**kw)
-        if r["trials_ok"] != args.trials:
+        if r["trials_ok"] != ntrials:
             raise RuntimeError(
-                f"rank {rank}: {args.trials - r['trials_ok']} trials aborted "
+                f"rank {rank}: {ntrials - r['trials_ok']} trials aborted "
                 f"(status {r['first_bad_status']})")
         return r["total_events"]
 
