@@ -401,6 +401,15 @@ void cmb_objectqueue_initialize(cmb_sim* s, cmb_objectqueue* q,
 uint64_t cmb_objectqueue_length(const cmb_sim* s, const cmb_objectqueue* q) {
     return (uint64_t)s->E->queues[dec(q)].len;
 }
+uint64_t cmb_objectqueue_position(const cmb_sim* s, const cmb_objectqueue* q,
+                                  const void* object) {
+    const auto& Q = s->E->queues[dec(q)];
+    for (int32_t i = 0; i < Q.len; ++i) {
+        const int32_t idx = (Q.head + i) % CModel::Cfg::QCAP;
+        if (Q.ring[idx] == (uint64_t)object) return (uint64_t)i + 1;
+    }
+    return 0;
+}
 void cmb_objectqueue_recording_start(cmb_sim* s, cmb_objectqueue* q) {
     auto& Q = s->E->queues[dec(q)];
     Q.recording = 1;
@@ -854,22 +863,34 @@ double cmb_wtdsummary_variance(const cmb_wtdsummary* s) {
 
 void cmb_logger_flags_on(uint32_t flags) { logger_flags_on(flags); }
 void cmb_logger_flags_off(uint32_t flags) { logger_flags_off(flags); }
-void cmb_logger_info(cmb_sim* s, const char* fmt, ...) {
+static cmb_timeformatter_func* g_timefmt = nullptr;
+void cmb_logger_timeformatter_set(cmb_timeformatter_func* fmt) {
+    g_timefmt = fmt;
+}
+static void apply_timefmt(cmb_sim* s) {
     logger_ctx().sim_time = s->E->now;
+    if (g_timefmt) {
+        static thread_local char buf[48];
+        g_timefmt(s->E->now, buf, sizeof(buf));
+        logger_ctx().who = buf;
+    }
+}
+void cmb_logger_info(cmb_sim* s, const char* fmt, ...) {
+    apply_timefmt(s);
     va_list ap;
     va_start(ap, fmt);
     logger_vlog(LOG_INFO, "info", fmt, ap);
     va_end(ap);
 }
 void cmb_logger_warning(cmb_sim* s, const char* fmt, ...) {
-    logger_ctx().sim_time = s->E->now;
+    apply_timefmt(s);
     va_list ap;
     va_start(ap, fmt);
     logger_vlog(LOG_WARNING, "warning", fmt, ap);
     va_end(ap);
 }
 void cmb_logger_error(cmb_sim* s, const char* fmt, ...) {
-    logger_ctx().sim_time = s->E->now;
+    apply_timefmt(s);
     va_list ap;
     va_start(ap, fmt);
     logger_vlog(LOG_ERROR, "error", fmt, ap);

@@ -263,6 +263,11 @@ cmb_objectqueue* cmb_objectqueue_create(cmb_sim* sim);
 void cmb_objectqueue_initialize(cmb_sim* sim, cmb_objectqueue* q,
                                 const char* name, int32_t capacity);
 uint64_t cmb_objectqueue_length(const cmb_sim* sim, const cmb_objectqueue* q);
+/* 1-based position of `object` in the queue; 0 = not present (reference
+ * cmb_objectqueue_position) */
+uint64_t cmb_objectqueue_position(const cmb_sim* sim,
+                                  const cmb_objectqueue* q,
+                                  const void* object);
 void cmb_objectqueue_recording_start(cmb_sim* sim, cmb_objectqueue* q);
 void cmb_objectqueue_recording_stop(cmb_sim* sim, cmb_objectqueue* q);
 /* time-weighted length stats while recording: mean/stddev/min/max */
@@ -396,6 +401,10 @@ double cmb_wtdsummary_variance(const cmb_wtdsummary* s);
 #define CMB_LOGGER_INFO (1u << 3)
 void cmb_logger_flags_on(uint32_t flags);
 void cmb_logger_flags_off(uint32_t flags);
+/* pluggable sim-time formatter (reference cmb_logger_timeformatter_set);
+ * fmt writes into buf (>= 32 bytes); NULL restores the default */
+typedef void (cmb_timeformatter_func)(double t, char* buf, size_t bufsz);
+void cmb_logger_timeformatter_set(cmb_timeformatter_func* fmt);
 void cmb_logger_info(cmb_sim* sim, const char* fmt, ...);
 void cmb_logger_warning(cmb_sim* sim, const char* fmt, ...);
 void cmb_logger_error(cmb_sim* sim, const char* fmt, ...); /* abandons trial */
