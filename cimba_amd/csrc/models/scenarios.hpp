@@ -47,6 +47,8 @@ struct Scenario : cmb::ModelBase {
         W_STALE_GRANT = 13,
         W_ABANDON = 14,
         W_POOL_PREEMPT = 15,
+        W_HEAP_OVERFLOW = 16,   // schedule past MAX_EV -> ST_HEAP_FULL
+        W_QUEUE_OVERFLOW = 17,  // put past physical QCAP -> ST_QUEUE_FULL
     };
 
     struct Params {
@@ -133,6 +135,8 @@ struct Scenario : cmb::ModelBase {
                             // error path; device: Engine::fail)
         F_POOL_PREEMPTOR,   // hold d, preempt a units of pool 0, hold b,
                             // release
+        F_OVERFLOWER,       // schedule a user events (heap abort path)
+        F_Q_FLOODER,        // put a objects without a consumer
     };
 
     template <class E_>
@@ -208,6 +212,20 @@ struct Scenario : cmb::ModelBase {
             if (CMB_SIG() != cmb::SIG_SUCCESS) trace(E, me, sigtag(E, self));
             CMB_POOL_RELEASE(0, E.pool_holding(0, me));
             trace(E, me, T_REL);
+            CMB_END();
+        }
+        case F_OVERFLOWER: {
+            CMB_BEGIN();
+            for (f.b = 0; f.b < f.a && E.status == cmb::ST_OK; ++f.b)
+                E.schedule(cmb::EV_USER, 0, 0, 0, 1e9 + (double)f.b, 0);
+            CMB_END();
+        }
+        case F_Q_FLOODER: {
+            CMB_BEGIN();
+            for (f.b = 0; f.b < f.a; ++f.b) {
+                CMB_QPUT(0, (uint64_t)f.b);
+                if (CMB_SIG() != cmb::SIG_SUCCESS) break;
+            }
             CMB_END();
         }
         case F_POOL_PREEMPTOR: {
@@ -452,6 +470,15 @@ struct Scenario : cmb::ModelBase {
         case W_ABANDON:
             sp(E, 0, F_HOLDER, 0, 0, 0, 1.5);   // a bystander process
             sp(E, 1, F_ABANDONER, 0, 0, 0, 0.0);
+            break;
+        case W_HEAP_OVERFLOW:
+            sp(E, 0, F_OVERFLOWER, 0, /*count*/ 100, 0, 0.0);
+            break;
+        case W_QUEUE_OVERFLOW:
+            // queue limit CMB_UNLIMITED but physical QCAP=16: the 17th
+            // put must abort the trial, not deadlock
+            E.queues[0].limit = cmb::CMB_UNLIMITED;
+            sp(E, 0, F_Q_FLOODER, 0, /*count*/ 64, 0, 0.0);
             break;
         case W_POOL_PREEMPT:
             // cap 4: p0 (pri 0) holds 3 for 10; p1 (pri 5) preempts 2 at

@@ -63,3 +63,18 @@ def test_scenario_gpu_matches_host(which):
         f"GPU scenario {which}: {g['trace']} != {EXPECTED[which]}")
     h = ca._C.scenario_host(which)
     assert g["events"] == h["events"]
+
+
+@pytest.mark.parametrize("which,status", [(16, 1), (17, 2)])
+def test_failure_paths_abort_cleanly(which, status):
+    """Capacity overflow = trial abort with a status code (SURVEY.md §5.3
+    per-block trial-abort flag), never a hang or corruption."""
+    r = ca._C.scenario_host(which)
+    assert r["status"] == status  # ST_HEAP_FULL=1 / ST_QUEUE_FULL=2
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("which,status", [(16, 1), (17, 2)])
+def test_failure_paths_abort_cleanly_gpu(which, status):
+    r = ca._C.scenario_gpu(which)
+    assert r["status"] == status
