@@ -286,8 +286,11 @@ uint64_t cmb_event_pattern_cancel(cmb_sim* s, cmb_event_func* action,
                     (!object || u.obj == object);
         }
         if (match) {
+            const uint32_t h = q.e[i].handle;
             uev_free_slot(*s->E, (int)q.e[i].b);
             q.remove_at(i);
+            if (s->E->n_event_waiters)
+                s->E->wake_event_waiters(h, SIG_CANCELLED);
             ++cnt;
         } else {
             ++i;
