@@ -23,15 +23,15 @@ using namespace cmb;
 
 namespace {
 
-constexpr int C_MAX_PROC = 256;
+constexpr int C_MAX_PROC = 4096;
 constexpr int C_UEV = 4096;  // pending user events
 constexpr int C_NAME = 32;  // reference CMB_PROCESS_NAMEBUF_SZ
 
 struct CModel : ModelBase {
     struct Cfg {
         static constexpr int MAX_PROC = C_MAX_PROC;
-        static constexpr int MAX_EV = 8192;
-        static constexpr int TIMERS = 2;
+        static constexpr int MAX_EV = 16384;
+        static constexpr int TIMERS = 4;
         static constexpr int NUM_QUEUES = 16;
         static constexpr int QCAP = 8192;
         static constexpr int NUM_RES = 16;
@@ -383,6 +383,24 @@ void cmb_timer_arm_(cmb_sim* s, cmb_process* p, double delay, int64_t sig) {
 }
 void cmb_timer_disarm_(cmb_sim* s, cmb_process* p) {
     s->E->timer_cancel(s->E->procs[dec(p)], 0);
+}
+bool cmb_process_timer_add(cmb_sim* s, cmb_process* p, int slot,
+                           double delay, int64_t sig) {
+    if (slot < 0 || slot >= CModel::Cfg::TIMERS) return false;
+    return s->E->timer_add(s->E->procs[dec(p)], slot, delay, sig);
+}
+void cmb_process_timer_cancel(cmb_sim* s, cmb_process* p, int slot) {
+    if (slot < 0 || slot >= CModel::Cfg::TIMERS) return;
+    s->E->timer_cancel(s->E->procs[dec(p)], slot);
+}
+void cmb_process_timer_clear(cmb_sim* s, cmb_process* p) {
+    for (int t = 0; t < CModel::Cfg::TIMERS; ++t)
+        s->E->timer_cancel(s->E->procs[dec(p)], t);
+}
+bool cmb_process_timer_pending(const cmb_sim* s, const cmb_process* p,
+                               int slot) {
+    if (slot < 0 || slot >= CModel::Cfg::TIMERS) return false;
+    return s->E->procs[dec(p)].timers[slot] != 0;
 }
 int cmb_sim_ok_(const cmb_sim* s) { return s->E->status == ST_OK; }
 

@@ -116,6 +116,19 @@ void cmb_wait_event_setup_(cmb_sim* sim, cmb_process* p, uint64_t handle);
 void cmb_proc_finish_(cmb_sim* sim, cmb_process* p);
 void cmb_timer_arm_(cmb_sim* sim, cmb_process* p, double delay, int64_t sig);
 void cmb_timer_disarm_(cmb_sim* sim, cmb_process* p);
+
+/* ---- multiple concurrent timers per process (reference
+ * cmb_process_timer_add/set/cancel/clear, cmb_process.c:514-580).
+ * Slot 0 is reserved for the blocking-call timeout (cmb_timer_arm_);
+ * user slots are 1..CMB_PROCESS_TIMERS-1.  A firing timer wakes the
+ * process's current blocking call with `sig`. ---- */
+#define CMB_PROCESS_TIMERS 4
+bool cmb_process_timer_add(cmb_sim* sim, cmb_process* p, int slot,
+                           double delay, int64_t sig);
+void cmb_process_timer_cancel(cmb_sim* sim, cmb_process* p, int slot);
+void cmb_process_timer_clear(cmb_sim* sim, cmb_process* p);
+bool cmb_process_timer_pending(const cmb_sim* sim, const cmb_process* p,
+                               int slot);
 int cmb_sim_ok_(const cmb_sim* sim);
 
 /* queue/resource try+wait plumbing */

@@ -64,6 +64,23 @@ def test_capi_mg1_resource_condition(libcimba, tmp_path):
     assert "M/G/1 avg system time" in out.stdout
 
 
+def test_capi_1000_processes_and_timers(libcimba, tmp_path):
+    # 1000 concurrent processes + multi-slot heartbeat timers (the
+    # reference's AWACS-scale process count on the host engine)
+    exe = str(tmp_path / "manyprocs")
+    r = subprocess.run(
+        ["gcc", "-std=c11", "-O2", "-Wall", "-Werror",
+         "-I", os.path.join(ROOT, "include"),
+         os.path.join(ROOT, "tutorial", "manyprocs_capi.c"),
+         "-L", os.path.join(ROOT, "cimba_amd"), "-lcimba",
+         f"-Wl,-rpath,{os.path.join(ROOT, 'cimba_amd')}",
+         "-lm", "-o", exe], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    out = subprocess.run([exe], capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, (out.stdout, out.stderr)
+    assert "1000-process run" in out.stdout
+
+
 def test_capi_header_is_c_clean(tmp_path):
     # header must compile as plain C99 without the library
     src = tmp_path / "hdr.c"
