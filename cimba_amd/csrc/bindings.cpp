@@ -794,14 +794,18 @@ PYBIND11_MODULE(_C, m) {
         .def("stddev", &WtdSummary::stddev)
         .def("minimum", [](const WtdSummary& s) { return s.mn; })
         .def("maximum", [](const WtdSummary& s) { return s.mx; })
+        .def("skewness", &WtdSummary::skewness)
+        .def("kurtosis", &WtdSummary::kurtosis)
         .def("raw", [](const WtdSummary& s) {
-            return py::make_tuple(s.n, s.sumw, s.mean, s.m2, s.mn, s.mx);
+            return py::make_tuple(s.n, s.sumw, s.mean, s.m2, s.m3, s.m4,
+                                  s.mn, s.mx);
         })
         .def_static("from_raw", [](double n, double sumw, double mean,
-                                   double m2, double mn, double mx) {
+                                   double m2, double m3, double m4,
+                                   double mn, double mx) {
             WtdSummary s;
-            s.n = n; s.sumw = sumw; s.mean = mean; s.m2 = m2;
-            s.mn = mn; s.mx = mx;
+            s.n = n; s.sumw = sumw; s.mean = mean; s.m2 = m2; s.m3 = m3;
+            s.m4 = m4; s.mn = mn; s.mx = mx;
             return s;
         });
 }

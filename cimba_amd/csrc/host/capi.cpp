@@ -7,12 +7,14 @@
 // cmb_condition.c, cmb_random.c, cmb_datasummary.c, cmb_logger.c).
 // engine headers first: cimba.h defines CMB_* macros (incl. CMB_UNLIMITED)
 // that would otherwise collide with the engine's identifiers
+#include "cimba/dataset.hpp"
 #include "cimba/engine.hpp"
 #include "cimba/logger.hpp"
 #include "cimba/runner.hpp"
 
 #include "../../../include/cimba.h"
 
+#include <algorithm>
 #include <atomic>
 #include <cstring>
 #include <ctime>
@@ -56,11 +58,16 @@ struct CModel : ModelBase {
         cmb_process_func* fn[C_MAX_PROC];
         void* ctx[C_MAX_PROC];
         char name[C_MAX_PROC][C_NAME];
+        void* exit_value[C_MAX_PROC];
         cmb_demand_func* dem_fn[C_MAX_PROC];
         void* dem_ctx[C_MAX_PROC];
         UEv uev[C_UEV];
         int32_t uev_free;  // freelist head
         int32_t nproc, nq, npq, nres, npool, nbuf, ncond;
+        char qname[Cfg::NUM_QUEUES][C_NAME];
+        char rname[Cfg::NUM_RES][C_NAME];
+        char plname[Cfg::NUM_POOLS][C_NAME];
+        char bname[Cfg::NUM_BUFS][C_NAME];
     };
 
     template <class E_>
@@ -154,6 +161,8 @@ void CModel::on_event(E_& E, const EvEntry& ev) {
 std::atomic<uint64_t> g_next{0};
 std::atomic<uint64_t> g_total{0};
 int g_default_threads = 0;
+thread_local void* g_thread_ctx = nullptr;
+thread_local int g_thread_id = 0;
 void (*g_thread_init)(int) = nullptr;
 void (*g_thread_exit)(int) = nullptr;
 void (*g_trial_cleanup)(uint64_t) = nullptr;
@@ -177,6 +186,7 @@ uint64_t cimba_run(void* experiment, uint64_t n, size_t size,
     std::atomic<uint64_t> failed{0};
 
     auto worker = [&](int widx) {
+        g_thread_id = widx;
         if (g_thread_init) g_thread_init(widx);
         auto store = std::make_unique<CStorage>();
         auto eng = std::make_unique<CEngine>(*store);
@@ -215,6 +225,18 @@ uint64_t cimba_run(void* experiment, uint64_t n, size_t size,
 }
 
 void cimba_threads_use(int nthreads) { g_default_threads = nthreads; }
+uint64_t cimba_run_experiment(void* experiment, uint64_t n, size_t size,
+                              cimba_trial_func* trial_fn,
+                              uint64_t master_seed, int nthreads) {
+    return cimba_run(experiment, n, size, trial_fn, master_seed, nthreads);
+}
+uint32_t cimba_trial_index(const cmb_sim* s) { return s->E->trial_index; }
+uint64_t cimba_trials_total(void) { return g_total.load(); }
+int cimba_threads_num(void) { return g_default_threads; }
+void cimba_thread_context_set(void* ctx) { g_thread_ctx = ctx; }
+void* cimba_thread_context(void) { return g_thread_ctx; }
+int cimba_thread_id(void) { return g_thread_id; }
+const char* cimba_version(void) { return "cimba-mi355x 0.1.0"; }
 uint64_t cimba_trials_remaining(void) {
     const uint64_t next = g_next.load(), total = g_total.load();
     return next >= total ? 0 : total - next;
@@ -302,6 +324,53 @@ uint64_t cmb_event_pattern_cancel(cmb_sim* s, cmb_event_func* action,
 void cmb_event_queue_execute(cmb_sim* s) {
     s->E->run(1.0e308, UINT64_C(0xFFFFFFFFFFFFFFFF));
 }
+bool cmb_event_execute_next(cmb_sim* s) { return s->E->dispatch_one(); }
+uint64_t cmb_event_queue_count(const cmb_sim* s) {
+    return (uint64_t)s->E->evq.n;
+}
+bool cmb_event_queue_is_empty(const cmb_sim* s) { return s->E->evq.empty(); }
+void cmb_event_queue_clear(cmb_sim* s) {
+    auto& q = s->E->evq;
+    for (int32_t i = 0; i < q.n; ++i)
+        if (q.e[i].kind == EV_USER) uev_free_slot(*s->E, (int)q.e[i].b);
+    q.reset();
+}
+static int32_t find_slot_(const cmb_sim* s, uint64_t handle) {
+    const auto& q = s->E->evq;
+    for (int32_t i = 0; i < q.n; ++i)
+        if (q.e[i].handle == (uint32_t)handle) return i;
+    return -1;
+}
+bool cmb_event_is_scheduled(const cmb_sim* s, uint64_t handle) {
+    return find_slot_(s, handle) >= 0;
+}
+double cmb_event_time(const cmb_sim* s, uint64_t handle) {
+    const int32_t i = find_slot_(s, handle);
+    return i >= 0 ? s->E->evq.e[i].t : -1.0;
+}
+int cmb_event_priority(const cmb_sim* s, uint64_t handle) {
+    const int32_t i = find_slot_(s, handle);
+    if (i < 0) return 0;
+    return 32767 - (int)(s->E->evq.e[i].pseq >> 48);
+}
+bool cmb_event_reprioritize(cmb_sim* s, uint64_t handle, int priority) {
+    const int32_t i = find_slot_(s, handle);
+    if (i < 0) return false;
+    const double t = s->E->evq.e[i].t;
+    return s->E->event_reschedule((uint32_t)handle, t, priority);
+}
+uint64_t cmb_event_pattern_find(cmb_sim* s, cmb_event_func* action,
+                                void* subject, void* object) {
+    const auto& q = s->E->evq;
+    for (int32_t i = 0; i < q.n; ++i) {
+        if (q.e[i].kind != EV_USER) continue;
+        const auto& u = s->E->globals.uev[(int)q.e[i].b];
+        if ((!action || u.fn == action) && (!subject || u.subj == subject) &&
+            (!object || u.obj == object))
+            return q.e[i].handle;
+    }
+    return 0;
+}
 void cmb_event_queue_execute_until(cmb_sim* s, double until) {
     s->E->run(until, UINT64_C(0xFFFFFFFFFFFFFFFF));
 }
@@ -355,6 +424,49 @@ int64_t cmb_process_signal(const cmb_sim* s, const cmb_process* p) {
 }
 void* cmb_process_context(const cmb_sim* s, const cmb_process* p) {
     return s->E->globals.ctx[dec(p)];
+}
+void cmb_process_kill(cmb_sim* s, cmb_process* p) { s->E->proc_stop(dec(p)); }
+int cmb_process_status(const cmb_sim* s, const cmb_process* p) {
+    return (int)s->E->procs[dec(p)].state;
+}
+void cmb_process_name_set(cmb_sim* s, cmb_process* p, const char* name) {
+    std::strncpy(s->E->globals.name[dec(p)], name ? name : "", C_NAME - 1);
+    s->E->globals.name[dec(p)][C_NAME - 1] = 0;
+}
+cmb_process* cmb_process_create(cmb_sim* s) {
+    CEngine& E = *s->E;
+    const int idx = E.proc_alloc();
+    if (idx < 0) {
+        E.fail(ST_BAD_STATE);
+        return nullptr;
+    }
+    E.globals.fn[idx] = nullptr;
+    E.globals.ctx[idx] = nullptr;
+    E.globals.exit_value[idx] = nullptr;
+    return enc<cmb_process>(idx);
+}
+void cmb_process_initialize(cmb_sim* s, cmb_process* p, const char* name,
+                            cmb_process_func* fn, void* ctx, int priority) {
+    CEngine& E = *s->E;
+    const int idx = dec(p);
+    E.proc_init(idx, 0, priority);
+    E.globals.fn[idx] = fn;
+    E.globals.ctx[idx] = ctx;
+    cmb_process_name_set(s, p, name);
+}
+void cmb_process_terminate(cmb_sim*, cmb_process*) {}
+void cmb_process_destroy(cmb_sim*, cmb_process*) {}
+void cmb_process_exit_value_set_(cmb_sim* s, cmb_process* p, void* value) {
+    s->E->globals.exit_value[dec(p)] = value;
+}
+void* cmb_process_exit_value(const cmb_sim* s, const cmb_process* p) {
+    return s->E->globals.exit_value[dec(p)];
+}
+bool cmb_process_timer_set(cmb_sim* s, cmb_process* p, int slot,
+                           double delay, int64_t sig) {
+    if (slot < 0 || slot >= CModel::Cfg::TIMERS) return false;
+    s->E->timer_cancel(s->E->procs[dec(p)], slot);
+    return s->E->timer_add(s->E->procs[dec(p)], slot, delay, sig);
 }
 
 int cmb_proc_pc_(const cmb_sim* s, const cmb_process* p) {
@@ -415,9 +527,20 @@ cmb_objectqueue* cmb_objectqueue_create(cmb_sim* s) {
     return enc<cmb_objectqueue>(E.globals.nq++);
 }
 void cmb_objectqueue_initialize(cmb_sim* s, cmb_objectqueue* q,
-                                const char* /*name*/, int32_t capacity) {
+                                const char* name, int32_t capacity) {
     s->E->queues[dec(q)].limit =
         capacity == CMB_UNLIMITED ? CMB_UNLIMITED : capacity;
+    std::strncpy(s->E->globals.qname[dec(q)], name ? name : "", C_NAME - 1);
+    s->E->globals.qname[dec(q)][C_NAME - 1] = 0;
+}
+const char* cmb_objectqueue_name(const cmb_sim* s, const cmb_objectqueue* q) {
+    return s->E->globals.qname[dec(q)];
+}
+uint64_t cmb_objectqueue_space(const cmb_sim* s, const cmb_objectqueue* q) {
+    const auto& Q = s->E->queues[dec(q)];
+    const int32_t lim =
+        Q.limit == CMB_UNLIMITED ? CModel::Cfg::QCAP : Q.limit;
+    return (uint64_t)(lim - Q.len);
 }
 uint64_t cmb_objectqueue_length(const cmb_sim* s, const cmb_objectqueue* q) {
     return (uint64_t)s->E->queues[dec(q)].len;
@@ -489,6 +612,95 @@ uint64_t cmb_priorityqueue_length(const cmb_sim* s,
                                   const cmb_priorityqueue* q) {
     return (uint64_t)s->E->pqueues[dec(q)].len;
 }
+uint64_t cmb_priorityqueue_space(const cmb_sim* s,
+                                 const cmb_priorityqueue* q) {
+    const auto& Q = s->E->pqueues[dec(q)];
+    return (uint64_t)(Q.limit - Q.len);
+}
+uint64_t cmb_priorityqueue_position(const cmb_sim* s,
+                                    const cmb_priorityqueue* q,
+                                    const void* object) {
+    const auto& Q = s->E->pqueues[dec(q)];
+    for (int32_t i = 0; i < Q.len; ++i) {
+        if (Q.val[i] == (uint64_t)object) {
+            uint64_t pos = 1;
+            for (int32_t j = 0; j < Q.len; ++j)
+                if (Q.key[j] < Q.key[i]) ++pos;
+            return pos;
+        }
+    }
+    return 0;
+}
+static void pq_remove_at_(cmb_sim* s, int qi, int32_t i) {
+    auto& Q = s->E->pqueues[qi];
+    --Q.len;
+    Q.key[i] = Q.key[Q.len];
+    Q.val[i] = Q.val[Q.len];
+    for (int32_t r = Q.len / 2 - 1; r >= 0; --r) {  // full re-heapify
+        int32_t n0 = r;
+        for (;;) {
+            int32_t cle = 2 * n0 + 1;
+            if (cle >= Q.len) break;
+            if (cle + 1 < Q.len && Q.key[cle + 1] < Q.key[cle]) ++cle;
+            if (Q.key[n0] <= Q.key[cle]) break;
+            std::swap(Q.key[n0], Q.key[cle]);
+            std::swap(Q.val[n0], Q.val[cle]);
+            n0 = cle;
+        }
+    }
+}
+bool cmb_priorityqueue_cancel(cmb_sim* s, cmb_priorityqueue* q,
+                              const void* object) {
+    auto& Q = s->E->pqueues[dec(q)];
+    for (int32_t i = 0; i < Q.len; ++i) {
+        if (Q.val[i] == (uint64_t)object) {
+            pq_remove_at_(s, dec(q), i);
+            s->E->guard_signal(Q.g_rear);  // space freed
+            return true;
+        }
+    }
+    return false;
+}
+bool cmb_priorityqueue_reprioritize(cmb_sim* s, cmb_priorityqueue* q,
+                                    const void* object, int priority) {
+    auto& Q = s->E->pqueues[dec(q)];
+    for (int32_t i = 0; i < Q.len; ++i) {
+        if (Q.val[i] == (uint64_t)object) {
+            pq_remove_at_(s, dec(q), i);
+            const uint64_t key =
+                ((uint64_t)(uint16_t)(32767 - priority) << 32) | Q.seq++;
+            int32_t j = Q.len++;
+            while (j > 0) {
+                const int32_t par = (j - 1) / 2;
+                if (Q.key[par] <= key) break;
+                Q.key[j] = Q.key[par];
+                Q.val[j] = Q.val[par];
+                j = par;
+            }
+            Q.key[j] = key;
+            Q.val[j] = (uint64_t)object;
+            return true;
+        }
+    }
+    return false;
+}
+void cmb_priorityqueue_recording_start(cmb_sim* s, cmb_priorityqueue* q) {
+    auto& Q = s->E->pqueues[dec(q)];
+    Q.recording = 1;
+    Q.len_stats.reset();
+    Q.t_last = s->E->now;
+}
+void cmb_priorityqueue_recording_stop(cmb_sim* s, cmb_priorityqueue* q) {
+    s->E->pqueues[dec(q)].recording = 0;
+}
+void cmb_priorityqueue_report_print(cmb_sim* s, const cmb_priorityqueue* q,
+                                    FILE* out) {
+    if (!out) out = stderr;
+    auto Q = s->E->pqueues[dec(q)];  // copy
+    Q.len_stats.add((double)Q.len, s->E->now - Q.t_last);
+    fprintf(out, "pqueue report @ t=%.6f: length=%d mean=%.4f sd=%.4f\n",
+            s->E->now, Q.len, Q.len_stats.mean, Q.len_stats.stddev());
+}
 bool cmb_pqueue_try_put_(cmb_sim* s, cmb_priorityqueue* q, cmb_process* p,
                          void* object, int priority) {
     return s->E->pq_try_put(dec(q), s->E->procs[dec(p)], (uint64_t)object,
@@ -522,7 +734,13 @@ cmb_resource* cmb_resource_create(cmb_sim* s) {
     }
     return enc<cmb_resource>(E.globals.nres++);
 }
-void cmb_resource_initialize(cmb_sim*, cmb_resource*, const char*) {}
+void cmb_resource_initialize(cmb_sim* s, cmb_resource* r, const char* name) {
+    std::strncpy(s->E->globals.rname[dec(r)], name ? name : "", C_NAME - 1);
+    s->E->globals.rname[dec(r)][C_NAME - 1] = 0;
+}
+const char* cmb_resource_name(const cmb_sim* s, const cmb_resource* r) {
+    return s->E->globals.rname[dec(r)];
+}
 void cmb_resource_release(cmb_sim* s, cmb_resource* r, cmb_process*) {
     s->E->resource_release(dec(r));
 }
@@ -532,6 +750,22 @@ bool cmb_resource_in_use(const cmb_sim* s, const cmb_resource* r) {
 cmb_process* cmb_resource_holder(const cmb_sim* s, const cmb_resource* r) {
     const int h = s->E->resources[dec(r)].holder;
     return h < 0 ? nullptr : enc<cmb_process>(h);
+}
+void cmb_resource_recording_stop(cmb_sim* s, cmb_resource* r) {
+    s->E->resources[dec(r)].recording = 0;
+}
+void cmb_resource_start_recording(cmb_sim* s, cmb_resource* r) {
+    cmb_resource_recording_start(s, r);
+}
+void cmb_resource_stop_recording(cmb_sim* s, cmb_resource* r) {
+    cmb_resource_recording_stop(s, r);
+}
+bool cmb_resource_available(const cmb_sim* s, const cmb_resource* r) {
+    return s->E->resources[dec(r)].holder < 0;
+}
+bool cmb_resource_held_by_process(const cmb_sim* s, const cmb_resource* r,
+                                  const cmb_process* p) {
+    return s->E->resources[dec(r)].holder == (int16_t)dec(p);
 }
 void cmb_resource_recording_start(cmb_sim* s, cmb_resource* r) {
     auto& R = s->E->resources[dec(r)];
@@ -568,8 +802,14 @@ cmb_resourcepool* cmb_resourcepool_create(cmb_sim* s) {
     return enc<cmb_resourcepool>(E.globals.npool++);
 }
 void cmb_resourcepool_initialize(cmb_sim* s, cmb_resourcepool* r,
-                                 const char* /*name*/, int32_t capacity) {
+                                 const char* name, int32_t capacity) {
     s->E->pools[dec(r)].capacity = capacity;
+    std::strncpy(s->E->globals.plname[dec(r)], name ? name : "", C_NAME - 1);
+    s->E->globals.plname[dec(r)][C_NAME - 1] = 0;
+}
+const char* cmb_resourcepool_get_name(const cmb_sim* s,
+                                      const cmb_resourcepool* r) {
+    return s->E->globals.plname[dec(r)];
 }
 void cmb_resourcepool_release(cmb_sim* s, cmb_resourcepool* r,
                               cmb_process* holder, int32_t amount) {
@@ -593,6 +833,32 @@ int32_t cmb_resourcepool_in_use(const cmb_sim* s, const cmb_resourcepool* r) {
 int32_t cmb_resourcepool_available(const cmb_sim* s,
                                    const cmb_resourcepool* r) {
     return s->E->pools[dec(r)].capacity - s->E->pools[dec(r)].in_use;
+}
+int32_t cmb_resourcepool_held(const cmb_sim* s, const cmb_resourcepool* r) {
+    return s->E->pools[dec(r)].in_use;
+}
+int32_t cmb_resourcepool_held_by_process(const cmb_sim* s,
+                                         const cmb_resourcepool* r,
+                                         const cmb_process* p) {
+    return s->E->pool_holding(dec(r), dec(p));
+}
+void cmb_resourcepool_start_recording(cmb_sim* s, cmb_resourcepool* r) {
+    auto& P = s->E->pools[dec(r)];
+    P.recording = 1;
+    P.use_stats.reset();
+    P.t_last = s->E->now;
+}
+void cmb_resourcepool_stop_recording(cmb_sim* s, cmb_resourcepool* r) {
+    s->E->pools[dec(r)].recording = 0;
+}
+void cmb_resourcepool_stats(cmb_sim* s, const cmb_resourcepool* r,
+                            double out4[4]) {
+    auto P = s->E->pools[dec(r)];  // copy
+    P.use_stats.add((double)P.in_use, s->E->now - P.t_last);
+    out4[0] = P.use_stats.mean;
+    out4[1] = P.use_stats.stddev();
+    out4[2] = P.use_stats.mn;
+    out4[3] = P.use_stats.mx;
 }
 int32_t cmb_pool_try_take_(cmb_sim* s, cmb_resourcepool* r, cmb_process* p,
                            int32_t want) {
@@ -622,8 +888,10 @@ cmb_buffer* cmb_buffer_create(cmb_sim* s) {
     }
     return enc<cmb_buffer>(E.globals.nbuf++);
 }
-void cmb_buffer_initialize(cmb_sim* s, cmb_buffer* b, const char* /*name*/,
+void cmb_buffer_initialize(cmb_sim* s, cmb_buffer* b, const char* name,
                            int64_t capacity, int64_t initial_level) {
+    std::strncpy(s->E->globals.bname[dec(b)], name ? name : "", C_NAME - 1);
+    s->E->globals.bname[dec(b)][C_NAME - 1] = 0;
     auto& B = s->E->buffers[dec(b)];
     B.capacity = capacity == CMB_UNLIMITED ? INT64_MAX / 2 : capacity;
     B.level = initial_level;
@@ -633,6 +901,38 @@ int64_t cmb_buffer_level(const cmb_sim* s, const cmb_buffer* b) {
 }
 int64_t cmb_buffer_capacity(const cmb_sim* s, const cmb_buffer* b) {
     return s->E->buffers[dec(b)].capacity;
+}
+const char* cmb_buffer_name(const cmb_sim* s, const cmb_buffer* b) {
+    return s->E->globals.bname[dec(b)];
+}
+int64_t cmb_buffer_space(const cmb_sim* s, const cmb_buffer* b) {
+    const auto& B = s->E->buffers[dec(b)];
+    return B.capacity - B.level;
+}
+void cmb_buffer_recording_start(cmb_sim* s, cmb_buffer* b) {
+    auto& B = s->E->buffers[dec(b)];
+    B.recording = 1;
+    B.level_stats.reset();
+    B.t_last = s->E->now;
+}
+void cmb_buffer_recording_stop(cmb_sim* s, cmb_buffer* b) {
+    s->E->buffers[dec(b)].recording = 0;
+}
+void cmb_buffer_stats(cmb_sim* s, const cmb_buffer* b, double out4[4]) {
+    auto B = s->E->buffers[dec(b)];  // copy
+    B.level_stats.add((double)B.level, s->E->now - B.t_last);
+    out4[0] = B.level_stats.mean;
+    out4[1] = B.level_stats.stddev();
+    out4[2] = B.level_stats.mn;
+    out4[3] = B.level_stats.mx;
+}
+void cmb_buffer_print_report(cmb_sim* s, const cmb_buffer* b, FILE* out) {
+    if (!out) out = stderr;
+    double st[4];
+    cmb_buffer_stats(s, b, st);
+    fprintf(out,
+            "buffer report @ t=%.6f: level=%lld mean=%.4f sd=%.4f\n",
+            s->E->now, (long long)s->E->buffers[dec(b)].level, st[0], st[1]);
 }
 bool cmb_buffer_try_get_(cmb_sim* s, cmb_buffer* b, cmb_process* p,
                          int64_t amount) {
@@ -674,6 +974,51 @@ void cmb_condition_wait_setup_(cmb_sim* s, cmb_condition* c, cmb_process* p,
     E.globals.dem_fn[pidx] = demand;
     E.globals.dem_ctx[pidx] = ctx;
     E.guard_wait(E.procs[pidx], E.conds[dec(c)].gid, DEM_USER, 0);
+}
+bool cmb_condition_cancel(cmb_sim* s, cmb_condition* c, cmb_process* p) {
+    CEngine& E = *s->E;
+    auto& pr = E.procs[dec(p)];
+    if (pr.await_kind != AW_GUARD || pr.gid != E.conds[dec(c)].gid)
+        return false;
+    E.proc_interrupt(dec(p), SIG_CANCELLED);
+    return true;
+}
+void cmb_condition_subscribe_resource(cmb_sim* s, cmb_condition* c,
+                                      cmb_resource* r) {
+    s->E->condition_observe(dec(c), s->E->resources[dec(r)].gid);
+}
+void cmb_condition_subscribe_queue(cmb_sim* s, cmb_condition* c,
+                                   cmb_objectqueue* q) {
+    s->E->condition_observe(dec(c), s->E->queues[dec(q)].g_front);
+}
+void cmb_condition_subscribe_pool(cmb_sim* s, cmb_condition* c,
+                                  cmb_resourcepool* r) {
+    s->E->condition_observe(dec(c), s->E->pools[dec(r)].gid);
+}
+void cmb_condition_unsubscribe_all(cmb_sim* s, cmb_condition* c) {
+    CEngine& E = *s->E;
+    for (int g = 0; g < CEngine::NGUARD; ++g)
+        if (E.guards[g].observer == (int16_t)dec(c)) E.guards[g].observer = -1;
+}
+
+/* resource guard = a condition whose signal() evaluates the front waiter
+ * only (the reference's guard contract) */
+cmb_resourceguard* cmb_resourceguard_create(cmb_sim* s) {
+    return (cmb_resourceguard*)cmb_condition_create(s);
+}
+void cmb_resourceguard_initialize(cmb_sim*, cmb_resourceguard*,
+                                  const char*) {}
+bool cmb_resourceguard_signal(cmb_sim* s, cmb_resourceguard* g) {
+    return s->E->guard_signal(s->E->conds[dec(g)].gid);
+}
+bool cmb_resourceguard_cancel(cmb_sim* s, cmb_resourceguard* g,
+                              cmb_process* p) {
+    return cmb_condition_cancel(s, (cmb_condition*)g, p);
+}
+void cmb_guard_wait_setup_(cmb_sim* s, cmb_resourceguard* g, cmb_process* p,
+                           cmb_resourceguard_demand_func* fn, void* ctx) {
+    cmb_condition_wait_setup_(s, (cmb_condition*)g, p,
+                              (cmb_demand_func*)fn, ctx);
 }
 
 /* ---- debug dumps & reports ---- */
@@ -823,6 +1168,34 @@ int64_t cmb_random_discrete_nonuniform(cmb_sim* s, const double* w,
 int64_t cmb_random_loaded_dice(cmb_sim* s, const double* w, int64_t n) {
     return s->E->rng.loaded_dice(w, n);
 }
+uint64_t cmb_random_splitmix64(uint64_t* state) { return splitmix64(*state); }
+void cmb_random_initialize(cmb_sim* s, uint64_t seed) {
+    s->E->rng.seed(seed);
+    s->seed = seed;
+}
+struct cmb_alias_impl {
+    std::vector<double> prob;
+    std::vector<int32_t> alias;
+    int64_t n;
+};
+cmb_alias* cmb_random_alias_create(const double* weights, int64_t n) {
+    auto* a = new cmb_alias_impl;
+    a->n = n;
+    a->prob.resize((size_t)n);
+    a->alias.resize((size_t)n);
+    std::vector<int32_t> scratch((size_t)(2 * n));
+    alias_build(weights, n, a->prob.data(), a->alias.data(), scratch.data());
+    return (cmb_alias*)a;
+}
+void cmb_random_alias_destroy(cmb_alias* a) { delete (cmb_alias_impl*)a; }
+int64_t cmb_random_alias_sample(cmb_sim* s, const cmb_alias* va) {
+    const auto* a = (const cmb_alias_impl*)va;
+    AliasTable t{a->prob.data(), a->alias.data(), a->n};
+    return t.sample(s->E->rng);
+}
+int64_t cmb_random_alias_draw(cmb_sim* s, const cmb_alias* a) {
+    return cmb_random_alias_sample(s, a);
+}
 uint64_t cmb_random_hwseed(void) {
     // host entropy (reference RDSEED asm; /dev/urandom is the portable
     // equivalent — see docs/PARITY.md)
@@ -836,8 +1209,185 @@ uint64_t cmb_random_hwseed(void) {
     return x;
 }
 
+/* ---- dataset & timeseries (host) ---- */
+
+cmb_dataset* cmb_dataset_create(void) { return (cmb_dataset*)new Dataset; }
+void cmb_dataset_destroy(cmb_dataset* d) { delete (Dataset*)d; }
+void cmb_dataset_reset(cmb_dataset* d) { ((Dataset*)d)->clear(); }
+void cmb_dataset_add(cmb_dataset* d, double x) { ((Dataset*)d)->add(x); }
+uint64_t cmb_dataset_count(const cmb_dataset* d) {
+    return (uint64_t)((const Dataset*)d)->size();
+}
+double cmb_dataset_min(cmb_dataset* d) { return ((Dataset*)d)->quantile(0.0); }
+double cmb_dataset_max(cmb_dataset* d) { return ((Dataset*)d)->quantile(1.0); }
+double cmb_dataset_median(cmb_dataset* d) { return ((Dataset*)d)->median(); }
+double cmb_dataset_quantile(cmb_dataset* d, double q) {
+    return ((Dataset*)d)->quantile(q);
+}
+void cmb_dataset_sort(cmb_dataset* d) { ((Dataset*)d)->sort(); }
+void cmb_dataset_merge(cmb_dataset* d, const cmb_dataset* o) {
+    ((Dataset*)d)->merge(*(const Dataset*)o);
+}
+void cmb_dataset_copy(cmb_dataset* dst, const cmb_dataset* src) {
+    *(Dataset*)dst = *(const Dataset*)src;
+}
+struct cmb_datasummary cmb_dataset_summarize(const cmb_dataset* d) {
+    const DataSummary s = ((const Dataset*)d)->summarize();
+    cmb_datasummary out;
+    std::memcpy(&out, &s, sizeof(out));
+    return out;
+}
+void cmb_dataset_acf(const cmb_dataset* d, double* out, int maxlag) {
+    auto v = ((const Dataset*)d)->acf(maxlag);
+    for (int i = 0; i < maxlag; ++i) out[i] = v[(size_t)i];
+}
+void cmb_dataset_pacf(const cmb_dataset* d, double* out, int maxlag) {
+    auto v = ((const Dataset*)d)->pacf(maxlag);
+    for (int i = 0; i < maxlag; ++i) out[i] = v[(size_t)i];
+}
+void cmb_dataset_histogram(cmb_dataset* d, int nbins, int64_t* out_counts) {
+    auto hcounts = ((Dataset*)d)->histogram(nbins);
+    for (int i = 0; i < nbins; ++i) out_counts[i] = hcounts[(size_t)i];
+}
+void cmb_dataset_fivenum_print(cmb_dataset* d, FILE* out) {
+    if (!out) out = stdout;
+    double f[5];
+    ((Dataset*)d)->fivenum(f);
+    fprintf(out, "min %.6g  Q1 %.6g  median %.6g  Q3 %.6g  max %.6g\n",
+            f[0], f[1], f[2], f[3], f[4]);
+}
+void cmb_dataset_histogram_print(cmb_dataset* d, int nbins, FILE* out) {
+    if (!out) out = stdout;
+    auto h = ((Dataset*)d)->histogram(nbins);
+    int64_t peak = 1;
+    for (auto c : h) peak = c > peak ? c : peak;
+    for (int i = 0; i < nbins; ++i) {
+        const int bars = (int)((h[(size_t)i] * 50) / peak);
+        fprintf(out, "%3d | %-50.*s %lld\n", i, bars,
+                "##################################################",
+                (long long)h[(size_t)i]);
+    }
+}
+void cmb_dataset_correlogram_print(const cmb_dataset* d, int maxlag,
+                                   FILE* out) {
+    if (!out) out = stdout;
+    auto a = ((const Dataset*)d)->acf(maxlag);
+    auto p = ((const Dataset*)d)->pacf(maxlag);
+    const double ci = 1.96 / sqrt((double)((const Dataset*)d)->size());
+    fprintf(out, "lag      acf     pacf   (95%% CI +-%.4f)\n", ci);
+    for (int k = 0; k < maxlag; ++k)
+        fprintf(out, "%3d  %7.4f  %7.4f\n", k + 1, a[(size_t)k],
+                p[(size_t)k]);
+}
+void cmb_dataset_print(const cmb_dataset* d, FILE* out) {
+    if (!out) out = stdout;
+    const auto& v = ((const Dataset*)d)->values();
+    for (double x : v) fprintf(out, "%.9g\n", x);
+}
+
+struct CTimeseries {
+    Timeseries ts;
+    double end = -1.0;
+};
+static double ts_end_(const CTimeseries* t, double end_time) {
+    if (end_time >= 0.0) return end_time;
+    if (t->end >= 0.0) return t->end;
+    return t->ts.times().empty() ? 0.0 : t->ts.times().back();
+}
+cmb_timeseries* cmb_timeseries_create(void) {
+    return (cmb_timeseries*)new CTimeseries;
+}
+void cmb_timeseries_destroy(cmb_timeseries* t) { delete (CTimeseries*)t; }
+void cmb_timeseries_reset(cmb_timeseries* t) {
+    *(CTimeseries*)t = CTimeseries();
+}
+void cmb_timeseries_add(cmb_timeseries* t, double x, double time) {
+    ((CTimeseries*)t)->ts.add(x, time);
+}
+uint64_t cmb_timeseries_count(const cmb_timeseries* t) {
+    return (uint64_t)((const CTimeseries*)t)->ts.size();
+}
+void cmb_timeseries_finalize(cmb_timeseries* t, double end_time) {
+    ((CTimeseries*)t)->end = end_time;
+}
+void cmb_timeseries_copy(cmb_timeseries* dst, const cmb_timeseries* src) {
+    *(CTimeseries*)dst = *(const CTimeseries*)src;
+}
+struct cmb_wtdsummary cmb_timeseries_summarize(const cmb_timeseries* t,
+                                               double end_time) {
+    const auto* ct = (const CTimeseries*)t;
+    const WtdSummary s = ct->ts.summarize(ts_end_(ct, end_time));
+    cmb_wtdsummary out;
+    std::memcpy(&out, &s, sizeof(out));
+    return out;
+}
+double cmb_timeseries_median(const cmb_timeseries* t, double end_time) {
+    const auto* ct = (const CTimeseries*)t;
+    return ct->ts.median(ts_end_(ct, end_time));
+}
+static Dataset ts_values_(const CTimeseries* t) {
+    Dataset d;
+    for (double x : t->ts.values()) d.add(x);
+    return d;
+}
+double cmb_timeseries_min(const cmb_timeseries* t) {
+    Dataset d = ts_values_((const CTimeseries*)t);
+    return d.quantile(0.0);
+}
+double cmb_timeseries_max(const cmb_timeseries* t) {
+    Dataset d = ts_values_((const CTimeseries*)t);
+    return d.quantile(1.0);
+}
+void cmb_timeseries_fivenum_print(const cmb_timeseries* t, FILE* out) {
+    Dataset d = ts_values_((const CTimeseries*)t);
+    cmb_dataset_fivenum_print((cmb_dataset*)&d, out);
+}
+void cmb_timeseries_histogram_print(const cmb_timeseries* t, int nbins,
+                                    FILE* out) {
+    Dataset d = ts_values_((const CTimeseries*)t);
+    cmb_dataset_histogram_print((cmb_dataset*)&d, nbins, out);
+}
+void cmb_timeseries_sort_x(cmb_timeseries* t) {
+    auto* ct = (CTimeseries*)t;
+    std::vector<std::pair<double, double>> v;
+    for (size_t i = 0; i < ct->ts.size(); ++i)
+        v.emplace_back(ct->ts.values()[i], ct->ts.times()[i]);
+    std::sort(v.begin(), v.end());
+    Timeseries fresh;
+    for (auto& pr : v) fresh.add(pr.first, pr.second);
+    ct->ts = fresh;
+}
+void cmb_timeseries_sort_t(cmb_timeseries* t) {
+    auto* ct = (CTimeseries*)t;
+    std::vector<std::pair<double, double>> v;
+    for (size_t i = 0; i < ct->ts.size(); ++i)
+        v.emplace_back(ct->ts.times()[i], ct->ts.values()[i]);
+    std::sort(v.begin(), v.end());
+    Timeseries fresh;
+    for (auto& pr : v) fresh.add(pr.second, pr.first);
+    ct->ts = fresh;
+}
+void cmb_timeseries_print(const cmb_timeseries* t, FILE* out) {
+    if (!out) out = stdout;
+    const auto* ts = &((const CTimeseries*)t)->ts;
+    for (size_t i = 0; i < ts->size(); ++i)
+        fprintf(out, "%.9g %.9g\n", ts->times()[i], ts->values()[i]);
+}
+
 /* ---- summaries ---- */
 
+cmb_datasummary* cmb_datasummary_create(void) {
+    auto* s = new cmb_datasummary;
+    reinterpret_cast<DataSummary*>(s)->reset();
+    return s;
+}
+void cmb_datasummary_destroy(cmb_datasummary* s) { delete s; }
+cmb_wtdsummary* cmb_wtdsummary_create(void) {
+    auto* s = new cmb_wtdsummary;
+    reinterpret_cast<WtdSummary*>(s)->reset();
+    return s;
+}
+void cmb_wtdsummary_destroy(cmb_wtdsummary* s) { delete s; }
 void cmb_datasummary_initialize(cmb_datasummary* s) {
     reinterpret_cast<DataSummary*>(s)->reset();
 }
@@ -879,6 +1429,34 @@ double cmb_wtdsummary_mean(const cmb_wtdsummary* s) { return s->mean; }
 double cmb_wtdsummary_variance(const cmb_wtdsummary* s) {
     return reinterpret_cast<const WtdSummary*>(s)->variance();
 }
+double cmb_wtdsummary_stddev(const cmb_wtdsummary* s) {
+    return reinterpret_cast<const WtdSummary*>(s)->stddev();
+}
+double cmb_wtdsummary_skewness(const cmb_wtdsummary* s) {
+    return reinterpret_cast<const WtdSummary*>(s)->skewness();
+}
+double cmb_wtdsummary_kurtosis(const cmb_wtdsummary* s) {
+    return reinterpret_cast<const WtdSummary*>(s)->kurtosis();
+}
+double cmb_wtdsummary_count(const cmb_wtdsummary* s) { return s->n; }
+double cmb_wtdsummary_min(const cmb_wtdsummary* s) { return s->mn; }
+double cmb_wtdsummary_max(const cmb_wtdsummary* s) { return s->mx; }
+void cmb_wtdsummary_print(const cmb_wtdsummary* s, FILE* out) {
+    if (!out) out = stdout;
+    fprintf(out,
+            "wtd summary: n=%.0f w=%.6g mean=%.6g sd=%.6g min=%.6g "
+            "max=%.6g\n",
+            s->n, s->sumw, s->mean, cmb_wtdsummary_stddev(s), s->mn, s->mx);
+}
+void cmb_datasummary_print(const struct cmb_datasummary* s, FILE* out) {
+    if (!out) out = stdout;
+    fprintf(out,
+            "summary: n=%.0f mean=%.6g sd=%.6g skew=%.4f kurt=%.4f "
+            "min=%.6g max=%.6g\n",
+            s->n, s->mean, cmb_datasummary_stddev(s),
+            cmb_datasummary_skewness(s), cmb_datasummary_kurtosis(s), s->mn,
+            s->mx);
+}
 
 /* ---- logger ---- */
 
@@ -917,6 +1495,31 @@ void cmb_logger_error(cmb_sim* s, const char* fmt, ...) {
     logger_vlog(LOG_ERROR, "error", fmt, ap);
     va_end(ap);
     throw TrialAbandon{1};
+}
+void cmb_logger_vfprintf(cmb_sim* s, uint32_t flag, const char* fmt,
+                         va_list ap) {
+    apply_timefmt(s);
+    logger_vlog(flag, "user", fmt, ap);
+}
+void cmb_logger_fatal(cmb_sim* s, const char* fmt, ...) {
+    apply_timefmt(s);
+    va_list ap;
+    va_start(ap, fmt);
+    logger_vlog(LOG_FATAL, "fatal", fmt, ap);
+    va_end(ap);
+    abort();
+}
+void cmb_logger_user(cmb_sim* s, uint32_t flag, const char* fmt, ...) {
+    apply_timefmt(s);
+    va_list ap;
+    va_start(ap, fmt);
+    logger_vlog(flag, "user", fmt, ap);
+    va_end(ap);
+}
+void cmb_assert_failed_(const char* expr, const char* file, int line) {
+    fprintf(stderr, "cimba assertion failed: %s (%s:%d)\n", expr, file, line);
+    fflush(stderr);
+    abort();
 }
 
 }  // extern "C"

@@ -85,22 +85,28 @@ struct WtdSummary {
     double sumw;     // total weight
     double mean;     // weighted mean
     double m2;       // weighted sum of squared deviations
+    double m3, m4;   // weighted 3rd/4th central sums (skewness/kurtosis)
     double mn, mx;
 
     CMB_FORCEINLINE void reset() {
-        n = 0.0; sumw = 0.0; mean = 0.0; m2 = 0.0;
+        n = 0.0; sumw = 0.0; mean = 0.0; m2 = 0.0; m3 = 0.0; m4 = 0.0;
         mn = DBL_MAX; mx = -DBL_MAX;
     }
 
+    // weighted Pebay update (exact, associative with merge below)
     CMB_FORCEINLINE void add(double x, double w) {
         if (w <= 0.0) return;
         n += 1.0;
-        const double sw1 = sumw;
-        sumw += w;
-        const double delta = x - mean;
-        const double r = delta * w / sumw;
-        mean += r;
-        m2 += sw1 * delta * r;
+        const double wa = sumw, wx = wa + w;
+        const double d = x - mean;
+        const double dn = d * w / wx;
+        const double t1 = d * dn * wa;
+        mean += dn;
+        m4 += t1 * dn * dn * (wx * wx / (w * w) - 3.0 * wx / w + 3.0) +
+              6.0 * dn * dn * m2 - 4.0 * dn * m3;
+        m3 += t1 * dn * (wx / w - 2.0) - 3.0 * dn * m2;
+        m2 += t1;
+        sumw = wx;
         if (x < mn) mn = x;
         if (x > mx) mx = x;
     }
@@ -110,9 +116,19 @@ struct WtdSummary {
         if (sumw == 0.0) { const double na = n; *this = o; n += na; return; }
         const double wa = sumw, wb = o.sumw, wx = wa + wb;
         const double d = o.mean - mean;
-        m2 = m2 + o.m2 + d * d * wa * wb / wx;
+        const double d2 = d * d;
+        const double m2x = m2 + o.m2 + d2 * wa * wb / wx;
+        const double m3x = m3 + o.m3 +
+            d * d2 * wa * wb * (wa - wb) / (wx * wx) +
+            3.0 * d * (wa * o.m2 - wb * m2) / wx;
+        const double m4x = m4 + o.m4 +
+            d2 * d2 * wa * wb * (wa * wa - wa * wb + wb * wb) /
+                (wx * wx * wx) +
+            6.0 * d2 * (wa * wa * o.m2 + wb * wb * m2) / (wx * wx) +
+            4.0 * d * (wa * o.m3 - wb * m3) / wx;
         mean = (wa * mean + wb * o.mean) / wx;
         sumw = wx;
+        m2 = m2x; m3 = m3x; m4 = m4x;
         n += o.n;
         if (o.mn < mn) mn = o.mn;
         if (o.mx > mx) mx = o.mx;
@@ -120,6 +136,12 @@ struct WtdSummary {
 
     CMB_FORCEINLINE double variance() const { return sumw > 0.0 ? m2 / sumw : 0.0; }
     CMB_FORCEINLINE double stddev() const { return sqrt(variance()); }
+    CMB_FORCEINLINE double skewness() const {
+        return m2 > 0.0 ? sqrt(sumw) * m3 / pow(m2, 1.5) : 0.0;
+    }
+    CMB_FORCEINLINE double kurtosis() const {  // excess
+        return m2 > 0.0 ? sumw * m4 / (m2 * m2) - 3.0 : 0.0;
+    }
 };
 
 // Fixed-capacity (value, time) recorder for piecewise-constant state
