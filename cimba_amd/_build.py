@@ -19,6 +19,7 @@ SOURCES = [
     os.path.join(CSRC, "hip", "deskernel.hip"),
     os.path.join(CSRC, "hip", "rng_kernel.hip"),
     os.path.join(CSRC, "hip", "awacs_kernel.hip"),
+    os.path.join(CSRC, "hip", "multigpu.hip"),
 ]
 
 
@@ -92,7 +93,7 @@ def build(verbose=True, force=False):
         objs = list(ex.map(compile_one, SOURCES))
 
     so = ext_path()
-    cmd = [hipcc, "-shared", "-fPIC", "-o", so] + objs
+    cmd = [hipcc, "-shared", "-fPIC", "-o", so] + objs + ["-L/opt/rocm/lib", "-lrccl"]
     if verbose:
         print("[cimba_amd build]", " ".join(cmd), flush=True)
     subprocess.run(cmd, check=True)

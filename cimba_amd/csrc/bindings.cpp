@@ -60,6 +60,9 @@ int cimba_awacs_power_test_devinit(const void* params, uint64_t seed,
 int cimba_awacs_first_dwell_dbg(const void* params, uint64_t master_seed,
                                 int device, float* out_powers);
 int cimba_xlane_repro(int iters, int device, int* out64);
+int cimba_mm1_multigpu_rccl(uint64_t ntrials, double arr_mean,
+                            double srv_mean, uint64_t num_objects,
+                            uint64_t seed, int ndev, double* out10);
 }
 
 using cmb_models::AWACS;
@@ -733,6 +736,29 @@ PYBIND11_MODULE(_C, m) {
         .def("size", &Timeseries::size)
         .def("summarize", &Timeseries::summarize)
         .def("median", &Timeseries::median);
+    m.def("mm1_multigpu_rccl", [](uint64_t ntrials, uint64_t num_objects,
+                                  double arr_rate, double srv_rate,
+                                  uint64_t seed, int ndev) {
+        double o[10];
+        int rc = cimba_mm1_multigpu_rccl(ntrials, 1.0 / arr_rate,
+                                         1.0 / srv_rate, num_objects, seed,
+                                         ndev, o);
+        if (rc) throw std::runtime_error("rccl/hip error " +
+                                         std::to_string(rc));
+        py::dict d;
+        d["n"] = o[0];
+        d["mean_system_time"] = o[1];
+        d["var_system_time"] = o[2];
+        d["min"] = o[3];
+        d["max"] = o[4];
+        d["total_events"] = (uint64_t)o[5];
+        d["ndev"] = (int)o[6];
+        d["elapsed_ms"] = o[7];
+        d["events_per_sec"] = o[7] > 0 ? o[5] * 1000.0 / o[7] : 0.0;
+        return d;
+    }, py::arg("ntrials"), py::arg("num_objects"), py::arg("arr_rate") = 0.9,
+       py::arg("srv_rate") = 1.0, py::arg("seed") = 0x34f05c64d7ad598fULL,
+       py::arg("ndev") = -1);
     m.def("gpu_device_count", &gpu_device_count);
     m.def("gpu_sync", []() { return cimba_gpu_sync(); });
 

@@ -61,3 +61,16 @@ def test_bench_script_single_gpu():
     assert j["metric"] == "sim_events_per_sec"
     assert j["value"] > 0
     assert j["n_gpus"] == 1
+
+
+def test_native_rccl_multigpu_runner():
+    # single-process multi-GPU fan-out over RCCL (ndev=1 on this box:
+    # broadcast + allreduce run with nranks=1); must agree with the
+    # plain single-GPU path
+    r = ca._C.mm1_multigpu_rccl(ntrials=4096, num_objects=10000, seed=55)
+    s = ca.mm1_gpu(ntrials=4096, num_objects=10000, seed=55, device=0)
+    assert r["ndev"] >= 1
+    assert int(r["n"]) == 4096
+    assert r["total_events"] == s["total_events"]
+    assert abs(r["mean_system_time"] - s["avg_system_time"]) < 1e-9
+    assert 8.0 < r["mean_system_time"] < 12.0
