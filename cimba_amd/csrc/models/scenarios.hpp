@@ -49,6 +49,7 @@ struct Scenario : cmb::ModelBase {
         W_POOL_PREEMPT = 15,
         W_HEAP_OVERFLOW = 16,   // schedule past MAX_EV -> ST_HEAP_FULL
         W_QUEUE_OVERFLOW = 17,  // put past physical QCAP -> ST_QUEUE_FULL
+        W_COND_OBSERVER = 18,   // condition observes the resource guard
     };
 
     struct Params {
@@ -135,6 +136,7 @@ struct Scenario : cmb::ModelBase {
                             // error path; device: Engine::fail)
         F_POOL_PREEMPTOR,   // hold d, preempt a units of pool 0, hold b,
                             // release
+        F_COND_RES_FREE_WAITER,  // condition-wait until resource 0 free
         F_OVERFLOWER,       // schedule a user events (heap abort path)
         F_Q_FLOODER,        // put a objects without a consumer
     };
@@ -212,6 +214,12 @@ struct Scenario : cmb::ModelBase {
             if (CMB_SIG() != cmb::SIG_SUCCESS) trace(E, me, sigtag(E, self));
             CMB_POOL_RELEASE(0, E.pool_holding(0, me));
             trace(E, me, T_REL);
+            CMB_END();
+        }
+        case F_COND_RES_FREE_WAITER: {
+            CMB_BEGIN();
+            CMB_COND_WAIT(0, cmb::DEM_USER + 1, 0);
+            trace(E, me, T_WAKE);
             CMB_END();
         }
         case F_OVERFLOWER: {
@@ -371,6 +379,7 @@ struct Scenario : cmb::ModelBase {
     CMB_FORCEINLINE static bool demand(E_& E, int /*pidx*/, uint8_t kind,
                                        uint32_t ctx) {
         if (kind == cmb::DEM_USER) return E.globals.aux >= (int32_t)ctx;
+        if (kind == cmb::DEM_USER + 1) return E.resources[0].holder < 0;
         return false;
     }
 
@@ -479,6 +488,17 @@ struct Scenario : cmb::ModelBase {
             // put must abort the trial, not deadlock
             E.queues[0].limit = cmb::CMB_UNLIMITED;
             sp(E, 0, F_Q_FLOODER, 0, /*count*/ 64, 0, 0.0);
+            break;
+        case W_COND_OBSERVER:
+            // p1 waits on a condition whose predicate is "resource 0 is
+            // free"; the condition OBSERVES the resource's guard
+            // (reference observer registration,
+            // include/cmb_resourceguard.h:48-52): the release signal is
+            // forwarded and wakes p1 without anyone signaling the
+            // condition directly.  demand id DEM_USER+1 -> aux unused.
+            E.condition_observe(0, E.resources[0].gid);
+            sp(E, 0, F_RES_USER, 0, /*hold*/ 2, 0, 0.0);
+            sp(E, 1, F_COND_RES_FREE_WAITER, 0, 0, 0, 0.0);
             break;
         case W_POOL_PREEMPT:
             // cap 4: p0 (pri 0) holds 3 for 10; p1 (pri 5) preempts 2 at

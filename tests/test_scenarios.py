@@ -43,6 +43,9 @@ EXPECTED = {
     # lower-priority holder; victim's hold returns SIG_PREEMPTED (101)
     # and it releases only its remaining holding
     15: [(0.0, 10), (1.0, 1010), (1.0, 101), (1.0, 11), (3.0, 1011)],
+    # condition observing a resource guard: the release signal at t=2 is
+    # forwarded to the condition and wakes the predicate waiter
+    18: [(0.0, 10), (2.0, 11), (2.0, 1002)],
 }
 
 
