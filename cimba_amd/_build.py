@@ -84,6 +84,10 @@ def build(verbose=True, force=False):
     def compile_one(src):
         obj = os.path.join(objdir, os.path.basename(src) + ".o")
         cmd = [hipcc, "-c", src, "-o", obj] + cflags
+        if src.endswith(".hip") and os.environ.get("CIMBA_DEVICE_NDEBUG", "1") != "0":
+            # kernels ship assert-free (the reference quotes ~2x for
+            # stripping debug asserts); host TUs keep all three tiers
+            cmd.append("-DNDEBUG")
         if verbose:
             print("[cimba_amd build]", " ".join(cmd), flush=True)
         subprocess.run(cmd, check=True)
