@@ -74,3 +74,15 @@ def test_jobshop_gpu_matches_host():
     assert g["trials_ok"] == 32
     assert g["total_completed"] == h["total_completed"]
     assert abs(g["mean_makespan"] - h["mean_makespan"]) / h["mean_makespan"] < 0.02
+
+
+@pytest.mark.parametrize("rho", [0.5, 0.7, 0.9])
+def test_mg1_utilization_sweep(rho):
+    # the reference's M/G/1 experiment grid sweeps utilization
+    # (test_cimba.c: 5 utilizations); PK must hold at each point
+    r = ca.mg1_host(ntrials=6, num_objects=30_000, arr_rate=rho,
+                    srv_mean=1.0, srv_scv=1.0, dist=0, seed=77, threads=3)
+    theory = pk_system_time(rho, 1.0, 1.0)
+    tol = 0.06 if rho < 0.85 else 0.15  # heavier rho = slower convergence
+    assert abs(r["avg_system_time"] - theory) / theory < tol, (
+        rho, r["avg_system_time"], theory)
