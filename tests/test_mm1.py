@@ -49,3 +49,12 @@ def test_seed_replay_single_trial():
     assert s5 == s5b != ca.trial_seed(999, 6)
     again = ca.mm1_host(ntrials=8, num_objects=3000, seed=999, threads=1)
     assert again["per_trial_avg"][5] == full["per_trial_avg"][5]
+
+
+def test_models_registry():
+    from cimba_amd.models import MODELS, run
+
+    assert MODELS == ("awacs", "jobshop", "mg1", "mm1")
+    r = run("mm1", backend="cpu", ntrials=2, num_objects=2000, seed=4,
+            threads=1)
+    assert r["trials_ok"] == 2
