@@ -75,7 +75,8 @@ def main():
             ca.gpu_sync()
 
     def one_step(step_idx):
-        seed = ca.fmix64((args.seed ^ (rank << 32)) + step_idx + 1)
+        seed = ca.fmix64(((args.seed ^ (rank << 32)) + step_idx + 1)
+                         & 0xFFFFFFFFFFFFFFFF)
         ntrials = args.trials
         if args.model == "mm1":
             fn = ca.mm1_gpu if use_gpu else ca.mm1_host
