@@ -81,6 +81,24 @@ def test_capi_1000_processes_and_timers(libcimba, tmp_path):
     assert "1000-process run" in out.stdout
 
 
+def test_capi_api_tour(libcimba, tmp_path):
+    # assertion-checked tour of the wider surface: event introspection,
+    # pattern ops, multi-slot timers, queue position, alias tables,
+    # dataset/timeseries/summary C wrappers
+    exe = str(tmp_path / "api_tour")
+    r = subprocess.run(
+        ["gcc", "-std=c11", "-O1", "-g", "-Wall", "-Werror",
+         "-I", os.path.join(ROOT, "include"),
+         os.path.join(ROOT, "tutorial", "api_tour_capi.c"),
+         "-L", os.path.join(ROOT, "cimba_amd"), "-lcimba",
+         f"-Wl,-rpath,{os.path.join(ROOT, 'cimba_amd')}",
+         "-lm", "-o", exe], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    out = subprocess.run([exe], capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, (out.stdout, out.stderr)
+    assert "api tour OK" in out.stdout
+
+
 def test_capi_header_is_c_clean(tmp_path):
     # header must compile as plain C99 without the library
     src = tmp_path / "hdr.c"
