@@ -9,6 +9,7 @@
 #pragma once
 
 #include "../include/cimba/engine.hpp"
+#include "../include/cimba/stats.hpp"
 #if !defined(__HIP_DEVICE_COMPILE__)
 #include "../include/cimba/logger.hpp"
 #endif
@@ -50,6 +51,7 @@ struct Scenario : cmb::ModelBase {
         W_HEAP_OVERFLOW = 16,   // schedule past MAX_EV -> ST_HEAP_FULL
         W_QUEUE_OVERFLOW = 17,  // put past physical QCAP -> ST_QUEUE_FULL
         W_COND_OBSERVER = 18,   // condition observes the resource guard
+        W_TIMESERIES = 19,      // device-side TimeseriesRec recording
     };
 
     struct Params {
@@ -66,6 +68,7 @@ struct Scenario : cmb::ModelBase {
         int32_t n;
         int32_t aux;       // condition state variable
         uint32_t uev;      // user event handle for wait_event scenarios
+        cmb::TimeseriesRec<16> qlen;  // scenario 19: device-side series
     };
     struct Frame {
         int64_t a, b;
@@ -136,6 +139,8 @@ struct Scenario : cmb::ModelBase {
                             // error path; device: Engine::fail)
         F_POOL_PREEMPTOR,   // hold d, preempt a units of pool 0, hold b,
                             // release
+        F_TS_PUTTER,        // put at t=1,2,3 recording queue length
+        F_TS_GETTER,        // get at t=2.5, 4 recording queue length
         F_COND_RES_FREE_WAITER,  // condition-wait until resource 0 free
         F_OVERFLOWER,       // schedule a user events (heap abort path)
         F_Q_FLOODER,        // put a objects without a consumer
@@ -214,6 +219,25 @@ struct Scenario : cmb::ModelBase {
             if (CMB_SIG() != cmb::SIG_SUCCESS) trace(E, me, sigtag(E, self));
             CMB_POOL_RELEASE(0, E.pool_holding(0, me));
             trace(E, me, T_REL);
+            CMB_END();
+        }
+        case F_TS_PUTTER: {
+            CMB_BEGIN();
+            for (f.b = 0; f.b < 3; ++f.b) {
+                CMB_HOLD(1.0);
+                CMB_QPUT(0, (uint64_t)f.b);
+                g.qlen.add((double)E.queues[0].len, E.now);
+            }
+            CMB_END();
+        }
+        case F_TS_GETTER: {
+            CMB_BEGIN();
+            CMB_HOLD(2.5);
+            CMB_QGET(0, (uint64_t*)&f.a);
+            g.qlen.add((double)E.queues[0].len, E.now);
+            CMB_HOLD(1.5);
+            CMB_QGET(0, (uint64_t*)&f.a);
+            g.qlen.add((double)E.queues[0].len, E.now);
             CMB_END();
         }
         case F_COND_RES_FREE_WAITER: {
@@ -489,6 +513,11 @@ struct Scenario : cmb::ModelBase {
             E.queues[0].limit = cmb::CMB_UNLIMITED;
             sp(E, 0, F_Q_FLOODER, 0, /*count*/ 64, 0, 0.0);
             break;
+        case W_TIMESERIES:
+            E.globals.qlen.reset();
+            sp(E, 0, F_TS_PUTTER, 0, 0, 0, 0.0);
+            sp(E, 1, F_TS_GETTER, 0, 0, 0, 0.0);
+            break;
         case W_COND_OBSERVER:
             // p1 waits on a condition whose predicate is "resource 0 is
             // free"; the condition OBSERVES the resource's guard
@@ -519,6 +548,16 @@ struct Scenario : cmb::ModelBase {
         for (int i = 0; i < r.n; ++i) r.ev[i] = E.globals.ev[i];
         r.status = E.status;
         r.events = E.ev_dispatched;
+        // scenario 19: summarize the device-recorded series into the
+        // trace as (weighted-mean x 1000, count) pseudo-entries
+        if (E.params->which == W_TIMESERIES && r.n + 1 < 96) {
+            cmb::WtdSummary ws;
+            ws.reset();
+            E.globals.qlen.summarize(ws, 5.0);
+            r.ev[r.n].t = ws.mean;
+            r.ev[r.n].code = (int32_t)E.globals.qlen.len;
+            ++r.n;
+        }
     }
 };
 

@@ -81,3 +81,30 @@ def test_failure_paths_abort_cleanly(which, status):
 def test_failure_paths_abort_cleanly_gpu(which, status):
     r = ca._C.scenario_gpu(which)
     assert r["status"] == status
+
+
+def _expected_ts():
+    # puts at 1,2,3 (len 1,2,3 after; getter takes at 2.5 -> len 1, at 4 ->
+    # len 1... trace: put@1 len1, put@2 len2, get@2.5 len1, put@3 len2,
+    # get@4 len1; weighted over [1,5): 1*1+2*0.5+1*0.5+2*1+1*1 / 4
+    wmean = (1*1.0 + 2*0.5 + 1*0.5 + 2*1.0 + 1*1.0) / 4.0
+    return wmean, 5
+
+
+def test_scenario_timeseries_device_recorder_host():
+    r = ca._C.scenario_host(19)
+    assert r["status"] == 0
+    wmean, n = _expected_ts()
+    t, code = r["trace"][-1]
+    assert code == n
+    assert abs(t - wmean) < 1e-12
+
+
+@pytest.mark.gpu
+def test_scenario_timeseries_device_recorder_gpu():
+    r = ca._C.scenario_gpu(19)
+    assert r["status"] == 0
+    wmean, n = _expected_ts()
+    t, code = r["trace"][-1]
+    assert code == n
+    assert abs(t - wmean) < 1e-12
