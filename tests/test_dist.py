@@ -74,3 +74,46 @@ def test_shard_range():
         lo, hi = shard_range(10, r, 3)
         covered.extend(range(lo, hi))
     assert covered == list(range(10))
+
+
+def test_run_distributed_mm1_single_process():
+    # single-process fallback path of the distributed experiment runner
+    from cimba_amd.parallel import run_distributed_mm1
+
+    ds, events = run_distributed_mm1(8, 5000, seed=3, use_gpu=False)
+    assert ds.count() == 8
+    assert 7.0 < ds.mean() < 13.0
+    assert events > 8 * 5000
+
+
+def _dist_mm1_worker(rank, world, port, q):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+        "RANK": str(rank),
+        "WORLD_SIZE": str(world),
+    })
+    import torch.distributed as dist
+
+    dist.init_process_group(backend="gloo", rank=rank, world_size=world)
+    from cimba_amd.parallel import run_distributed_mm1
+
+    ds, events = run_distributed_mm1(8, 3000, seed=5, use_gpu=False)
+    if rank == 0:
+        q.put({"n": ds.count(), "mean": ds.mean(), "events": events})
+    dist.destroy_process_group()
+
+
+def test_run_distributed_mm1_gloo():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_dist_mm1_worker, args=(r, 2, 29513, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = q.get(timeout=180)
+    for p in procs:
+        p.join(timeout=60)
+    assert res["n"] == 8          # all shards merged
+    assert 7.0 < res["mean"] < 13.0
+    assert res["events"] > 8 * 3000
