@@ -125,6 +125,9 @@ def main():
         import torch
         te = torch.tensor([elapsed], dtype=torch.float64)
         tv = torch.tensor([float(events)], dtype=torch.float64)
+        if dist.get_backend() == "nccl":  # nccl reduces CUDA tensors only
+            te = te.cuda()
+            tv = tv.cuda()
         dist.all_reduce(te, op=dist.ReduceOp.MAX)
         dist.all_reduce(tv, op=dist.ReduceOp.SUM)
         elapsed = te.item()
