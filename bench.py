@@ -56,9 +56,9 @@ def main():
         import torch  # noqa: F811
         import torch.distributed as dist  # noqa: F811
         backend = "nccl" if (not args.host and torch.cuda.is_available()) else "gloo"
-        dist.init_process_group(backend=backend)
-        if backend == "nccl":
+        if backend == "nccl":  # bind the device before init (rank->GPU map)
             torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
 
     use_gpu = not args.host
     if use_gpu and ca.gpu_device_count() <= local_rank:
