@@ -326,12 +326,26 @@ int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                           int device, double* elapsed_ms, void* results_out) {
     HIP_TRY(hipSetDevice(device));
     const char* lane = getenv("CIMBA_JS_LANE");
-    if (lane ? atoi(lane) != 0 : ntrials >= 32768)  // HBM-lane variant:
-        // measured faster than scratch for JobShop's larger store
+    if (lane ? atoi(lane) != 0 : ntrials >= 32768) {
+        // HBM-lane variant: measured faster than scratch for JobShop's
+        // larger store; MINW knob as elsewhere
+        const char* lmw = getenv("CIMBA_JS_LANE_MINW");
+        const int lminw = lmw ? atoi(lmw) : 1;
+        if (lminw >= 3)
+            return run_trials_gpu_lane<JobShop, 3>(
+                *(const JobShop::Params*)params, ntrials, seed, 1.0e308,
+                UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
+                (JobShop::Result*)results_out, 2048u);
+        if (lminw == 2)
+            return run_trials_gpu_lane<JobShop, 2>(
+                *(const JobShop::Params*)params, ntrials, seed, 1.0e308,
+                UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
+                (JobShop::Result*)results_out, 2048u);
         return run_trials_gpu_lane<JobShop, 1>(
             *(const JobShop::Params*)params, ntrials, seed, 1.0e308,
             UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
             (JobShop::Result*)results_out, 2048u);
+    }
     const char* mw = getenv("CIMBA_JS_MINW");
     const int minw = mw ? atoi(mw) : 4;
     if (minw >= 4)
