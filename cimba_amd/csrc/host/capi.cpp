@@ -63,6 +63,8 @@ struct CModel : ModelBase {
         void* dem_ctx[C_MAX_PROC];
         UEv uev[C_UEV];
         int32_t uev_free;  // freelist head
+        uint32_t cur_event;  // most recently dispatched user event handle
+        int16_t cur_proc;    // currently executing process idx, -1 = none
         int32_t nproc, nq, npq, nres, npool, nbuf, ncond;
         char qname[Cfg::NUM_QUEUES][C_NAME];
         char rname[Cfg::NUM_RES][C_NAME];
@@ -113,6 +115,7 @@ void init_globals(CEngine& E) {
     for (int i = 0; i < C_UEV; ++i) g.uev[i].next_free = i + 1;
     g.uev[C_UEV - 1].next_free = -1;
     g.uev_free = 0;
+    g.cur_proc = -1;
 }
 
 int uev_alloc(CEngine& E, cmb_event_func* fn, void* subj, void* obj) {
@@ -137,7 +140,10 @@ template <class E_>
 void CModel::step(E_& E, int pidx) {
     cmb_sim* sim = E.params->sim;
     Globals& g = E.globals;
+    const int16_t prev = g.cur_proc;
+    g.cur_proc = (int16_t)pidx;
     if (g.fn[pidx]) g.fn[pidx](sim, enc<cmb_process>(pidx), g.ctx[pidx]);
+    g.cur_proc = prev;
 }
 
 template <class E_>
@@ -154,6 +160,7 @@ void CModel::on_event(E_& E, const EvEntry& ev) {
     const int s = (int)ev.b;
     UEv u = E.globals.uev[s];
     uev_free_slot(E, s);
+    E.globals.cur_event = ev.handle;
     if (u.fn) u.fn(E.params->sim, u.subj, u.obj);
 }
 
@@ -374,8 +381,21 @@ uint64_t cmb_event_pattern_find(cmb_sim* s, cmb_event_func* action,
 void cmb_event_queue_execute_until(cmb_sim* s, double until) {
     s->E->run(until, UINT64_C(0xFFFFFFFFFFFFFFFF));
 }
+uint64_t cmb_event_current(const cmb_sim* s) {
+    // reference cmb_event.h:189 — handle of the currently / most
+    // recently executed user event, 0 if none yet
+    return (uint64_t)s->E->globals.cur_event;
+}
 
 /* ---- processes ---- */
+
+cmb_process* cmb_process_current(const cmb_sim* s) {
+    // reference cmb_process.h:256 — the process whose body is executing,
+    // NULL from the dispatcher / trial function (no coroutines here: the
+    // trampoline in CModel::step tracks the running protothread instead)
+    const int16_t p = s->E->globals.cur_proc;
+    return p < 0 ? nullptr : enc<cmb_process>((int)p);
+}
 
 cmb_process* cmb_process_spawn(cmb_sim* s, const char* name,
                                cmb_process_func* fn, void* ctx,
@@ -1002,23 +1022,49 @@ void cmb_condition_unsubscribe_all(cmb_sim* s, cmb_condition* c) {
 }
 
 /* resource guard = a condition whose signal() evaluates the front waiter
- * only (the reference's guard contract) */
+ * only (the reference's guard contract).  A guard handle is either a
+ * condition (standalone guard) or a TAGGED engine guard id: the guards
+ * embedded in built-in objects (reference cmb_resource.h:205 /
+ * cmb_resourcepool.h:263 return `&rp->guard`) are exposed the same way
+ * here, as handles that signal / wait on / cancel against the object's
+ * own guard queue. */
+static const uintptr_t RG_TAG = (uintptr_t)1 << 30;
+static int rg_gid(const cmb_sim* s, const cmb_resourceguard* g) {
+    const uintptr_t v = (uintptr_t)g;
+    if (v & RG_TAG) return (int)((v & ~RG_TAG) - 1);
+    return s->E->conds[dec((const cmb_condition*)g)].gid;
+}
 cmb_resourceguard* cmb_resourceguard_create(cmb_sim* s) {
     return (cmb_resourceguard*)cmb_condition_create(s);
 }
 void cmb_resourceguard_initialize(cmb_sim*, cmb_resourceguard*,
                                   const char*) {}
+cmb_resourceguard* cmb_resource_guard(cmb_sim* s, cmb_resource* r) {
+    return (cmb_resourceguard*)(
+        ((uintptr_t)(s->E->resources[dec(r)].gid + 1)) | RG_TAG);
+}
+cmb_resourceguard* cmb_resourcepool_guard(cmb_sim* s, cmb_resourcepool* p) {
+    return (cmb_resourceguard*)(
+        ((uintptr_t)(s->E->pools[dec(p)].gid + 1)) | RG_TAG);
+}
 bool cmb_resourceguard_signal(cmb_sim* s, cmb_resourceguard* g) {
-    return s->E->guard_signal(s->E->conds[dec(g)].gid);
+    return s->E->guard_signal(rg_gid(s, g));
 }
 bool cmb_resourceguard_cancel(cmb_sim* s, cmb_resourceguard* g,
                               cmb_process* p) {
-    return cmb_condition_cancel(s, (cmb_condition*)g, p);
+    CEngine& E = *s->E;
+    auto& pr = E.procs[dec(p)];
+    if (pr.await_kind != AW_GUARD || pr.gid != rg_gid(s, g)) return false;
+    E.proc_interrupt(dec(p), SIG_CANCELLED);
+    return true;
 }
 void cmb_guard_wait_setup_(cmb_sim* s, cmb_resourceguard* g, cmb_process* p,
                            cmb_resourceguard_demand_func* fn, void* ctx) {
-    cmb_condition_wait_setup_(s, (cmb_condition*)g, p,
-                              (cmb_demand_func*)fn, ctx);
+    CEngine& E = *s->E;
+    const int pidx = dec(p);
+    E.globals.dem_fn[pidx] = (cmb_demand_func*)fn;
+    E.globals.dem_ctx[pidx] = ctx;
+    E.guard_wait(E.procs[pidx], rg_gid(s, g), DEM_USER, 0);
 }
 
 /* ---- debug dumps & reports ---- */
@@ -1346,6 +1392,12 @@ void cmb_timeseries_histogram_print(const cmb_timeseries* t, int nbins,
                                     FILE* out) {
     Dataset d = ts_values_((const CTimeseries*)t);
     cmb_dataset_histogram_print((cmb_dataset*)&d, nbins, out);
+}
+void cmb_timeseries_correlogram_print(const cmb_timeseries* t, int maxlag,
+                                      FILE* out) {
+    // reference cmb_timeseries.h:342 (delegates to the dataset version)
+    Dataset d = ts_values_((const CTimeseries*)t);
+    cmb_dataset_correlogram_print((cmb_dataset*)&d, maxlag, out);
 }
 void cmb_timeseries_sort_x(cmb_timeseries* t) {
     auto* ct = (CTimeseries*)t;

@@ -118,8 +118,14 @@ int cmb_event_priority(const cmb_sim* sim, uint64_t handle);
 bool cmb_event_reprioritize(cmb_sim* sim, uint64_t handle, int priority);
 uint64_t cmb_event_pattern_find(cmb_sim* sim, cmb_event_func* action,
                                 void* subject, void* object);
+/* handle of the currently / most recently executed event, 0 if none
+ * (reference cmb_event_current; takes the sim context instead of TLS) */
+uint64_t cmb_event_current(const cmb_sim* sim);
 
 /* ---- processes (reference include/cmb_process.h) ---- */
+/* currently executing process, NULL from the dispatcher / trial function
+ * (reference cmb_process_current; sim context instead of coroutine TLS) */
+cmb_process* cmb_process_current(const cmb_sim* sim);
 cmb_process* cmb_process_spawn(cmb_sim* sim, const char* name,
                                cmb_process_func* fn, void* ctx,
                                int priority);
@@ -500,6 +506,12 @@ typedef cmb_resourceguard_demand_func cmb_condition_demand_func;
 cmb_resourceguard* cmb_resourceguard_create(cmb_sim* sim);
 void cmb_resourceguard_initialize(cmb_sim* sim, cmb_resourceguard* g,
                                   const char* name);
+/* the guard EMBEDDED in a built-in object (reference cmb_resource_guard /
+ * cmb_resourcepool_guard return &obj->guard): signal it after changing
+ * external state a waiter's demand depends on, wait on it with
+ * CMB_RESOURCEGUARD_WAIT, cancel waiters with cmb_resourceguard_cancel */
+cmb_resourceguard* cmb_resource_guard(cmb_sim* sim, cmb_resource* r);
+cmb_resourceguard* cmb_resourcepool_guard(cmb_sim* sim, cmb_resourcepool* p);
 bool cmb_resourceguard_signal(cmb_sim* sim, cmb_resourceguard* g);
 bool cmb_resourceguard_cancel(cmb_sim* sim, cmb_resourceguard* g,
                               cmb_process* p);
@@ -620,6 +632,8 @@ double cmb_timeseries_max(const cmb_timeseries* t);
 void cmb_timeseries_fivenum_print(const cmb_timeseries* t, FILE* out);
 void cmb_timeseries_histogram_print(const cmb_timeseries* t, int nbins,
                                     FILE* out);
+void cmb_timeseries_correlogram_print(const cmb_timeseries* t, int maxlag,
+                                      FILE* out);
 /* x-sorted / t-sorted copies (reference sort_x/sort_t) */
 void cmb_timeseries_sort_x(cmb_timeseries* t);
 void cmb_timeseries_sort_t(cmb_timeseries* t);

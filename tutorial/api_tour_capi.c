@@ -35,6 +35,8 @@ static void tour_body(cmb_sim* sim, cmb_process* me, void* vctx) {
     struct tour_ctx* ctx = vctx;
     CMB_PROC_BEGIN(sim, me);
 
+    assert(cmb_process_current(sim) == me);
+
     /* ---- event introspection ---- */
     ctx->ev_a = cmb_event_schedule(sim, noop_event, (void*)1, NULL, 5.0, 0);
     ctx->ev_b = cmb_event_schedule(sim, noop_event, (void*)2, NULL, 7.0, 3);
@@ -77,9 +79,45 @@ static void tour_body(cmb_sim* sim, cmb_process* me, void* vctx) {
     CMB_PROC_END(sim, me);
 }
 
+/* ---- embedded resource guard: wait on cmb_resource_guard() with a
+ * custom demand, woken by the release-driven guard signal ---- */
+struct rg_ctx {
+    cmb_resource* res;
+    double t0;
+    int got;
+};
+
+static bool res_free_demand(cmb_sim* sim, void* vctx) {
+    const struct rg_ctx* c = vctx;
+    return cmb_resource_available(sim, c->res);
+}
+
+static void rg_holder(cmb_sim* sim, cmb_process* me, void* vctx) {
+    struct rg_ctx* c = vctx;
+    CMB_PROC_BEGIN(sim, me);
+    CMB_RESOURCE_ACQUIRE(sim, me, c->res);
+    CMB_HOLD(sim, me, 5.0);
+    cmb_resource_release(sim, c->res, me);
+    CMB_PROC_END(sim, me);
+}
+
+static void rg_waiter(cmb_sim* sim, cmb_process* me, void* vctx) {
+    struct rg_ctx* c = vctx;
+    CMB_PROC_BEGIN(sim, me);
+    CMB_HOLD(sim, me, 1.0); /* let the holder grab the resource first */
+    while (!cmb_resource_available(sim, c->res)) {
+        CMB_RESOURCEGUARD_WAIT(sim, me, cmb_resource_guard(sim, c->res),
+                               res_free_demand, c);
+    }
+    c->got = 1;
+    assert(fabs(cmb_time(sim) - c->t0 - 5.0) < 1e-12);
+    CMB_PROC_END(sim, me);
+}
+
 static void run_trial(cmb_sim* sim, void* vtrl) {
     (void)vtrl;
     static struct tour_ctx ctx;
+    assert(cmb_process_current(sim) == NULL); /* dispatcher context */
     ctx.q = cmb_objectqueue_create(sim);
     cmb_objectqueue_initialize(sim, ctx.q, "TourQ", 16);
     assert(cmb_objectqueue_space(sim, ctx.q) == 16);
@@ -87,6 +125,25 @@ static void run_trial(cmb_sim* sim, void* vtrl) {
     cmb_process_start(sim, p);
     cmb_event_queue_execute(sim);
     assert(g_fired == 1 + 2);  /* events a+b fired; c canceled */
+    assert(cmb_event_current(sim) == ctx.ev_b); /* b (t=7) ran last */
+
+    /* ---- built-in object guards ---- */
+    static struct rg_ctx rc;
+    rc.res = cmb_resource_create(sim);
+    cmb_resource_initialize(sim, rc.res, "RGRes");
+    rc.t0 = cmb_time(sim);
+    rc.got = 0;
+    cmb_process* ph = cmb_process_spawn(sim, "Holder", rg_holder, &rc, 0);
+    cmb_process* pw = cmb_process_spawn(sim, "Waiter", rg_waiter, &rc, 0);
+    cmb_process_start(sim, ph);
+    cmb_process_start(sim, pw);
+    cmb_event_queue_execute(sim);
+    assert(rc.got == 1);
+    /* pool guard handle + signal with no waiters = no grant */
+    cmb_resourcepool* pl = cmb_resourcepool_create(sim);
+    cmb_resourcepool_initialize(sim, pl, "RGPool", 4);
+    assert(cmb_resourcepool_guard(sim, pl) != NULL);
+    assert(!cmb_resourceguard_signal(sim, cmb_resourcepool_guard(sim, pl)));
 
     /* ---- per-trial RNG helpers ---- */
     double w[4] = {1, 2, 3, 4};
@@ -132,6 +189,15 @@ int main(void) {
     assert(fabs(cmb_wtdsummary_mean(&ws) - 2.0) < 1e-12); /* (1*2+3*2)/4 */
     assert(cmb_timeseries_min(ts) == 1.0 && cmb_timeseries_max(ts) == 3.0);
     cmb_timeseries_destroy(ts);
+
+    /* timeseries correlogram (delegates to the dataset version) */
+    cmb_timeseries* t2 = cmb_timeseries_create();
+    for (int i = 0; i < 64; i++)
+        cmb_timeseries_add(t2, (double)(i % 5), (double)i);
+    FILE* nul = fopen("/dev/null", "w");
+    cmb_timeseries_correlogram_print(t2, 5, nul);
+    fclose(nul);
+    cmb_timeseries_destroy(t2);
 
     /* weighted skewness sanity: symmetric data -> ~0 */
     cmb_wtdsummary* w2 = cmb_wtdsummary_create();
