@@ -50,6 +50,9 @@ def _sources_mtime():
             if f.endswith((".cpp", ".hpp", ".hip", ".h")):
                 latest = max(latest, os.path.getmtime(os.path.join(base, f)))
     latest = max(latest, os.path.getmtime(os.path.abspath(__file__)))
+    pub = os.path.join(os.path.dirname(ROOT), "include", "cimba.h")
+    if os.path.exists(pub):
+        latest = max(latest, os.path.getmtime(pub))
     return latest
 
 

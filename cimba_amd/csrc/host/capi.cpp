@@ -88,13 +88,51 @@ using CEngine = Engine<CModel>;
 
 }  // namespace
 
+/* host timeseries wrapper (also the cmb_timeseries representation) */
+struct CTimeseries {
+    Timeseries ts;
+    double end = -1.0;
+};
+
+/* host-side recorded histories (reference cmb_*_history: a timeseries of
+ * the object's level, appended on every successful state change while
+ * recording is on; the C API is host-only, so these live beside the sim
+ * instead of inside the device-shaped engine Storage) */
+struct SimHist {
+    CTimeseries* q[CModel::Cfg::NUM_QUEUES] = {};
+    CTimeseries* pq[CModel::Cfg::NUM_PQ] = {};
+    CTimeseries* res[CModel::Cfg::NUM_RES] = {};
+    CTimeseries* pool[CModel::Cfg::NUM_POOLS] = {};
+    CTimeseries* buf[CModel::Cfg::NUM_BUFS] = {};
+    void reset() {
+        auto z = [](CTimeseries** a, int n) {
+            for (int i = 0; i < n; ++i) {
+                delete a[i];
+                a[i] = nullptr;
+            }
+        };
+        z(q, CModel::Cfg::NUM_QUEUES);
+        z(pq, CModel::Cfg::NUM_PQ);
+        z(res, CModel::Cfg::NUM_RES);
+        z(pool, CModel::Cfg::NUM_POOLS);
+        z(buf, CModel::Cfg::NUM_BUFS);
+    }
+    ~SimHist() { reset(); }
+};
+
 struct cmb_sim {
     CEngine* E;
     CModel::Params params;
     uint64_t seed;
+    SimHist* hist = nullptr;
 };
 namespace {
 using CStorage = CEngine::Storage;
+
+// append a level sample at the current sim time, if a history is live
+void hist_add(const cmb_sim* s, CTimeseries* h, double v) {
+    if (h) h->ts.add(v, s->E->now);
+}
 }
 
 namespace {
@@ -200,6 +238,8 @@ uint64_t cimba_run(void* experiment, uint64_t n, size_t size,
         cmb_sim sim;
         sim.E = eng.get();
         sim.params.sim = &sim;
+        SimHist hist;
+        sim.hist = &hist;
         for (;;) {
             const uint64_t t = g_next.fetch_add(1);
             if (t >= n) break;
@@ -210,6 +250,7 @@ uint64_t cimba_run(void* experiment, uint64_t n, size_t size,
             logger_ctx().sim_time = 0.0;
             eng->init(&sim.params, seed, (uint32_t)t);
             init_globals(*eng);
+            hist.reset();
             try {
                 trial_fn(&sim, (char*)experiment + t * size);
             } catch (const TrialAbandon&) {
@@ -579,6 +620,15 @@ void cmb_objectqueue_recording_start(cmb_sim* s, cmb_objectqueue* q) {
     Q.recording = 1;
     Q.len_stats.reset();
     Q.t_last = s->E->now;
+    if (s->hist && !s->hist->q[dec(q)])
+        s->hist->q[dec(q)] = new CTimeseries;
+    if (s->hist) hist_add(s, s->hist->q[dec(q)], (double)Q.len);
+}
+cmb_timeseries* cmb_objectqueue_history(const cmb_sim* s,
+                                        const cmb_objectqueue* q) {
+    // reference cmb_objectqueue.h:226 — queue-length history (recorded
+    // while recording is on; NULL before the first recording_start)
+    return s->hist ? (cmb_timeseries*)s->hist->q[dec(q)] : NULL;
 }
 void cmb_objectqueue_recording_stop(cmb_sim* s, cmb_objectqueue* q) {
     s->E->queues[dec(q)].recording = 0;
@@ -595,13 +645,21 @@ void cmb_objectqueue_stats(cmb_sim* s, const cmb_objectqueue* q,
 
 bool cmb_queue_try_put_(cmb_sim* s, cmb_objectqueue* q, cmb_process* p,
                         void* object) {
-    return s->E->q_try_put(dec(q), s->E->procs[dec(p)], (uint64_t)object);
+    if (!s->E->q_try_put(dec(q), s->E->procs[dec(p)], (uint64_t)object))
+        return false;
+    const auto& Q = s->E->queues[dec(q)];
+    if (Q.recording && s->hist)
+        hist_add(s, s->hist->q[dec(q)], (double)Q.len);
+    return true;
 }
 bool cmb_queue_try_get_(cmb_sim* s, cmb_objectqueue* q, cmb_process* p,
                         void** object) {
     uint64_t v = 0;
     if (!s->E->q_try_get(dec(q), s->E->procs[dec(p)], &v)) return false;
     *object = (void*)v;
+    const auto& Q = s->E->queues[dec(q)];
+    if (Q.recording && s->hist)
+        hist_add(s, s->hist->q[dec(q)], (double)Q.len);
     return true;
 }
 void cmb_queue_wait_space_(cmb_sim* s, cmb_objectqueue* q, cmb_process* p) {
@@ -709,6 +767,14 @@ void cmb_priorityqueue_recording_start(cmb_sim* s, cmb_priorityqueue* q) {
     Q.recording = 1;
     Q.len_stats.reset();
     Q.t_last = s->E->now;
+    if (s->hist && !s->hist->pq[dec(q)])
+        s->hist->pq[dec(q)] = new CTimeseries;
+    if (s->hist) hist_add(s, s->hist->pq[dec(q)], (double)Q.len);
+}
+cmb_timeseries* cmb_priorityqueue_history(const cmb_sim* s,
+                                          const cmb_priorityqueue* q) {
+    // reference cmb_priorityqueue.h:258
+    return s->hist ? (cmb_timeseries*)s->hist->pq[dec(q)] : NULL;
 }
 void cmb_priorityqueue_recording_stop(cmb_sim* s, cmb_priorityqueue* q) {
     s->E->pqueues[dec(q)].recording = 0;
@@ -723,14 +789,22 @@ void cmb_priorityqueue_report_print(cmb_sim* s, const cmb_priorityqueue* q,
 }
 bool cmb_pqueue_try_put_(cmb_sim* s, cmb_priorityqueue* q, cmb_process* p,
                          void* object, int priority) {
-    return s->E->pq_try_put(dec(q), s->E->procs[dec(p)], (uint64_t)object,
-                            priority);
+    if (!s->E->pq_try_put(dec(q), s->E->procs[dec(p)], (uint64_t)object,
+                          priority))
+        return false;
+    const auto& Q = s->E->pqueues[dec(q)];
+    if (Q.recording && s->hist)
+        hist_add(s, s->hist->pq[dec(q)], (double)Q.len);
+    return true;
 }
 bool cmb_pqueue_try_get_(cmb_sim* s, cmb_priorityqueue* q, cmb_process* p,
                          void** object) {
     uint64_t v = 0;
     if (!s->E->pq_try_get(dec(q), s->E->procs[dec(p)], &v)) return false;
     *object = (void*)v;
+    const auto& Q = s->E->pqueues[dec(q)];
+    if (Q.recording && s->hist)
+        hist_add(s, s->hist->pq[dec(q)], (double)Q.len);
     return true;
 }
 void cmb_pqueue_wait_space_(cmb_sim* s, cmb_priorityqueue* q,
@@ -763,6 +837,8 @@ const char* cmb_resource_name(const cmb_sim* s, const cmb_resource* r) {
 }
 void cmb_resource_release(cmb_sim* s, cmb_resource* r, cmb_process*) {
     s->E->resource_release(dec(r));
+    const auto& R = s->E->resources[dec(r)];
+    if (R.recording && s->hist) hist_add(s, s->hist->res[dec(r)], 0.0);
 }
 bool cmb_resource_in_use(const cmb_sim* s, const cmb_resource* r) {
     return s->E->resources[dec(r)].holder >= 0;
@@ -792,6 +868,15 @@ void cmb_resource_recording_start(cmb_sim* s, cmb_resource* r) {
     R.recording = 1;
     R.busy.reset();
     R.t_last = s->E->now;
+    if (s->hist && !s->hist->res[dec(r)])
+        s->hist->res[dec(r)] = new CTimeseries;
+    if (s->hist)
+        hist_add(s, s->hist->res[dec(r)], R.holder >= 0 ? 1.0 : 0.0);
+}
+cmb_timeseries* cmb_resource_history(const cmb_sim* s,
+                                     const cmb_resource* r) {
+    // reference cmb_resource.h:236 — busy (0/1) history
+    return s->hist ? (cmb_timeseries*)s->hist->res[dec(r)] : NULL;
 }
 void cmb_resource_stats(cmb_sim* s, const cmb_resource* r, double out4[4]) {
     auto R = s->E->resources[dec(r)];  // copy
@@ -802,7 +887,10 @@ void cmb_resource_stats(cmb_sim* s, const cmb_resource* r, double out4[4]) {
     out4[3] = R.busy.mx;
 }
 bool cmb_resource_try_acquire_(cmb_sim* s, cmb_resource* r, cmb_process* p) {
-    return s->E->res_try_acquire(dec(r), s->E->procs[dec(p)]);
+    if (!s->E->res_try_acquire(dec(r), s->E->procs[dec(p)])) return false;
+    const auto& R = s->E->resources[dec(r)];
+    if (R.recording && s->hist) hist_add(s, s->hist->res[dec(r)], 1.0);
+    return true;
 }
 bool cmb_resource_try_preempt_(cmb_sim* s, cmb_resource* r, cmb_process* p) {
     return s->E->res_try_preempt(dec(r), s->E->procs[dec(p)]);
@@ -834,6 +922,9 @@ const char* cmb_resourcepool_get_name(const cmb_sim* s,
 void cmb_resourcepool_release(cmb_sim* s, cmb_resourcepool* r,
                               cmb_process* holder, int32_t amount) {
     s->E->pool_release_for(dec(r), dec(holder), amount);
+    const auto& P = s->E->pools[dec(r)];
+    if (P.recording && s->hist)
+        hist_add(s, s->hist->pool[dec(r)], (double)P.in_use);
 }
 int32_t cmb_resourcepool_holding(const cmb_sim* s, const cmb_resourcepool* r,
                                  const cmb_process* p) {
@@ -867,6 +958,14 @@ void cmb_resourcepool_start_recording(cmb_sim* s, cmb_resourcepool* r) {
     P.recording = 1;
     P.use_stats.reset();
     P.t_last = s->E->now;
+    if (s->hist && !s->hist->pool[dec(r)])
+        s->hist->pool[dec(r)] = new CTimeseries;
+    if (s->hist) hist_add(s, s->hist->pool[dec(r)], (double)P.in_use);
+}
+cmb_timeseries* cmb_resourcepool_get_history(const cmb_sim* s,
+                                             const cmb_resourcepool* r) {
+    // reference cmb_resourcepool.h:293 — units-in-use history
+    return s->hist ? (cmb_timeseries*)s->hist->pool[dec(r)] : NULL;
 }
 void cmb_resourcepool_stop_recording(cmb_sim* s, cmb_resourcepool* r) {
     s->E->pools[dec(r)].recording = 0;
@@ -882,11 +981,20 @@ void cmb_resourcepool_stats(cmb_sim* s, const cmb_resourcepool* r,
 }
 int32_t cmb_pool_try_take_(cmb_sim* s, cmb_resourcepool* r, cmb_process* p,
                            int32_t want) {
-    return s->E->pool_try_take(dec(r), s->E->procs[dec(p)], want);
+    const int32_t got = s->E->pool_try_take(dec(r), s->E->procs[dec(p)], want);
+    const auto& P = s->E->pools[dec(r)];
+    if (got > 0 && P.recording && s->hist)
+        hist_add(s, s->hist->pool[dec(r)], (double)P.in_use);
+    return got;
 }
 bool cmb_pool_try_take_all_(cmb_sim* s, cmb_resourcepool* r, cmb_process* p,
                             int32_t want) {
-    return s->E->pool_try_take_all(dec(r), s->E->procs[dec(p)], want);
+    if (!s->E->pool_try_take_all(dec(r), s->E->procs[dec(p)], want))
+        return false;
+    const auto& P = s->E->pools[dec(r)];
+    if (P.recording && s->hist)
+        hist_add(s, s->hist->pool[dec(r)], (double)P.in_use);
+    return true;
 }
 void cmb_pool_wait_(cmb_sim* s, cmb_resourcepool* r, cmb_process* p) {
     CEngine& E = *s->E;
@@ -934,6 +1042,13 @@ void cmb_buffer_recording_start(cmb_sim* s, cmb_buffer* b) {
     B.recording = 1;
     B.level_stats.reset();
     B.t_last = s->E->now;
+    if (s->hist && !s->hist->buf[dec(b)])
+        s->hist->buf[dec(b)] = new CTimeseries;
+    if (s->hist) hist_add(s, s->hist->buf[dec(b)], (double)B.level);
+}
+cmb_timeseries* cmb_buffer_history(const cmb_sim* s, const cmb_buffer* b) {
+    // reference cmb_buffer.h:230 — level history
+    return s->hist ? (cmb_timeseries*)s->hist->buf[dec(b)] : NULL;
 }
 void cmb_buffer_recording_stop(cmb_sim* s, cmb_buffer* b) {
     s->E->buffers[dec(b)].recording = 0;
@@ -956,11 +1071,21 @@ void cmb_buffer_print_report(cmb_sim* s, const cmb_buffer* b, FILE* out) {
 }
 bool cmb_buffer_try_get_(cmb_sim* s, cmb_buffer* b, cmb_process* p,
                          int64_t amount) {
-    return s->E->buf_try_get(dec(b), s->E->procs[dec(p)], amount);
+    if (!s->E->buf_try_get(dec(b), s->E->procs[dec(p)], amount))
+        return false;
+    const auto& B = s->E->buffers[dec(b)];
+    if (B.recording && s->hist)
+        hist_add(s, s->hist->buf[dec(b)], (double)B.level);
+    return true;
 }
 bool cmb_buffer_try_put_(cmb_sim* s, cmb_buffer* b, cmb_process* p,
                          int64_t amount) {
-    return s->E->buf_try_put(dec(b), s->E->procs[dec(p)], amount);
+    if (!s->E->buf_try_put(dec(b), s->E->procs[dec(p)], amount))
+        return false;
+    const auto& B = s->E->buffers[dec(b)];
+    if (B.recording && s->hist)
+        hist_add(s, s->hist->buf[dec(b)], (double)B.level);
+    return true;
 }
 void cmb_buffer_wait_level_(cmb_sim* s, cmb_buffer* b, cmb_process* p,
                             int64_t amount) {
@@ -1331,10 +1456,6 @@ void cmb_dataset_print(const cmb_dataset* d, FILE* out) {
     for (double x : v) fprintf(out, "%.9g\n", x);
 }
 
-struct CTimeseries {
-    Timeseries ts;
-    double end = -1.0;
-};
 static double ts_end_(const CTimeseries* t, double end_time) {
     if (end_time >= 0.0) return end_time;
     if (t->end >= 0.0) return t->end;

@@ -53,6 +53,7 @@ typedef struct cmb_resource cmb_resource;
 typedef struct cmb_resourcepool cmb_resourcepool;
 typedef struct cmb_buffer cmb_buffer;
 typedef struct cmb_condition cmb_condition;
+typedef struct cmb_timeseries cmb_timeseries;  /* full API further below */
 
 typedef void (cmb_process_func)(cmb_sim* sim, cmb_process* me, void* ctx);
 typedef void (cmb_event_func)(cmb_sim* sim, void* subject, void* object);
@@ -389,6 +390,10 @@ uint64_t cmb_objectqueue_position(const cmb_sim* sim,
                                   const void* object);
 void cmb_objectqueue_recording_start(cmb_sim* sim, cmb_objectqueue* q);
 void cmb_objectqueue_recording_stop(cmb_sim* sim, cmb_objectqueue* q);
+/* recorded queue-length history while recording is on (reference
+ * cmb_objectqueue_history); NULL before the first recording_start */
+cmb_timeseries* cmb_objectqueue_history(const cmb_sim* sim,
+                                        const cmb_objectqueue* q);
 /* time-weighted length stats while recording: mean/stddev/min/max */
 void cmb_objectqueue_stats(cmb_sim* sim, const cmb_objectqueue* q,
                            double out4[4]);
@@ -410,6 +415,8 @@ bool cmb_priorityqueue_reprioritize(cmb_sim* sim, cmb_priorityqueue* q,
                                     const void* object, int priority);
 void cmb_priorityqueue_recording_start(cmb_sim* sim, cmb_priorityqueue* q);
 void cmb_priorityqueue_recording_stop(cmb_sim* sim, cmb_priorityqueue* q);
+cmb_timeseries* cmb_priorityqueue_history(const cmb_sim* sim,
+                                          const cmb_priorityqueue* q);
 void cmb_priorityqueue_report_print(cmb_sim* sim, const cmb_priorityqueue* q,
                                     FILE* out);
 uint64_t cmb_objectqueue_space(const cmb_sim* sim, const cmb_objectqueue* q);
@@ -432,6 +439,9 @@ void cmb_resource_recording_stop(cmb_sim* sim, cmb_resource* r);
 /* reference aliases */
 void cmb_resource_start_recording(cmb_sim* sim, cmb_resource* r);
 void cmb_resource_stop_recording(cmb_sim* sim, cmb_resource* r);
+/* busy (0/1) history (reference cmb_resource_history) */
+cmb_timeseries* cmb_resource_history(const cmb_sim* sim,
+                                     const cmb_resource* r);
 bool cmb_resource_available(const cmb_sim* sim, const cmb_resource* r);
 bool cmb_resource_held_by_process(const cmb_sim* sim, const cmb_resource* r,
                                   const cmb_process* p);
@@ -458,6 +468,10 @@ int32_t cmb_resourcepool_held_by_process(const cmb_sim* sim,
                                          const cmb_process* p);
 void cmb_resourcepool_start_recording(cmb_sim* sim, cmb_resourcepool* r);
 void cmb_resourcepool_stop_recording(cmb_sim* sim, cmb_resourcepool* r);
+/* units-in-use history (reference cmb_resourcepool_get_history) */
+cmb_timeseries* cmb_resourcepool_get_history(const cmb_sim* sim,
+                                             const cmb_resourcepool* r);
+#define cmb_resourcepool_history cmb_resourcepool_get_history
 void cmb_resourcepool_stats(cmb_sim* sim, const cmb_resourcepool* r,
                             double out4[4]);
 
@@ -469,6 +483,8 @@ int64_t cmb_buffer_capacity(const cmb_sim* sim, const cmb_buffer* b);
 int64_t cmb_buffer_space(const cmb_sim* sim, const cmb_buffer* b);
 void cmb_buffer_recording_start(cmb_sim* sim, cmb_buffer* b);
 void cmb_buffer_recording_stop(cmb_sim* sim, cmb_buffer* b);
+/* level history (reference cmb_buffer_history) */
+cmb_timeseries* cmb_buffer_history(const cmb_sim* sim, const cmb_buffer* b);
 void cmb_buffer_stats(cmb_sim* sim, const cmb_buffer* b, double out4[4]);
 void cmb_buffer_print_report(cmb_sim* sim, const cmb_buffer* b, FILE* out);
 

@@ -133,12 +133,18 @@ static void run_trial(cmb_sim* sim, void* vtrl) {
     cmb_resource_initialize(sim, rc.res, "RGRes");
     rc.t0 = cmb_time(sim);
     rc.got = 0;
+    cmb_resource_recording_start(sim, rc.res);
     cmb_process* ph = cmb_process_spawn(sim, "Holder", rg_holder, &rc, 0);
     cmb_process* pw = cmb_process_spawn(sim, "Waiter", rg_waiter, &rc, 0);
     cmb_process_start(sim, ph);
     cmb_process_start(sim, pw);
     cmb_event_queue_execute(sim);
     assert(rc.got == 1);
+    /* recorded busy history: 0 at start, 1 at acquire, 0 at release */
+    cmb_timeseries* rh = cmb_resource_history(sim, rc.res);
+    assert(rh != NULL);
+    assert(cmb_timeseries_count(rh) == 3);
+    assert(cmb_timeseries_min(rh) == 0.0 && cmb_timeseries_max(rh) == 1.0);
     /* pool guard handle + signal with no waiters = no grant */
     cmb_resourcepool* pl = cmb_resourcepool_create(sim);
     cmb_resourcepool_initialize(sim, pl, "RGPool", 4);
