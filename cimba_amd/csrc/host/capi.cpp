@@ -1194,15 +1194,30 @@ void cmb_guard_wait_setup_(cmb_sim* s, cmb_resourceguard* g, cmb_process* p,
 
 /* ---- debug dumps & reports ---- */
 
-void cmb_event_queue_print(cmb_sim* s, FILE* out) {
+void cmb_event_queue_print_formatted(cmb_sim* s, FILE* out,
+                                     cmb_event_print_formatter* epf) {
+    // reference cmb_event_queue_print(FILE*, cmb_event_print_formatter*):
+    // the formatter labels user events from (action, subject, object)
     if (!out) out = stderr;
     auto& q = s->E->evq;
     fprintf(out, "event queue @ t=%.6f: %d pending\n", s->E->now, q.n);
     for (int32_t i = 0; i < q.n; ++i) {
         const auto& e = q.e[i];
-        fprintf(out, "  [%2d] t=%.6f kind=%u a=%u handle=%u\n", i, e.t,
-                (unsigned)e.kind, (unsigned)e.a, e.handle);
+        const char* label = NULL;
+        if (epf && e.kind == EV_USER) {
+            const auto& u = s->E->globals.uev[(int)e.b];
+            label = epf(u.fn, u.subj, u.obj);
+        }
+        if (label)
+            fprintf(out, "  [%2d] t=%.6f handle=%u %s\n", i, e.t, e.handle,
+                    label);
+        else
+            fprintf(out, "  [%2d] t=%.6f kind=%u a=%u handle=%u\n", i, e.t,
+                    (unsigned)e.kind, (unsigned)e.a, e.handle);
     }
+}
+void cmb_event_queue_print(cmb_sim* s, FILE* out) {
+    cmb_event_queue_print_formatted(s, out, NULL);
 }
 
 void cmb_resource_print_report(cmb_sim* s, const cmb_resource* r, FILE* out) {

@@ -543,6 +543,14 @@ void cmb_guard_wait_setup_(cmb_sim* sim, cmb_resourceguard* g,
 /* ---- debug dumps & reports (reference cmb_event_queue_print,
  * cmb_resource_print_report et al.) ---- */
 void cmb_event_queue_print(cmb_sim* sim, FILE* out);
+/* user-event labeler for queue dumps (reference cmb_event_print_formatter):
+ * return a static/thread-local string naming the event, or NULL to fall
+ * back to the default kind/handle line */
+typedef const char* (cmb_event_print_formatter)(cmb_event_func* action,
+                                                const void* subject,
+                                                const void* object);
+void cmb_event_queue_print_formatted(cmb_sim* sim, FILE* out,
+                                     cmb_event_print_formatter* epf);
 void cmb_resource_print_report(cmb_sim* sim, const cmb_resource* r,
                                FILE* out);
 void cmb_resourcepool_print_report(cmb_sim* sim, const cmb_resourcepool* r,

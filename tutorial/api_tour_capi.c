@@ -79,6 +79,13 @@ static void tour_body(cmb_sim* sim, cmb_process* me, void* vctx) {
     CMB_PROC_END(sim, me);
 }
 
+static const char* tour_fmt(cmb_event_func* a, const void* s,
+                            const void* o) {
+    (void)a;
+    (void)o;
+    return s == (const void*)7 ? "noop(7)" : NULL;
+}
+
 /* ---- embedded resource guard: wait on cmb_resource_guard() with a
  * custom demand, woken by the release-driven guard signal ---- */
 struct rg_ctx {
@@ -145,6 +152,13 @@ static void run_trial(cmb_sim* sim, void* vtrl) {
     assert(rh != NULL);
     assert(cmb_timeseries_count(rh) == 3);
     assert(cmb_timeseries_min(rh) == 0.0 && cmb_timeseries_max(rh) == 1.0);
+    /* formatted event-queue dump */
+    uint64_t ev = cmb_event_schedule(sim, noop_event, (void*)7, NULL, 1e9, 0);
+    FILE* nul2 = fopen("/dev/null", "w");
+    cmb_event_queue_print_formatted(sim, nul2, tour_fmt);
+    fclose(nul2);
+    assert(cmb_event_cancel(sim, ev));
+
     /* pool guard handle + signal with no waiters = no grant */
     cmb_resourcepool* pl = cmb_resourcepool_create(sim);
     cmb_resourcepool_initialize(sim, pl, "RGPool", 4);
