@@ -20,6 +20,7 @@ SOURCES = [
     os.path.join(CSRC, "hip", "rng_kernel.hip"),
     os.path.join(CSRC, "hip", "awacs_kernel.hip"),
     os.path.join(CSRC, "hip", "multigpu.hip"),
+    os.path.join(CSRC, "hip", "terrain_kernel.hip"),
 ]
 
 

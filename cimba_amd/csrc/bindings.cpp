@@ -11,6 +11,7 @@
 #include "cimba/logger.hpp"
 #include "cimba/runner.hpp"
 #include "cimba/stats.hpp"
+#include "cimba/terrain.hpp"
 #include "../models/mm1.hpp"
 #include "../models/mg1.hpp"
 #include "../models/jobshop.hpp"
@@ -63,6 +64,19 @@ int cimba_xlane_repro(int iters, int device, int* out64);
 int cimba_mm1_multigpu_rccl(uint64_t ntrials, double arr_mean,
                             double srv_mean, uint64_t num_objects,
                             uint64_t seed, int ndev, double* out10);
+int cimba_terrain_gpu_build(int cols, int rows, double base, double amp,
+                            int octaves, uint64_t seed, int device,
+                            void** handle_out);
+int cimba_terrain_gpu_stats(void* handle, int cols, int rows, double out4[4]);
+int cimba_terrain_gpu_sample(void* handle, int cols, int rows, double base,
+                             double amp, int octaves, uint64_t seed,
+                             const float* xs, const float* ys, float* out,
+                             uint32_t n);
+int cimba_terrain_gpu_los(void* handle, int cols, int rows, double base,
+                          double amp, int octaves, uint64_t seed,
+                          const float* queries, uint8_t* vis, uint32_t nq,
+                          int nsteps, double* elapsed_ms);
+int cimba_terrain_gpu_free(void* handle);
 }
 
 using cmb_models::AWACS;
@@ -606,6 +620,96 @@ static uint64_t py_sfc64_raw(uint64_t seed, uint64_t skip) {
     return r.next();
 }
 
+// ---- terrain + LOS (reference tut_5_2.cu terrain stack; the host build
+// below is the fp32 numerics reference for the GPU kernels) ----
+
+static cmb::TerrainDesc terrain_desc_(int cols, int rows, double base,
+                                      double amp, int octaves,
+                                      uint64_t seed) {
+    return cmb::TerrainDesc{cols,          rows,        0.0f, 0.0f, 1.0f,
+                            1.0f,          (float)base, (float)amp,
+                            octaves,       seed};
+}
+
+static py::dict terrain_host(int cols, int rows, double base, double amp,
+                             int octaves, uint64_t seed,
+                             std::vector<float> xs, std::vector<float> ys,
+                             std::vector<float> queries, int nsteps) {
+    const cmb::TerrainDesc T = terrain_desc_(cols, rows, base, amp, octaves,
+                                             seed);
+    std::vector<float> h((size_t)cols * rows);
+    for (int r = 0; r < rows; ++r)
+        for (int c = 0; c < cols; ++c)
+            h[(size_t)r * cols + c] = cmb::th_texel_height(T, c, r);
+    double s1 = 0, s2 = 0, mn = 1e308, mx = -1e308;
+    for (float v : h) {
+        const double x = (double)v;
+        s1 += x;
+        s2 += x * x;
+        mn = x < mn ? x : mn;
+        mx = x > mx ? x : mx;
+    }
+    const double n = (double)h.size();
+    std::vector<float> samples(xs.size());
+    for (size_t i = 0; i < xs.size(); ++i)
+        samples[i] = cmb::th_sample(h.data(), T, xs[i], ys[i]);
+    std::vector<int> vis(queries.size() / 6);
+    for (size_t i = 0; i < vis.size(); ++i) {
+        const float* q = &queries[i * 6];
+        vis[i] = cmb::th_los_clear(h.data(), T, q[0], q[1], q[2], q[3], q[4],
+                                   q[5], nsteps)
+                     ? 1
+                     : 0;
+    }
+    py::dict d;
+    d["stats"] = std::vector<double>{s1 / n, s2 / n - (s1 / n) * (s1 / n),
+                                     mn, mx};
+    d["samples"] = samples;
+    d["vis"] = vis;
+    py::array_t<float> hm({rows, cols});
+    std::memcpy(hm.mutable_data(), h.data(), h.size() * sizeof(float));
+    d["heights"] = hm;
+    return d;
+}
+
+static py::dict terrain_gpu(int cols, int rows, double base, double amp,
+                            int octaves, uint64_t seed,
+                            std::vector<float> xs, std::vector<float> ys,
+                            std::vector<float> queries, int nsteps,
+                            int device) {
+    void* hdl = nullptr;
+    int rc = cimba_terrain_gpu_build(cols, rows, base, amp, octaves, seed,
+                                     device, &hdl);
+    if (rc != 0) throw std::runtime_error("terrain build failed");
+    double st[4];
+    rc = cimba_terrain_gpu_stats(hdl, cols, rows, st);
+    if (rc != 0) throw std::runtime_error("terrain stats failed");
+    std::vector<float> samples(xs.size());
+    if (!xs.empty()) {
+        rc = cimba_terrain_gpu_sample(hdl, cols, rows, base, amp, octaves,
+                                      seed, xs.data(), ys.data(),
+                                      samples.data(), (uint32_t)xs.size());
+        if (rc != 0) throw std::runtime_error("terrain sample failed");
+    }
+    const uint32_t nq = (uint32_t)(queries.size() / 6);
+    std::vector<uint8_t> v8(nq);
+    double los_ms = 0.0;
+    if (nq) {
+        rc = cimba_terrain_gpu_los(hdl, cols, rows, base, amp, octaves, seed,
+                                   queries.data(), v8.data(), nq, nsteps,
+                                   &los_ms);
+        if (rc != 0) throw std::runtime_error("terrain los failed");
+    }
+    cimba_terrain_gpu_free(hdl);
+    std::vector<int> vis(v8.begin(), v8.end());
+    py::dict d;
+    d["stats"] = std::vector<double>{st[0], st[1], st[2], st[3]};
+    d["samples"] = samples;
+    d["vis"] = vis;
+    d["los_ms"] = los_ms;
+    return d;
+}
+
 PYBIND11_MODULE(_C, m) {
     m.doc() = "cimba_amd native engine (MI355X / gfx950)";
 
@@ -771,6 +875,20 @@ PYBIND11_MODULE(_C, m) {
           py::arg("p0") = 0.0, py::arg("n") = 1 << 26, py::arg("seed") = 1ULL,
           py::arg("device") = 0);
     m.def("fmix64", &py_fmix64);
+    m.def("terrain_host", &terrain_host, py::arg("cols"), py::arg("rows"),
+          py::arg("base") = 0.0, py::arg("amp") = 1000.0,
+          py::arg("octaves") = 6, py::arg("seed") = 1ULL,
+          py::arg("xs") = std::vector<float>{},
+          py::arg("ys") = std::vector<float>{},
+          py::arg("queries") = std::vector<float>{},
+          py::arg("nsteps") = 128);
+    m.def("terrain_gpu", &terrain_gpu, py::arg("cols"), py::arg("rows"),
+          py::arg("base") = 0.0, py::arg("amp") = 1000.0,
+          py::arg("octaves") = 6, py::arg("seed") = 1ULL,
+          py::arg("xs") = std::vector<float>{},
+          py::arg("ys") = std::vector<float>{},
+          py::arg("queries") = std::vector<float>{},
+          py::arg("nsteps") = 128, py::arg("device") = 0);
     // seed-replay tooling (reference seed discipline):
     // any trial is reproducible from (master_seed, index)
     m.def("trial_seed", [](uint64_t master, uint64_t idx) {
