@@ -344,29 +344,32 @@ uint64_t cmb_event_pattern_count(cmb_sim* s, cmb_event_func* action,
 
 uint64_t cmb_event_pattern_cancel(cmb_sim* s, cmb_event_func* action,
                                   void* subject, void* object) {
+    // restart the scan after every removal: remove_at(i) refills slot i
+    // with the last heap entry and sift_up can carry that not-yet-examined
+    // entry ABOVE i, where a single ascending scan would miss it (same
+    // fix as HashHeap::pattern_cancel; caught by tests/test_hashheap.py)
     uint64_t cnt = 0;
     auto& q = s->E->evq;
-    int32_t i = 0;
-    while (i < q.n) {
-        bool match = false;
-        if (q.e[i].kind == EV_USER) {
+    for (;;) {
+        int32_t hit = -1;
+        for (int32_t i = 0; i < q.n; ++i) {
+            if (q.e[i].kind != EV_USER) continue;
             const auto& u = s->E->globals.uev[(int)q.e[i].b];
-            match = (!action || u.fn == action) &&
-                    (!subject || u.subj == subject) &&
-                    (!object || u.obj == object);
+            if ((!action || u.fn == action) &&
+                (!subject || u.subj == subject) &&
+                (!object || u.obj == object)) {
+                hit = i;
+                break;
+            }
         }
-        if (match) {
-            const uint32_t h = q.e[i].handle;
-            uev_free_slot(*s->E, (int)q.e[i].b);
-            q.remove_at(i);
-            if (s->E->n_event_waiters)
-                s->E->wake_event_waiters(h, SIG_CANCELLED);
-            ++cnt;
-        } else {
-            ++i;
-        }
+        if (hit < 0) return cnt;
+        const uint32_t h = q.e[hit].handle;
+        uev_free_slot(*s->E, (int)q.e[hit].b);
+        q.remove_at(hit);
+        if (s->E->n_event_waiters)
+            s->E->wake_event_waiters(h, SIG_CANCELLED);
+        ++cnt;
     }
-    return cnt;
 }
 
 void cmb_event_queue_execute(cmb_sim* s) {

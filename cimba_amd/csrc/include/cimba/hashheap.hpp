@@ -154,19 +154,25 @@ struct HashHeap {
 
     CMB_FORCEINLINE int32_t pattern_cancel(uint16_t kind, uint16_t a, bool match_b,
                                            uint64_t b) {
+        // restart the scan after every removal: remove_at(i) refills slot i
+        // with the last entry, and sift_up can carry that NOT-yet-examined
+        // entry above i where a single ascending scan would never revisit
+        // it (caught by the tests/test_hashheap.py fuzzer)
         int32_t cnt = 0;
-        int32_t i = 0;
-        while (i < n) {
-            if ((kind == 0xFFFF || e[i].kind == kind) &&
-                (a == 0xFFFF || e[i].a == a) && (!match_b || e[i].b == b)) {
-                remove_at(i);
-                ++cnt;
-                // re-examine slot i (a new entry moved in)
-            } else {
-                ++i;
+        for (;;) {
+            int32_t hit = -1;
+            for (int32_t i = 0; i < n; ++i) {
+                if ((kind == 0xFFFF || e[i].kind == kind) &&
+                    (a == 0xFFFF || e[i].a == a) &&
+                    (!match_b || e[i].b == b)) {
+                    hit = i;
+                    break;
+                }
             }
+            if (hit < 0) return cnt;
+            remove_at(hit);
+            ++cnt;
         }
-        return cnt;
     }
 
     CMB_FORCEINLINE uint32_t pattern_find(uint16_t kind, uint16_t a, bool match_b,
