@@ -9,9 +9,11 @@ Sized so the profiled kernel runs ~1 s: enough for stable averages,
 cheap on the GPU budget.  Counter summaries are committed under
 profiles/ (r01_lane_divergence.md).
 """
+import os
 import sys
 
-import cimba_amd as ca
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import cimba_amd as ca  # noqa: E402
 
 model = sys.argv[1] if len(sys.argv) > 1 else "mg1"
 if model == "mg1":

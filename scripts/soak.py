@@ -5,10 +5,12 @@ burn-ins recorded in profiles/r01_mm1_engine.md.
 
 Usage: python scripts/soak.py [seconds]
 """
+import os
 import sys
 import time
 
-import cimba_amd as ca
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import cimba_amd as ca  # noqa: E402
 
 budget = float(sys.argv[1]) if len(sys.argv) > 1 else 120.0
 t0 = time.time()
