@@ -86,3 +86,18 @@ def test_mg1_utilization_sweep(rho):
     tol = 0.06 if rho < 0.85 else 0.15  # heavier rho = slower convergence
     assert abs(r["avg_system_time"] - theory) / theory < tol, (
         rho, r["avg_system_time"], theory)
+
+
+def test_mg1_grid_script():
+    # the reference's showcase experiment (test_cimba.c grid) must run
+    # and PK-validate end-to-end on the host backend
+    import os
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "scripts", "mg1_grid.py"),
+         "--backend", "cpu", "--reps", "2"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    assert "worst PK error" in r.stdout
