@@ -79,3 +79,55 @@ def test_tsan_multithreaded_executive(tmp_path):
     out = subprocess.run([exe], capture_output=True, text=True, timeout=900,
                          env=env)
     assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-3000:])
+
+
+@pytest.mark.skipif(not os.path.exists(CLANGXX), reason="no clang++")
+def test_asan_prodcons_stress(tmp_path):
+    """ASan/UBSan over the randomized producer/consumer stress model —
+    blocking queue paths, guard grants and timeout wakes under the
+    multithreaded executive, with the sanitizer watching every
+    engine-state access."""
+    import re
+    pcf = open(os.path.join(ROOT, "tests", "test_prodcons_fuzz.py")).read()
+    HARNESS = re.search(r'HARNESS = r"""(.*?)"""', pcf, re.S).group(1)
+    src = tmp_path / "pcfuzz_asan.cpp"
+    src.write_text(HARNESS)
+    exe = str(tmp_path / "pcfuzz_asan")
+    r = subprocess.run(
+        [CLANGXX, "-std=c++17", "-O1", "-g",
+         "-fsanitize=address,undefined", "-fno-sanitize-recover=all",
+         "-I", os.path.join(ROOT, "cimba_amd", "csrc", "include"),
+         str(src),
+         os.path.join(ROOT, "cimba_amd", "csrc", "host", "support.cpp"),
+         "-o", exe, "-lpthread"], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr[-3000:]
+    env = dict(os.environ, ASAN_OPTIONS="detect_leaks=1:abort_on_error=1")
+    out = subprocess.run([exe, "7", "60"], capture_output=True, text=True,
+                         timeout=900, env=env)
+    assert out.returncode == 0, (out.stdout[-1000:], out.stderr[-3000:])
+    assert "prodcons fuzz OK" in out.stdout
+
+
+@pytest.mark.skipif(not os.path.exists(CLANGXX), reason="no clang++")
+def test_tsan_prodcons_stress(tmp_path):
+    """TSan over the same stress model running on all worker threads:
+    each worker owns a private engine, so the only shared state is the
+    work-claim counter and the report atomics — TSan proves it."""
+    import re
+    pcf = open(os.path.join(ROOT, "tests", "test_prodcons_fuzz.py")).read()
+    HARNESS = re.search(r'HARNESS = r"""(.*?)"""', pcf, re.S).group(1)
+    src = tmp_path / "pcfuzz_tsan.cpp"
+    src.write_text(HARNESS)
+    exe = str(tmp_path / "pcfuzz_tsan")
+    r = subprocess.run(
+        [CLANGXX, "-std=c++17", "-O1", "-g", "-fsanitize=thread",
+         "-I", os.path.join(ROOT, "cimba_amd", "csrc", "include"),
+         str(src),
+         os.path.join(ROOT, "cimba_amd", "csrc", "host", "support.cpp"),
+         "-o", exe, "-lpthread"], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr[-3000:]
+    env = dict(os.environ, TSAN_OPTIONS="halt_on_error=1")
+    out = subprocess.run([exe, "7", "60"], capture_output=True, text=True,
+                         timeout=900, env=env)
+    assert out.returncode == 0, (out.stdout[-1000:], out.stderr[-3000:])
+    assert "prodcons fuzz OK" in out.stdout
