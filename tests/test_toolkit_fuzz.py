@@ -70,7 +70,7 @@ int main(int argc, char** argv) {
     // guard stays empty and the no-queue-jump gate is always open
     for (int i = 0; i < 3; ++i) {
         const int pidx = eng.proc_alloc();
-        eng.proc_init(pidx, 0, 0);
+        eng.proc_init(pidx, 0, i - 1);  // priorities -1/0/1: preempt order
     }
     eng.queues[0].limit = 24;  // below QCAP: exercises the limit path
     eng.pqueues[0].limit = 24;
@@ -91,7 +91,7 @@ int main(int argc, char** argv) {
     for (uint64_t op = 0; op < nops; ++op) {
         const int who = (int)r.below(3);
         auto& p = eng.procs[who];
-        switch (r.below(10)) {
+        switch (r.below(11)) {
         case 0: {  // queue put
             const uint64_t v = next_val++;
             const bool ok = eng.q_try_put(0, p, v);
@@ -168,6 +168,35 @@ int main(int argc, char** argv) {
             const int32_t amt = 1 + (int32_t)r.below((uint32_t)mheld[who]);
             eng.pool_release(0, p, amt);
             mheld[who] -= amt;
+            break;
+        }
+        case 9: {  // pool preempt: all-or-nothing from lower-pri holders
+            const int32_t want = 1 + (int32_t)r.below(5);
+            const int32_t in_use = mheld[0] + mheld[1] + mheld[2];
+            const int32_t free_units = 10 - in_use;
+            int32_t reclaimable = 0;
+            for (int i = 0; i < 3; ++i)
+                if (i != who && (i - 1) < (who - 1))  // strictly lower pri
+                    reclaimable += mheld[i];
+            const bool expect = free_units + reclaimable >= want;
+            const bool ok = eng.pool_try_preempt(0, p, want);
+            CHECK(ok == expect);
+            if (ok) {
+                int32_t need = want - free_units;
+                while (need > 0) {  // lowest priority first, then index
+                    int victim = -1;
+                    for (int i = 0; i < 3; ++i) {
+                        if (i == who || mheld[i] <= 0) continue;
+                        if ((i - 1) >= (who - 1)) continue;
+                        if (victim < 0 || (i - 1) < (victim - 1)) victim = i;
+                    }
+                    const int32_t take =
+                        mheld[victim] < need ? mheld[victim] : need;
+                    mheld[victim] -= take;
+                    need -= take;
+                }
+                mheld[who] += want;
+            }
             break;
         }
         default: {  // invariants snapshot
