@@ -44,6 +44,14 @@ def build():
                   "power": dhex(r["sum_power"])}
     g["scenarios"] = {str(w): ca._C.scenario_host(w)["trace"]
                       for w in list(range(1, 14)) + [15, 16, 17, 18, 19]}
+    t = ca._C.terrain_host(64, 48, base=50.0, amp=400.0, octaves=5,
+                           seed=SEED, xs=[3.25, 40.0], ys=[7.5, 20.0],
+                           queries=[0, 0, 9000, 63, 47, 9000], nsteps=64)
+    g["terrain"] = {
+        "stats": [dhex(v) for v in t["stats"]],
+        "samples": [struct.pack("<f", v).hex() for v in t["samples"]],
+        "vis": t["vis"],
+    }
     return g
 
 
