@@ -43,12 +43,30 @@ def main():
                          "BASELINE configs 3-4")
     args = ap.parse_args()
 
+    # Self-launch: `python bench.py --gpus N` with no external launcher
+    # re-execs under torch.distributed.run, one rank per GPU over RCCL
+    # (VERDICT r01 item 1).  When the driver already launched us through
+    # torch.distributed.run, WORLD_SIZE is set and we skip this.
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        import socket
+        import subprocess
+
+        with socket.socket() as s:  # free rendezvous port on loopback
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        cmd = [sys.executable, "-m", "torch.distributed.run",
+               "--nnodes=1", f"--nproc-per-node={args.gpus}",
+               "--master-addr=127.0.0.1", f"--master-port={port}",
+               os.path.abspath(__file__)] + sys.argv[1:]
+        sys.exit(subprocess.call(cmd))
+
     import cimba_amd as ca
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    n_gpus = world if world > 1 else args.gpus
+    # n_gpus reports the ACTUAL world size, never the requested one
+    n_gpus = world
 
     dist = None
     torch = None
@@ -75,8 +93,10 @@ def main():
             ca.gpu_sync()
 
     def one_step(step_idx):
-        seed = ca.fmix64(((args.seed ^ (rank << 32)) + step_idx + 1)
-                         & 0xFFFFFFFFFFFFFFFF)
+        # same master seed on every rank; ranks are disjoint via trial_base
+        # (global trial indices), so the union over ranks is the same trial
+        # set a 1-rank run of world*trials would simulate
+        seed = ca.fmix64((args.seed + step_idx + 1) & 0xFFFFFFFFFFFFFFFF)
         ntrials = args.trials
         if args.model == "mm1":
             fn = ca.mm1_gpu if use_gpu else ca.mm1_host
@@ -97,6 +117,7 @@ def main():
             fn = ca._C.awacs_gpu if use_gpu else ca._C.awacs_host
             kw = dict(ntrials=ntrials, duration=min(args.objects / 250.0, 60.0),
                       ntargets=1000, seed=seed)
+        kw["trial_base"] = rank * ntrials
         if use_gpu:
             kw["device"] = local_rank
         else:

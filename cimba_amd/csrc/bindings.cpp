@@ -38,21 +38,25 @@ struct Mm1GpuOut {
     int32_t pad_;
 };
 int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
-                      uint64_t num_objects, uint64_t seed, int device,
-                      double until, uint64_t max_events, Mm1GpuOut* out);
+                      uint64_t num_objects, uint64_t seed,
+                      uint64_t trial_base, int device, double until,
+                      uint64_t max_events, Mm1GpuOut* out);
 int cimba_gpu_device_count(int* n);
 int cimba_gpu_sync(void);
 int cimba_scenario_gpu_run(int which, void* result_out);
 int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
-                      int device, double* elapsed_ms, void* results_out);
-int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
-                          int device, double* elapsed_ms, void* results_out);
+                      uint64_t trial_base, int device, double* elapsed_ms,
+                      void* results_out);
+int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params,
+                          uint64_t seed, uint64_t trial_base, int device,
+                          double* elapsed_ms, void* results_out);
 int cimba_sample_gpu(int dist, double p0, uint64_t n, uint64_t seed,
                      int device, double* host_out, double* elapsed_ms);
 int cimba_sample_moments_gpu(int dist, double p0, uint64_t n, uint64_t seed,
                              int device, double* out7, double* elapsed_ms);
 int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
-                        int device, double* elapsed_ms, void* results_out);
+                        uint64_t trial_base, int device, double* elapsed_ms,
+                        void* results_out);
 int cimba_awacs_power_test(const void* params, uint64_t seed, int device,
                            float* out_powers, int* nt_out);
 int cimba_awacs_power_test_devinit(const void* params, uint64_t seed,
@@ -134,20 +138,21 @@ static py::dict awacs_aggregate(const std::vector<AWACS::Result>& res,
 
 static py::dict awacs_host(uint64_t ntrials, double duration, double dwell,
                            double maneuver_mean, int ntargets, uint64_t seed,
-                           int threads) {
+                           int threads, uint64_t trial_base) {
     AWACS::Params p = make_awacs_params(duration, dwell, maneuver_mean,
                                         ntargets, 50000.0, 250.0, 2.0e15);
     std::vector<AWACS::Result> res(ntrials);
     {
         py::gil_scoped_release nogil;
-        run_host<AWACS>(p, seed, ntrials, threads, res.data());
+        run_host<AWACS>(p, seed, ntrials, threads, res.data(), {}, nullptr,
+                        trial_base);
     }
     return awacs_aggregate(res, -1.0);
 }
 
 static py::dict awacs_gpu(uint64_t ntrials, double duration, double dwell,
                           double maneuver_mean, int ntargets, uint64_t seed,
-                          int device) {
+                          int device, uint64_t trial_base) {
     AWACS::Params p = make_awacs_params(duration, dwell, maneuver_mean,
                                         ntargets, 50000.0, 250.0, 2.0e15);
     std::vector<AWACS::Result> res(ntrials);
@@ -155,7 +160,8 @@ static py::dict awacs_gpu(uint64_t ntrials, double duration, double dwell,
     int rc;
     {
         py::gil_scoped_release nogil;
-        rc = cimba_awacs_gpu_run(ntrials, &p, seed, device, &ms, res.data());
+        rc = cimba_awacs_gpu_run(ntrials, &p, seed, trial_base, device, &ms,
+                                 res.data());
     }
     if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
     return awacs_aggregate(res, ms);
@@ -306,26 +312,30 @@ static py::dict mg1_aggregate(const std::vector<MG1::Result>& res,
 
 static py::dict mg1_host(uint64_t ntrials, uint64_t num_objects,
                          double arr_rate, double srv_mean, double srv_scv,
-                         int dist, uint64_t seed, int threads) {
+                         int dist, uint64_t seed, int threads,
+                         uint64_t trial_base) {
     MG1::Params p{1.0 / arr_rate, srv_mean, srv_scv, num_objects, dist, 0};
     std::vector<MG1::Result> res(ntrials);
     {
         py::gil_scoped_release nogil;
-        run_host<MG1>(p, seed, ntrials, threads, res.data());
+        run_host<MG1>(p, seed, ntrials, threads, res.data(), {}, nullptr,
+                      trial_base);
     }
     return mg1_aggregate(res, -1.0);
 }
 
 static py::dict mg1_gpu(uint64_t ntrials, uint64_t num_objects,
                         double arr_rate, double srv_mean, double srv_scv,
-                        int dist, uint64_t seed, int device) {
+                        int dist, uint64_t seed, int device,
+                        uint64_t trial_base) {
     MG1::Params p{1.0 / arr_rate, srv_mean, srv_scv, num_objects, dist, 0};
     std::vector<MG1::Result> res(ntrials);
     double ms = 0.0;
     int rc;
     {
         py::gil_scoped_release nogil;
-        rc = cimba_mg1_gpu_run(ntrials, &p, seed, device, &ms, res.data());
+        rc = cimba_mg1_gpu_run(ntrials, &p, seed, trial_base, device, &ms,
+                               res.data());
     }
     if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
     return mg1_aggregate(res, ms);
@@ -379,25 +389,29 @@ static JobShop::Params make_jobshop_params(uint64_t entities, int njobs,
 }
 
 static py::dict jobshop_host(uint64_t ntrials, uint64_t entities, int njobs,
-                             double think_mean, uint64_t seed, int threads) {
+                             double think_mean, uint64_t seed, int threads,
+                             uint64_t trial_base) {
     JobShop::Params p = make_jobshop_params(entities, njobs, think_mean);
     std::vector<JobShop::Result> res(ntrials);
     {
         py::gil_scoped_release nogil;
-        run_host<JobShop>(p, seed, ntrials, threads, res.data());
+        run_host<JobShop>(p, seed, ntrials, threads, res.data(), {}, nullptr,
+                          trial_base);
     }
     return jobshop_aggregate(res, -1.0);
 }
 
 static py::dict jobshop_gpu(uint64_t ntrials, uint64_t entities, int njobs,
-                            double think_mean, uint64_t seed, int device) {
+                            double think_mean, uint64_t seed, int device,
+                            uint64_t trial_base) {
     JobShop::Params p = make_jobshop_params(entities, njobs, think_mean);
     std::vector<JobShop::Result> res(ntrials);
     double ms = 0.0;
     int rc;
     {
         py::gil_scoped_release nogil;
-        rc = cimba_jobshop_gpu_run(ntrials, &p, seed, device, &ms, res.data());
+        rc = cimba_jobshop_gpu_run(ntrials, &p, seed, trial_base, device, &ms,
+                                   res.data());
     }
     if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
     return jobshop_aggregate(res, ms);
@@ -458,12 +472,14 @@ static py::dict scenario_run_host(int which, uint64_t ntrials, int threads) {
 }
 
 static py::dict mm1_host(uint64_t ntrials, uint64_t num_objects, double arr_rate,
-                         double srv_rate, uint64_t seed, int threads) {
+                         double srv_rate, uint64_t seed, int threads,
+                         uint64_t trial_base) {
     MM1::Params p{1.0 / arr_rate, 1.0 / srv_rate, num_objects};
     std::vector<MM1::Result> res(ntrials);
     {
         py::gil_scoped_release nogil;
-        run_host<MM1>(p, seed, ntrials, threads, res.data());
+        run_host<MM1>(p, seed, ntrials, threads, res.data(), {}, nullptr,
+                      trial_base);
     }
     uint64_t ev = 0, objs = 0, ok = 0;
     double wait = 0.0;
@@ -491,13 +507,14 @@ static py::dict mm1_host(uint64_t ntrials, uint64_t num_objects, double arr_rate
 }
 
 static py::dict mm1_gpu(uint64_t ntrials, uint64_t num_objects, double arr_rate,
-                        double srv_rate, uint64_t seed, int device) {
+                        double srv_rate, uint64_t seed, int device,
+                        uint64_t trial_base) {
     Mm1GpuOut o;
     int rc;
     {
         py::gil_scoped_release nogil;
         rc = cimba_mm1_gpu_run(ntrials, 1.0 / arr_rate, 1.0 / srv_rate,
-                               num_objects, seed, device, 1.0e308,
+                               num_objects, seed, trial_base, device, 1.0e308,
                                UINT64_C(0xFFFFFFFFFFFFFFFF), &o);
     }
     if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
@@ -715,34 +732,40 @@ PYBIND11_MODULE(_C, m) {
 
     m.def("mm1_host", &mm1_host, py::arg("ntrials"), py::arg("num_objects"),
           py::arg("arr_rate") = 0.9, py::arg("srv_rate") = 1.0,
-          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("threads") = 0);
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("threads") = 0,
+          py::arg("trial_base") = 0);
     m.def("mm1_gpu", &mm1_gpu, py::arg("ntrials"), py::arg("num_objects"),
           py::arg("arr_rate") = 0.9, py::arg("srv_rate") = 1.0,
-          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0);
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0,
+          py::arg("trial_base") = 0);
     m.def("mg1_host", &mg1_host, py::arg("ntrials"), py::arg("num_objects"),
           py::arg("arr_rate") = 0.8, py::arg("srv_mean") = 1.0,
           py::arg("srv_scv") = 1.0, py::arg("dist") = 1,
-          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("threads") = 0);
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("threads") = 0,
+          py::arg("trial_base") = 0);
     m.def("mg1_gpu", &mg1_gpu, py::arg("ntrials"), py::arg("num_objects"),
           py::arg("arr_rate") = 0.8, py::arg("srv_mean") = 1.0,
           py::arg("srv_scv") = 1.0, py::arg("dist") = 1,
-          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0);
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0,
+          py::arg("trial_base") = 0);
     m.def("jobshop_host", &jobshop_host, py::arg("ntrials"),
           py::arg("entities") = 10000, py::arg("njobs") = 24,
           py::arg("think_mean") = 0.5, py::arg("seed") = 0x34f05c64d7ad598fULL,
-          py::arg("threads") = 0);
+          py::arg("threads") = 0, py::arg("trial_base") = 0);
     m.def("jobshop_gpu", &jobshop_gpu, py::arg("ntrials"),
           py::arg("entities") = 10000, py::arg("njobs") = 24,
           py::arg("think_mean") = 0.5, py::arg("seed") = 0x34f05c64d7ad598fULL,
-          py::arg("device") = 0);
+          py::arg("device") = 0, py::arg("trial_base") = 0);
     m.def("awacs_host", &awacs_host, py::arg("ntrials"),
           py::arg("duration") = 60.0, py::arg("dwell") = 0.04,
           py::arg("maneuver_mean") = 5.0, py::arg("ntargets") = 1000,
-          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("threads") = 0);
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("threads") = 0,
+          py::arg("trial_base") = 0);
     m.def("awacs_gpu", &awacs_gpu, py::arg("ntrials"),
           py::arg("duration") = 60.0, py::arg("dwell") = 0.04,
           py::arg("maneuver_mean") = 5.0, py::arg("ntargets") = 1000,
-          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0);
+          py::arg("seed") = 0x34f05c64d7ad598fULL, py::arg("device") = 0,
+          py::arg("trial_base") = 0);
     m.def("xlane_repro", [](int iters, int device) {
         std::vector<int> o(64);
         int rc = cimba_xlane_repro(iters, device, o.data());

@@ -142,9 +142,11 @@ __device__ __noinline__ int engine_phase(EngA& E, int lane) {
 __device__ __noinline__ void engine_init_phase(EngA& E, int lane,
                                                const AWACS::Params* dP,
                                                uint64_t master_seed,
+                                               uint64_t trial_base,
                                                uint32_t trial) {
     if (lane != 0) return;
-    E.init(dP, cmb::trial_seed(master_seed, trial), trial);
+    E.init(dP, cmb::trial_seed(master_seed, trial_base + trial),
+           (uint32_t)(trial_base + trial));
     AWACS::setup(E);
 }
 
@@ -161,7 +163,7 @@ __device__ __noinline__ void engine_finish_phase(EngA& E, int lane,
 
 __global__ __launch_bounds__(256) void awacs_kernel(
     const AWACS::Params* __restrict__ dP, uint64_t master_seed,
-    uint32_t ntrials, AWACS::Result* __restrict__ out,
+    uint64_t trial_base, uint32_t ntrials, AWACS::Result* __restrict__ out,
     StA* __restrict__ stores, float* __restrict__ dbg, int scalar_phys) {
     const int lane = (int)(threadIdx.x & 63);
     const uint32_t wslot =
@@ -173,7 +175,7 @@ __global__ __launch_bounds__(256) void awacs_kernel(
 
     EngA E(stores[wslot]);  // lane 0's context is the live engine
     for (uint32_t trial = wslot; trial < ntrials; trial += nwaves) {
-        engine_init_phase(E, lane, dP, master_seed, trial);
+        engine_init_phase(E, lane, dP, master_seed, trial_base, trial);
         for (;;) {
             const int phase_word =
                 __builtin_amdgcn_readfirstlane(engine_phase(E, lane));
@@ -309,7 +311,8 @@ int cimba_xlane_repro(int iters, int device, int* out64) {
 }
 
 int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
-                        int device, double* elapsed_ms, void* results_out) {
+                        uint64_t trial_base, int device, double* elapsed_ms,
+                        void* results_out) {
     HIP_TRY(hipSetDevice(device));
     const AWACS::Params& P = *(const AWACS::Params*)params;
     const uint32_t want_waves = (uint32_t)ntrials;
@@ -330,7 +333,7 @@ int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
     HIP_TRY(hipEventRecord(t0));
     const char* sp = getenv("CIMBA_AWACS_SCALAR");
     hipLaunchKernelGGL(awacs_kernel, dim3(blocks), dim3(256), 0, 0, d_P,
-                       seed, (uint32_t)ntrials, d_out, d_eng,
+                       seed, trial_base, (uint32_t)ntrials, d_out, d_eng,
                        (float*)nullptr, sp ? atoi(sp) : 0);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
@@ -364,7 +367,7 @@ int cimba_awacs_first_dwell_dbg(const void* params, uint64_t master_seed,
     HIP_TRY(hipMalloc(&d_dbg, sizeof(float) * AWACS::MAX_T));
     HIP_TRY(hipMemset(d_dbg, 0, sizeof(float) * AWACS::MAX_T));
     hipLaunchKernelGGL(awacs_kernel, dim3(1), dim3(256), 0, 0, d_P,
-                       master_seed, 1u, d_out, d_eng, d_dbg, 0);
+                       master_seed, 0ull, 1u, d_out, d_eng, d_dbg, 0);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipMemcpy(out_powers, d_dbg, sizeof(float) * AWACS::MAX_T,
                       hipMemcpyDeviceToHost));

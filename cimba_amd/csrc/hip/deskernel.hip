@@ -31,8 +31,9 @@ using cmb_models::Scenario;
 // MINW = requested waves/SIMD (caps the register allocator; occupancy knob)
 template <class Model, int WPB, int MINW = 1>
 __global__ __launch_bounds__(WPB * 64, MINW) __attribute__((flatten)) void trial_kernel(
-    typename Model::Params P, uint64_t master_seed, uint32_t ntrials,
-    double until, uint64_t max_events, typename Model::Result* __restrict__ out) {
+    typename Model::Params P, uint64_t master_seed, uint64_t trial_base,
+    uint32_t ntrials, double until, uint64_t max_events,
+    typename Model::Result* __restrict__ out) {
     // arrays in LDS; the Engine context (clock, seq, handles, status, RNG
     // state, heap size) is a per-lane local -> register-resident
     __shared__ typename Engine<Model>::Storage st[WPB];
@@ -42,7 +43,8 @@ __global__ __launch_bounds__(WPB * 64, MINW) __attribute__((flatten)) void trial
     const uint32_t stride = gridDim.x * WPB;
     for (uint32_t trial = blockIdx.x * WPB + (uint32_t)w; trial < ntrials;
          trial += stride) {
-        E.init(&P, cmb::trial_seed(master_seed, trial), trial);
+        E.init(&P, cmb::trial_seed(master_seed, trial_base + trial),
+               (uint32_t)(trial_base + trial));
         Model::setup(E);
         E.run(until, max_events);
         Model::finish(E, out[trial]);
@@ -64,8 +66,9 @@ static_assert(sizeof(Engine<JobShop>::Storage) * 4 < 64 * 1024,
 // host-side launcher: upload params, launch, copy per-trial results back
 template <class Model, int WPB, int MINW = 1>
 int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
-                   uint64_t seed, double until, uint64_t max_events,
-                   double* elapsed_ms, typename Model::Result* host_out) {
+                   uint64_t seed, uint64_t trial_base, double until,
+                   uint64_t max_events, double* elapsed_ms,
+                   typename Model::Result* host_out) {
     using Result = typename Model::Result;
     Result* d_out = nullptr;
     HIP_TRY(hipMalloc(&d_out, sizeof(Result) * ntrials));
@@ -77,8 +80,8 @@ int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
     hipLaunchKernelGGL((trial_kernel<Model, WPB, MINW>), dim3(grid),
-                       dim3(WPB * 64), 0, 0, P, seed, (uint32_t)ntrials,
-                       until, max_events, d_out);
+                       dim3(WPB * 64), 0, 0, P, seed, trial_base,
+                       (uint32_t)ntrials, until, max_events, d_out);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -101,14 +104,16 @@ int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
 // stream.  Measured A/B against the wave-per-trial kernel in profiles/.
 template <class Model, int MINW = 1>
 __global__ __launch_bounds__(256, MINW) __attribute__((flatten)) void lane_trial_kernel(
-    typename Model::Params P, uint64_t master_seed, uint32_t ntrials,
-    double until, uint64_t max_events, typename Model::Result* __restrict__ out,
+    typename Model::Params P, uint64_t master_seed, uint64_t trial_base,
+    uint32_t ntrials, double until, uint64_t max_events,
+    typename Model::Result* __restrict__ out,
     typename Engine<Model>::Storage* __restrict__ stores) {
     const uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
     const uint32_t stride = gridDim.x * blockDim.x;
     Engine<Model> E(stores[gid]);
     for (uint32_t trial = gid; trial < ntrials; trial += stride) {
-        E.init(&P, cmb::trial_seed(master_seed, trial), trial);
+        E.init(&P, cmb::trial_seed(master_seed, trial_base + trial),
+               (uint32_t)(trial_base + trial));
         Model::setup(E);
         E.run(until, max_events);
         Model::finish(E, out[trial]);
@@ -121,15 +126,16 @@ __global__ __launch_bounds__(256, MINW) __attribute__((flatten)) void lane_trial
 // into wide transactions: the AoS->SoA transposition for free.
 template <class Model, int MINW = 1>
 __global__ __launch_bounds__(256, MINW) void lane_scratch_kernel(
-    typename Model::Params P, uint64_t master_seed, uint32_t ntrials,
-    double until, uint64_t max_events,
+    typename Model::Params P, uint64_t master_seed, uint64_t trial_base,
+    uint32_t ntrials, double until, uint64_t max_events,
     typename Model::Result* __restrict__ out) {
     const uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
     const uint32_t stride = gridDim.x * blockDim.x;
     typename Engine<Model>::Storage st;  // per-lane scratch (HW-swizzled)
     Engine<Model> E(st);
     for (uint32_t trial = gid; trial < ntrials; trial += stride) {
-        E.init(&P, cmb::trial_seed(master_seed, trial), trial);
+        E.init(&P, cmb::trial_seed(master_seed, trial_base + trial),
+               (uint32_t)(trial_base + trial));
         Model::setup(E);
         E.run(until, max_events);
         Model::finish(E, out[trial]);
@@ -138,7 +144,8 @@ __global__ __launch_bounds__(256, MINW) void lane_scratch_kernel(
 
 template <class Model, int MINW>
 int run_trials_gpu_lane_scratch(const typename Model::Params& P,
-                                uint64_t ntrials, uint64_t seed, double until,
+                                uint64_t ntrials, uint64_t seed,
+                                uint64_t trial_base, double until,
                                 uint64_t max_events, double* elapsed_ms,
                                 typename Model::Result* host_out,
                                 uint32_t blocks) {
@@ -152,8 +159,8 @@ int run_trials_gpu_lane_scratch(const typename Model::Params& P,
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
     hipLaunchKernelGGL((lane_scratch_kernel<Model, MINW>), dim3(grid),
-                       dim3(256), 0, 0, P, seed, (uint32_t)ntrials, until,
-                       max_events, d_out);
+                       dim3(256), 0, 0, P, seed, trial_base,
+                       (uint32_t)ntrials, until, max_events, d_out);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -170,9 +177,9 @@ int run_trials_gpu_lane_scratch(const typename Model::Params& P,
 
 template <class Model, int MINW>
 int run_trials_gpu_lane(const typename Model::Params& P, uint64_t ntrials,
-                        uint64_t seed, double until, uint64_t max_events,
-                        double* elapsed_ms, typename Model::Result* host_out,
-                        uint32_t blocks) {
+                        uint64_t seed, uint64_t trial_base, double until,
+                        uint64_t max_events, double* elapsed_ms,
+                        typename Model::Result* host_out, uint32_t blocks) {
     using Result = typename Model::Result;
     using St = typename Engine<Model>::Storage;
     const uint32_t want = (uint32_t)((ntrials + 255) / 256);
@@ -186,8 +193,8 @@ int run_trials_gpu_lane(const typename Model::Params& P, uint64_t ntrials,
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
     hipLaunchKernelGGL((lane_trial_kernel<Model, MINW>), dim3(grid),
-                       dim3(256), 0, 0, P, seed, (uint32_t)ntrials, until,
-                       max_events, d_out, d_st);
+                       dim3(256), 0, 0, P, seed, trial_base,
+                       (uint32_t)ntrials, until, max_events, d_out, d_st);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -231,8 +238,9 @@ struct Mm1GpuOut {
 };
 
 int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
-                      uint64_t num_objects, uint64_t seed, int device,
-                      double until, uint64_t max_events, Mm1GpuOut* out) {
+                      uint64_t num_objects, uint64_t seed,
+                      uint64_t trial_base, int device, double until,
+                      uint64_t max_events, Mm1GpuOut* out) {
     HIP_TRY(hipSetDevice(device));
     MM1::Params P{arr_mean, srv_mean, num_objects};
     std::vector<MM1::Result> res(ntrials);
@@ -247,11 +255,11 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
         const char* lb = getenv("CIMBA_MM1_LANE_BLOCKS");
         const uint32_t blocks = lb ? (uint32_t)atoi(lb) : 2048u;
         if (lane && atoi(lane) == 1)  // explicit HBM-lane variant
-            rc = run_trials_gpu_lane<MM1, 1>(P, ntrials, seed, until,
+            rc = run_trials_gpu_lane<MM1, 1>(P, ntrials, seed, trial_base, until,
                                              max_events, &out->elapsed_ms,
                                              res.data(), blocks);
         else  // default: scratch storage (HW lane-interleaved -> coalesced)
-            rc = run_trials_gpu_lane_scratch<MM1, 1>(P, ntrials, seed, until,
+            rc = run_trials_gpu_lane_scratch<MM1, 1>(P, ntrials, seed, trial_base, until,
                                                      max_events,
                                                      &out->elapsed_ms,
                                                      res.data(), blocks);
@@ -259,17 +267,17 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
     }
     switch (minw) {
         case 5:
-            rc = run_trials_gpu<MM1, 4, 5>(P, ntrials, seed, until,
+            rc = run_trials_gpu<MM1, 4, 5>(P, ntrials, seed, trial_base, until,
                                            max_events, &out->elapsed_ms,
                                            res.data());
             break;
         case 6:
-            rc = run_trials_gpu<MM1, 4, 6>(P, ntrials, seed, until,
+            rc = run_trials_gpu<MM1, 4, 6>(P, ntrials, seed, trial_base, until,
                                            max_events, &out->elapsed_ms,
                                            res.data());
             break;
         default:
-            rc = run_trials_gpu<MM1, 4, 4>(P, ntrials, seed, until,
+            rc = run_trials_gpu<MM1, 4, 4>(P, ntrials, seed, trial_base, until,
                                            max_events, &out->elapsed_ms,
                                            res.data());
     }
@@ -294,36 +302,38 @@ aggregate:
 
 // MG1: results array provided by caller (per-trial)
 int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
-                      int device, double* elapsed_ms, void* results_out) {
+                      uint64_t trial_base, int device, double* elapsed_ms,
+                      void* results_out) {
     HIP_TRY(hipSetDevice(device));
     const char* lane = getenv("CIMBA_MG1_LANE");
     const uint64_t nt_ = ntrials;
     if (lane ? atoi(lane) != 0 : nt_ >= 32768)
         return run_trials_gpu_lane_scratch<MG1, 1>(
-            *(const MG1::Params*)params, ntrials, seed, 1.0e308,
+            *(const MG1::Params*)params, ntrials, seed, trial_base, 1.0e308,
             UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
             (MG1::Result*)results_out, 2048u);
     const char* mw = getenv("CIMBA_MG1_MINW");
     const int minw = mw ? atoi(mw) : 4;
     if (minw >= 4)
         return run_trials_gpu<MG1, 4, 4>(*(const MG1::Params*)params, ntrials,
-                                         seed, 1.0e308,
+                                         seed, trial_base, 1.0e308,
                                          UINT64_C(0xFFFFFFFFFFFFFFFF),
                                          elapsed_ms,
                                          (MG1::Result*)results_out);
     if (minw == 3)
         return run_trials_gpu<MG1, 4, 3>(*(const MG1::Params*)params, ntrials,
-                                         seed, 1.0e308,
+                                         seed, trial_base, 1.0e308,
                                          UINT64_C(0xFFFFFFFFFFFFFFFF),
                                          elapsed_ms,
                                          (MG1::Result*)results_out);
-    return run_trials_gpu<MG1, 4>(*(const MG1::Params*)params, ntrials, seed,
+    return run_trials_gpu<MG1, 4>(*(const MG1::Params*)params, ntrials, seed, trial_base,
                                   1.0e308, UINT64_C(0xFFFFFFFFFFFFFFFF),
                                   elapsed_ms, (MG1::Result*)results_out);
 }
 
-int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
-                          int device, double* elapsed_ms, void* results_out) {
+int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params,
+                          uint64_t seed, uint64_t trial_base, int device,
+                          double* elapsed_ms, void* results_out) {
     HIP_TRY(hipSetDevice(device));
     const char* lane = getenv("CIMBA_JS_LANE");
     if (lane ? atoi(lane) != 0 : ntrials >= 32768) {
@@ -333,16 +343,16 @@ int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
         const int lminw = lmw ? atoi(lmw) : 1;
         if (lminw >= 3)
             return run_trials_gpu_lane<JobShop, 3>(
-                *(const JobShop::Params*)params, ntrials, seed, 1.0e308,
+                *(const JobShop::Params*)params, ntrials, seed, trial_base, 1.0e308,
                 UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
                 (JobShop::Result*)results_out, 2048u);
         if (lminw == 2)
             return run_trials_gpu_lane<JobShop, 2>(
-                *(const JobShop::Params*)params, ntrials, seed, 1.0e308,
+                *(const JobShop::Params*)params, ntrials, seed, trial_base, 1.0e308,
                 UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
                 (JobShop::Result*)results_out, 2048u);
         return run_trials_gpu_lane<JobShop, 1>(
-            *(const JobShop::Params*)params, ntrials, seed, 1.0e308,
+            *(const JobShop::Params*)params, ntrials, seed, trial_base, 1.0e308,
             UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
             (JobShop::Result*)results_out, 2048u);
     }
@@ -350,11 +360,11 @@ int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
     const int minw = mw ? atoi(mw) : 4;
     if (minw >= 4)
         return run_trials_gpu<JobShop, 4, 4>(
-            *(const JobShop::Params*)params, ntrials, seed, 1.0e308,
+            *(const JobShop::Params*)params, ntrials, seed, trial_base, 1.0e308,
             UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
             (JobShop::Result*)results_out);
     return run_trials_gpu<JobShop, 4>(*(const JobShop::Params*)params, ntrials,
-                                      seed, 1.0e308,
+                                      seed, trial_base, 1.0e308,
                                       UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
                                       (JobShop::Result*)results_out);
 }

@@ -46,11 +46,15 @@ struct RunReport {
     uint64_t abandoned = 0;  // subset of failed: TrialAbandon thrown
 };
 
+// `trial_base` offsets the GLOBAL trial index: seeds derive from
+// trial_base + t, so a run sharded across ranks/devices simulates the
+// identical trial set as a single-device run of the same total — any
+// world size reproduces bit-identical per-trial streams.
 template <class Model>
 RunReport run_host(const typename Model::Params& params, uint64_t master_seed,
                    uint64_t ntrials, int nthreads,
                    typename Model::Result* out, RunLimits limits = {},
-                   const RunHooks* hooks = nullptr) {
+                   const RunHooks* hooks = nullptr, uint64_t trial_base = 0) {
     if (nthreads <= 0) {
         nthreads = (int)std::thread::hardware_concurrency();
         if (nthreads <= 0) nthreads = 1;
@@ -70,11 +74,12 @@ RunReport run_host(const typename Model::Params& params, uint64_t master_seed,
         for (;;) {
             const uint64_t t = next.fetch_add(1, std::memory_order_relaxed);
             if (t >= ntrials) break;
-            const uint64_t seed = trial_seed(master_seed, t);
-            logger_ctx().trial = (uint32_t)t;
+            const uint64_t gt = trial_base + t;  // global trial index
+            const uint64_t seed = trial_seed(master_seed, gt);
+            logger_ctx().trial = (uint32_t)gt;
             logger_ctx().seed = seed;
             logger_ctx().sim_time = 0.0;
-            eng->init(&params, seed, (uint32_t)t);
+            eng->init(&params, seed, (uint32_t)gt);
             try {
                 Model::setup(*eng);
                 eng->run(limits.until, limits.max_events);

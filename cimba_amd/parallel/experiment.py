@@ -82,12 +82,14 @@ def run_distributed_mm1(ntrials_total, num_objects, seed, use_gpu=None,
     if use_gpu is None:
         use_gpu = gpu_device_count() > 0
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    # trial_base seeds each trial by its GLOBAL index, so any world size
+    # simulates the identical trial set (bit-identical per-trial streams)
     if use_gpu:
         r = mm1_gpu(ntrials=hi - lo, num_objects=num_objects,
-                    seed=seed + lo, device=local_rank)
+                    seed=seed, device=local_rank, trial_base=lo)
     else:
         r = mm1_host(ntrials=hi - lo, num_objects=num_objects,
-                     seed=seed + lo, threads=0)
+                     seed=seed, threads=0, trial_base=lo)
 
     ds = DataSummary()
     if "per_trial_avg" in r:

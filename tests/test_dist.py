@@ -117,3 +117,43 @@ def test_run_distributed_mm1_gloo():
     assert res["n"] == 8          # all shards merged
     assert 7.0 < res["mean"] < 13.0
     assert res["events"] > 8 * 3000
+
+
+def test_trial_base_world_invariance():
+    """Sharding by trial_base simulates the identical global trial set:
+    two half-shards reproduce the full run bit-exactly (ADVICE r01 #1)."""
+    import cimba_amd as ca
+
+    full = ca.mm1_host(ntrials=16, num_objects=500, seed=1234, threads=1)
+    a = ca.mm1_host(ntrials=8, num_objects=500, seed=1234, threads=1,
+                    trial_base=0)
+    b = ca.mm1_host(ntrials=8, num_objects=500, seed=1234, threads=1,
+                    trial_base=8)
+    assert a["total_events"] + b["total_events"] == full["total_events"]
+    assert a["total_objects"] + b["total_objects"] == full["total_objects"]
+    assert (a["per_trial_avg"] + b["per_trial_avg"]) == full["per_trial_avg"]
+
+
+def test_bench_self_launch_host():
+    """`python bench.py --gpus 2 --host` must self-launch 2 ranks (no
+    external torch.distributed.run) and report the ACTUAL world size
+    (VERDICT r01 item 1)."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.pop("WORLD_SIZE", None)
+    env.pop("RANK", None)
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "bench.py"), "--gpus", "2",
+         "--host", "--steps", "1", "--warmup", "0", "--trials", "8",
+         "--objects", "200"],
+        capture_output=True, text=True, timeout=600, env=env, cwd=root)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    rep = json.loads(line)
+    assert rep["n_gpus"] == 2
+    assert rep["value"] > 0
+    assert rep["config"]["parallelism"] == "trial-parallel dp2"
