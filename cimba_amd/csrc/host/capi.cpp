@@ -335,8 +335,9 @@ uint64_t cmb_event_pattern_count(cmb_sim* s, cmb_event_func* action,
     for (int32_t i = 0; i < q.n; ++i) {
         if (q.e[i].kind != EV_USER) continue;
         const auto& u = s->E->globals.uev[(int)q.e[i].b];
-        if ((!action || u.fn == action) && (!subject || u.subj == subject) &&
-            (!object || u.obj == object))
+        if ((action == CMB_ANY_ACTION || u.fn == action) &&
+            (subject == CMB_ANY_SUBJECT || u.subj == subject) &&
+            (object == CMB_ANY_OBJECT || u.obj == object))
             ++cnt;
     }
     return cnt;
@@ -355,9 +356,9 @@ uint64_t cmb_event_pattern_cancel(cmb_sim* s, cmb_event_func* action,
         for (int32_t i = 0; i < q.n; ++i) {
             if (q.e[i].kind != EV_USER) continue;
             const auto& u = s->E->globals.uev[(int)q.e[i].b];
-            if ((!action || u.fn == action) &&
-                (!subject || u.subj == subject) &&
-                (!object || u.obj == object)) {
+            if ((action == CMB_ANY_ACTION || u.fn == action) &&
+                (subject == CMB_ANY_SUBJECT || u.subj == subject) &&
+                (object == CMB_ANY_OBJECT || u.obj == object)) {
                 hit = i;
                 break;
             }
@@ -416,8 +417,9 @@ uint64_t cmb_event_pattern_find(cmb_sim* s, cmb_event_func* action,
     for (int32_t i = 0; i < q.n; ++i) {
         if (q.e[i].kind != EV_USER) continue;
         const auto& u = s->E->globals.uev[(int)q.e[i].b];
-        if ((!action || u.fn == action) && (!subject || u.subj == subject) &&
-            (!object || u.obj == object))
+        if ((action == CMB_ANY_ACTION || u.fn == action) &&
+            (subject == CMB_ANY_SUBJECT || u.subj == subject) &&
+            (object == CMB_ANY_OBJECT || u.obj == object))
             return q.e[i].handle;
     }
     return 0;
@@ -528,7 +530,7 @@ void* cmb_process_exit_value(const cmb_sim* s, const cmb_process* p) {
 }
 bool cmb_process_timer_set(cmb_sim* s, cmb_process* p, int slot,
                            double delay, int64_t sig) {
-    if (slot < 0 || slot >= CModel::Cfg::TIMERS) return false;
+    if (slot < 1 || slot >= CModel::Cfg::TIMERS) return false;  // 0 reserved
     s->E->timer_cancel(s->E->procs[dec(p)], slot);
     return s->E->timer_add(s->E->procs[dec(p)], slot, delay, sig);
 }
@@ -562,20 +564,20 @@ void cmb_timer_disarm_(cmb_sim* s, cmb_process* p) {
 }
 bool cmb_process_timer_add(cmb_sim* s, cmb_process* p, int slot,
                            double delay, int64_t sig) {
-    if (slot < 0 || slot >= CModel::Cfg::TIMERS) return false;
+    if (slot < 1 || slot >= CModel::Cfg::TIMERS) return false;  // 0 reserved
     return s->E->timer_add(s->E->procs[dec(p)], slot, delay, sig);
 }
 void cmb_process_timer_cancel(cmb_sim* s, cmb_process* p, int slot) {
-    if (slot < 0 || slot >= CModel::Cfg::TIMERS) return;
+    if (slot < 1 || slot >= CModel::Cfg::TIMERS) return;  // 0 reserved
     s->E->timer_cancel(s->E->procs[dec(p)], slot);
 }
 void cmb_process_timer_clear(cmb_sim* s, cmb_process* p) {
-    for (int t = 0; t < CModel::Cfg::TIMERS; ++t)
+    for (int t = 1; t < CModel::Cfg::TIMERS; ++t)  // slot 0 reserved
         s->E->timer_cancel(s->E->procs[dec(p)], t);
 }
 bool cmb_process_timer_pending(const cmb_sim* s, const cmb_process* p,
                                int slot) {
-    if (slot < 0 || slot >= CModel::Cfg::TIMERS) return false;
+    if (slot < 1 || slot >= CModel::Cfg::TIMERS) return false;  // 0 reserved
     return s->E->procs[dec(p)].timers[slot] != 0;
 }
 int cmb_sim_ok_(const cmb_sim* s) { return s->E->status == ST_OK; }

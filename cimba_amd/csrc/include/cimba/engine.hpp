@@ -119,7 +119,9 @@ struct ProcRec {
     uint8_t pad0_;
     uint16_t pad1_;
     uint32_t demand_ctx;
-    uint32_t wseq;      // FIFO tie-break within a guard
+    uint64_t wseq;      // FIFO tie-break within a guard; 64-bit so an
+                        // EV_GRANT can never alias a new wait on the same
+                        // guard, whatever the trial length
     double entry_t;     // guard entry time (ordering: pri desc, entry asc, seq asc)
     uint32_t timers[NT];  // pending timer event handles; 0 = free slot
     int16_t waiters_head;  // procs waiting for me to finish (wait_process)
@@ -641,10 +643,10 @@ struct Engine {
         p.demand_kind = demand;
         p.demand_ctx = ctx;
         p.entry_t = now;
-        p.wseq = (uint32_t)seq++;
+        p.wseq = seq++;
         p.gnext = guards[gid].head;
         guards[gid].head = (int16_t)pidx_of(&p);
-        await_setup(p, AW_GUARD, (uint32_t)p.wseq);
+        await_setup(p, AW_GUARD, (uint32_t)p.wseq);  // key is informational
     }
 
     CMB_FORCEINLINE void guard_unlink(ProcT& p) {
@@ -708,7 +710,7 @@ struct Engine {
         bool granted = false;
         if (w >= 0 && eval_demand(procs[w])) {
             ProcT& p = procs[w];
-            schedule(EV_GRANT, (uint16_t)w, (uint32_t)gid, (uint64_t)p.wseq, now,
+            schedule(EV_GRANT, (uint16_t)w, (uint32_t)gid, p.wseq, now,
                      p.priority);
             granted = true;
         }
@@ -725,7 +727,7 @@ struct Engine {
         for (int16_t i = guards[gid].head; i >= 0; i = procs[i].gnext) {
             if (eval_demand(procs[i])) {
                 ProcT& p = procs[i];
-                schedule(EV_GRANT, (uint16_t)i, (uint32_t)gid, (uint64_t)p.wseq,
+                schedule(EV_GRANT, (uint16_t)i, (uint32_t)gid, p.wseq,
                          now, p.priority);
                 ++cnt;
             }
@@ -1011,7 +1013,7 @@ struct Engine {
             case EV_GRANT: {
                 ProcT& p = procs[ev.a];
                 if (p.state == PS_RUNNING && p.await_kind == AW_GUARD &&
-                    p.wseq == (uint32_t)ev.b && p.gid == (int16_t)ev.c) {
+                    p.wseq == ev.b && p.gid == (int16_t)ev.c) {
                     guard_unlink(p);
                     p.await_kind = AW_NONE;
                     p.g_granted = 1;

@@ -8,7 +8,7 @@
 // A replication completes when TOTAL_ENTITIES jobs have finished their
 // route.  Exercises pool contention, partial acquisition and the guard
 // priority ordering under load; throughput/utilization sanity is checked
-// host-side (tests/test_jobshop.py).
+// host-side (tests/test_mg1_jobshop.py).
 #pragma once
 
 #include "../include/cimba/engine.hpp"

@@ -99,7 +99,13 @@ uint64_t cmb_event_schedule(cmb_sim* sim, cmb_event_func* action,
 bool cmb_event_cancel(cmb_sim* sim, uint64_t handle);
 bool cmb_event_reschedule(cmb_sim* sim, uint64_t handle, double time,
                           int priority);
-/* wildcard pattern ops; pass NULL action/subject/object as ANY */
+/* wildcard pattern ops.  Pass CMB_ANY_ACTION / CMB_ANY_SUBJECT /
+ * CMB_ANY_OBJECT to match any value in that position (reference
+ * include/cmb_event.h wildcard contract); NULL is a CONCRETE value and
+ * matches only events scheduled with a NULL in that position. */
+#define CMB_ANY_ACTION ((cmb_event_func*)~(uintptr_t)0)
+#define CMB_ANY_SUBJECT ((void*)~(uintptr_t)0)
+#define CMB_ANY_OBJECT ((void*)~(uintptr_t)0)
 uint64_t cmb_event_pattern_count(cmb_sim* sim, cmb_event_func* action,
                                  void* subject, void* object);
 uint64_t cmb_event_pattern_cancel(cmb_sim* sim, cmb_event_func* action,
@@ -174,8 +180,11 @@ void cmb_timer_disarm_(cmb_sim* sim, cmb_process* p);
 /* ---- multiple concurrent timers per process (reference
  * cmb_process_timer_add/set/cancel/clear, cmb_process.c:514-580).
  * Slot 0 is reserved for the blocking-call timeout (cmb_timer_arm_);
- * user slots are 1..CMB_PROCESS_TIMERS-1.  A firing timer wakes the
- * process's current blocking call with `sig`. ---- */
+ * user slots are 1..CMB_PROCESS_TIMERS-1 and the public wrappers REJECT
+ * slot 0 (timer_add/set return false; cancel/pending treat it as absent;
+ * clear skips it), so a user timer can never hijack the next CMB_HOLD.
+ * A firing timer wakes the process's current blocking call with
+ * `sig`. ---- */
 #define CMB_PROCESS_TIMERS 4
 bool cmb_process_timer_add(cmb_sim* sim, cmb_process* p, int slot,
                            double delay, int64_t sig);

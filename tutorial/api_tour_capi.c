@@ -49,9 +49,12 @@ static void tour_body(cmb_sim* sim, cmb_process* me, void* vctx) {
     assert(fabs(cmb_event_time(sim, ctx->ev_a) - 6.0) < 1e-12);
     assert(cmb_event_reprioritize(sim, ctx->ev_b, -2));
     assert(cmb_event_priority(sim, ctx->ev_b) == -2);
-    assert(cmb_event_pattern_count(sim, noop_event, NULL, NULL) == 3);
-    assert(cmb_event_pattern_find(sim, NULL, (void*)4, NULL) == ctx->ev_c);
-    assert(cmb_event_pattern_cancel(sim, NULL, (void*)4, NULL) == 1);
+    assert(cmb_event_pattern_count(sim, noop_event, CMB_ANY_SUBJECT,
+                                   CMB_ANY_OBJECT) == 3);
+    assert(cmb_event_pattern_find(sim, CMB_ANY_ACTION, (void*)4,
+                                  CMB_ANY_OBJECT) == ctx->ev_c);
+    assert(cmb_event_pattern_cancel(sim, CMB_ANY_ACTION, (void*)4,
+                                    CMB_ANY_OBJECT) == 1);
     assert(!cmb_event_is_scheduled(sim, ctx->ev_c));
 
     /* ---- multi-slot timers: slot 1 fires first, slot 2 canceled ---- */
