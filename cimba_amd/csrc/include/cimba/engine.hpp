@@ -1045,6 +1045,73 @@ struct Engine {
         return true;
     }
 
+    // ---- context spill/restore + path peek: support for the
+    // path-converged lane kernel (hip/deskernel.hip conv_lane_kernel).
+    // A lane time-multiplexes K trials; the hot scalar context of a
+    // parked trial lives in this POD alongside its Storage. ----
+
+    struct Ctx {
+        double now;
+        uint64_t ev_dispatched;
+        uint64_t seq;
+        uint32_t next_handle;
+        int32_t status;
+        int32_t n_event_waiters;
+        uint32_t trial_index;
+        Rng rng;
+        int32_t evn;
+        int32_t pad_;
+    };
+
+    CMB_FORCEINLINE void ctx_save(Ctx& c) const {
+        c.now = now;
+        c.ev_dispatched = ev_dispatched;
+        c.seq = seq;
+        c.next_handle = next_handle;
+        c.status = status;
+        c.n_event_waiters = n_event_waiters;
+        c.trial_index = trial_index;
+        c.rng = rng;
+        c.evn = evq.n;
+    }
+
+    CMB_FORCEINLINE void ctx_load(const Ctx& c, const Params* p) {
+        now = c.now;
+        ev_dispatched = c.ev_dispatched;
+        seq = c.seq;
+        next_handle = c.next_handle;
+        status = c.status;
+        n_event_waiters = c.n_event_waiters;
+        trial_index = c.trial_index;
+        params = p;
+        rng = c.rng;
+        evq.n = c.evn;
+    }
+
+    // Identify the code path the NEXT dispatch_one() will take:
+    // (event kind, owning process function, its resume pc).  Lanes whose
+    // peeked paths are equal execute the same instruction stream through
+    // dispatch + model step, so the converged kernel votes on this id.
+    // PATH_DONE = the trial is finished under (until, max_events) — the
+    // same conditions as run()'s loop.
+    static constexpr uint32_t PATH_DONE = 0xFFFFFFFEu;
+    static constexpr uint32_t PATH_DEAD = 0xFFFFFFFFu;  // no trial in slot
+
+    CMB_FORCEINLINE uint32_t peek_path(double until,
+                                       uint64_t max_events) const {
+        if (status != ST_OK || evq.n == 0 || ev_dispatched >= max_events)
+            return PATH_DONE;
+        const EvEntry& ev = evq.top();
+        if (ev.t > until) return PATH_DONE;
+        uint32_t func = 0, pc = 0;
+        if (ev.kind >= EV_PROC_START && ev.kind <= EV_RESUME) {
+            const ProcT& p = procs[ev.a];
+            func = p.func;
+            pc = (uint16_t)p.pc;
+        }
+        return (((uint32_t)ev.kind & 0x7Fu) << 24) | (func << 16) | pc;
+    }
+
     // run until the event queue drains (reference cmb_event_queue_execute,
     // cmb_event.c:402) or a limit is hit
     CMB_FORCEINLINE void run(double until, uint64_t max_events) {

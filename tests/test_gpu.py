@@ -50,6 +50,51 @@ def test_lane_kernel_matches_wave_kernel():
     assert a["total_events"] == b["total_events"]
 
 
+def test_conv_kernel_matches_scratch_kernel():
+    """Path-converged K-trials-per-lane kernel: scheduling BETWEEN trials
+    differs, event order WITHIN a trial cannot — totals must be bitwise
+    equal to the plain scratch-lane kernel, for every K."""
+    import os
+    os.environ["CIMBA_MM1_LANE"] = "2"
+    try:
+        ref = ca.mm1_gpu(ntrials=2048, num_objects=2000, seed=99, device=0)
+    finally:
+        del os.environ["CIMBA_MM1_LANE"]
+    for k in (1, 2, 4, 8):
+        os.environ["CIMBA_MM1_LANE"] = "3"
+        os.environ["CIMBA_CONV_K"] = str(k)
+        try:
+            c = ca.mm1_gpu(ntrials=2048, num_objects=2000, seed=99, device=0)
+        finally:
+            del os.environ["CIMBA_MM1_LANE"]
+            del os.environ["CIMBA_CONV_K"]
+        assert c["trials_ok"] == 2048, (k, c)
+        assert c["total_wait"] == ref["total_wait"], k
+        assert c["total_events"] == ref["total_events"], k
+
+
+def test_conv_kernel_mg1_jobshop_match():
+    import os
+    for var, fn, kw in [
+        ("CIMBA_MG1_LANE", ca.mg1_gpu,
+         dict(ntrials=1024, num_objects=2000, arr_rate=0.8, srv_mean=1.0,
+              srv_scv=0.25, dist=3, seed=7, device=0)),
+        ("CIMBA_JS_LANE", ca.jobshop_gpu,
+         dict(ntrials=512, entities=1000, njobs=24, seed=7, device=0)),
+    ]:
+        os.environ[var] = "2" if var == "CIMBA_MG1_LANE" else "1"
+        try:
+            ref = fn(**kw)
+        finally:
+            del os.environ[var]
+        os.environ[var] = "3"
+        try:
+            c = fn(**kw)
+        finally:
+            del os.environ[var]
+        assert c["total_events"] == ref["total_events"], var
+
+
 def test_bench_script_single_gpu():
     out = subprocess.run(
         [sys.executable, "bench.py", "--gpus", "1", "--steps", "1",
