@@ -403,10 +403,37 @@ int run_conv_auto(const typename Model::Params& P, uint64_t ntrials,
                   uint64_t seed, uint64_t trial_base, double until,
                   uint64_t max_events, double* elapsed_ms,
                   typename Model::Result* host_out) {
+    // measured (gpurun_out/r2_ab1.log): K=1 wins on M/M/1 — at a fixed
+    // trial pool, higher K cuts the grid K-fold and the lost wave
+    // parallelism outweighs the extra slot choice; K=1 still converges
+    // the wave on ONE voted path per iteration (idle lanes are cheaper
+    // than serialized paths).
     const char* ke = getenv("CIMBA_CONV_K");
-    const int k = ke ? atoi(ke) : 4;
+    const int k = ke ? atoi(ke) : 1;
     const char* be = getenv("CIMBA_CONV_BLOCKS");
     const uint32_t blocks = be ? (uint32_t)atoi(be) : 0u;
+    const char* me = getenv("CIMBA_CONV_MINW");
+    const int minw = me ? atoi(me) : 1;
+    if (minw >= 4) {
+        if (k >= 8)
+            return run_trials_gpu_conv<Model, 8, 4>(P, ntrials, seed,
+                                                    trial_base, until,
+                                                    max_events, elapsed_ms,
+                                                    host_out, blocks);
+        if (k >= 4)
+            return run_trials_gpu_conv<Model, 4, 4>(P, ntrials, seed,
+                                                    trial_base, until,
+                                                    max_events, elapsed_ms,
+                                                    host_out, blocks);
+        if (k >= 2)
+            return run_trials_gpu_conv<Model, 2, 4>(P, ntrials, seed,
+                                                    trial_base, until,
+                                                    max_events, elapsed_ms,
+                                                    host_out, blocks);
+        return run_trials_gpu_conv<Model, 1, 4>(P, ntrials, seed, trial_base,
+                                                until, max_events, elapsed_ms,
+                                                host_out, blocks);
+    }
     if (k >= 8)
         return run_trials_gpu_conv<Model, 8, 1>(P, ntrials, seed, trial_base,
                                                 until, max_events, elapsed_ms,
@@ -526,7 +553,7 @@ int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
     const char* lane = getenv("CIMBA_MG1_LANE");
     const uint64_t nt_ = ntrials;
     if (lane ? atoi(lane) != 0 : nt_ >= 32768) {
-        const int lane_mode = lane ? atoi(lane) : 3;
+        const int lane_mode = lane ? atoi(lane) : 2;  // conv pending A/B
         if (lane_mode == 3)
             return run_conv_auto<MG1>(*(const MG1::Params*)params, ntrials,
                                       seed, trial_base, 1.0e308,
@@ -562,7 +589,9 @@ int cimba_jobshop_gpu_run(uint64_t ntrials, const void* params,
     HIP_TRY(hipSetDevice(device));
     const char* lane = getenv("CIMBA_JS_LANE");
     if (lane ? atoi(lane) != 0 : ntrials >= 32768) {
-        const int lane_mode = lane ? atoi(lane) : 3;
+        // conv-in-scratch measured 3x slower for JobShop's large Storage
+        // (gpurun_out/r2_ab1.log) — HBM-lane stays the default
+        const int lane_mode = lane ? atoi(lane) : 1;
         if (lane_mode == 3)
             return run_conv_auto<JobShop>(
                 *(const JobShop::Params*)params, ntrials, seed, trial_base,

@@ -16,14 +16,16 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import cimba_amd as ca  # noqa: E402
 
 model = sys.argv[1] if len(sys.argv) > 1 else "mg1"
-if model == "mg1":
+if model == "mm1":
+    r = ca.mm1_gpu(ntrials=262144, num_objects=5000, seed=11, device=0)
+elif model == "mg1":
     r = ca.mg1_gpu(ntrials=262144, num_objects=5000, arr_rate=0.8,
                    srv_mean=1.0, srv_scv=0.25, dist=1, seed=11, device=0)
 elif model == "jobshop":
     r = ca.jobshop_gpu(ntrials=131072, entities=1000, njobs=12, seed=11,
                        device=0)
 else:
-    sys.exit("model must be mg1 or jobshop")
+    sys.exit("model must be mm1, mg1 or jobshop")
 ev = r["total_events"]
 ms = r["elapsed_ms"]
 print(f"{model}: {ev} events in {ms:.1f} ms = {ev/ms/1e6:.2f} G ev/s, "
