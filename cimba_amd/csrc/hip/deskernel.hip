@@ -148,170 +148,100 @@ __global__ __launch_bounds__(256, MINW) void lane_scratch_kernel(
 // ---------------------------------------------------------------------------
 // Path-converged lane kernel: the divergence fix for the lane regime.
 //
-// The plain lane kernels run 64 independent trials per wave; each dispatch
-// step the 64 lanes sit at random (event-kind, process-func, resume-pc)
-// points, so the wave serializes over every distinct path present and
-// VALUUtilization measured 20-28% (profiles/r01_lane_divergence.md).
-// Here each lane owns K trials (Storage + parked Ctx in scratch, which the
-// hardware interleaves per lane -> same-field access coalesces).  Each
-// iteration:
-//   1. every lane peeks the dispatch path of its K trials (Engine::peek_path)
-//   2. the wave votes: candidate paths are polled from the lanes
-//      (__shfl from the first lane of each unseen-proposal group) and the
-//      path that the most lanes can serve wins (ballot+popcount)
-//   3. every lane that has a trial on the winning path dispatches ONE event
-//      from it; the few lanes without a match idle for that iteration
-// With P(majority path) ~ 0.45 on M/M/1 and K=4, ~90% of lanes serve every
-// iteration and the wave executes ONE path body instead of ~3 — recovering
-// most of the 2-3x that divergence costs.  Trial results are bitwise
-// identical to the other kernels: scheduling BETWEEN trials changes, the
-// event order WITHIN a trial cannot.
+// The plain lane kernels run 64 independent trials per wave; at each
+// dispatch step the 64 lanes sit at random (event-kind, process-func,
+// resume-pc) points, so the wave serializes over every distinct path
+// present and VALUUtilization measured 20-28%
+// (profiles/r01_lane_divergence.md).  Here each lane still owns ONE trial
+// in scratch, but dispatch is gated by a wave vote: every iteration the
+// lanes peek their next dispatch path (Engine::peek_path), poll up to 4
+// distinct candidates (__shfl from the first lane of each unseen group),
+// and ONLY the lanes on the most-popular path dispatch.  Idle minority
+// lanes cost less than serializing all paths: measured 3.93 G ev/s vs
+// 3.33 G for the ungated scratch kernel (gpurun_out/r2_ab1.log).
+//
+// Trial refill is wave-synchronous via the stride loop — all lanes of a
+// wave start their next trial together, so the low-utilization tail is
+// one trial-length VARIANCE, not a full trial.  (The earlier design —
+// K parked trials per lane with per-lane atomic refill — measured 21%
+// WORSE at any batch that needs refill: staggered per-lane trial ages
+// leave each wave grinding out stragglers at ~50% liveness after the
+// pool empties.  gpurun_out/r2_sweep3 data; kept here as the measured
+// justification for the no-refill shape.)
 // ---------------------------------------------------------------------------
-template <class Model, int K, int MINW = 1>
+template <class Model, int MINW = 1>
 __global__ __launch_bounds__(256, MINW) void conv_lane_kernel(
     typename Model::Params P, uint64_t master_seed, uint64_t trial_base,
     uint32_t ntrials, double until, uint64_t max_events,
-    typename Model::Result* __restrict__ out,
-    uint32_t* __restrict__ next_trial) {
+    typename Model::Result* __restrict__ out) {
     using Eng = Engine<Model>;
-    using Ctx = typename Eng::Ctx;
     constexpr uint32_t DONE = Eng::PATH_DONE;
-    constexpr uint32_t DEAD = Eng::PATH_DEAD;
-
-    typename Eng::Storage st[K];  // per-lane scratch (HW-swizzled)
-    Ctx ctx[K];
-    uint32_t cur[K];   // trial index per slot; ~0u = empty slot
-    uint32_t path[K];
-
-    // claim + start one trial in slot k; returns its peeked path or DEAD
-    auto start_trial = [&](int k) -> uint32_t {
-        const uint32_t t = atomicAdd(next_trial, 1u);
-        if (t >= ntrials) {
-            cur[k] = ~0u;
-            return DEAD;
-        }
-        Eng E(st[k]);
-        E.init(&P, cmb::trial_seed(master_seed, trial_base + t),
-               (uint32_t)(trial_base + t));
-        Model::setup(E);
-        cur[k] = t;
-        uint32_t pth = E.peek_path(until, max_events);
-        while (pth == DONE) {  // degenerate zero-event trial
-            if (!E.evq.empty() && E.evq.top().t > until) E.now = until;
-            Model::finish(E, out[t]);
-            const uint32_t t2 = atomicAdd(next_trial, 1u);
-            if (t2 >= ntrials) {
-                cur[k] = ~0u;
-                return DEAD;
-            }
-            E.init(&P, cmb::trial_seed(master_seed, trial_base + t2),
-                   (uint32_t)(trial_base + t2));
+    const uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+    const uint32_t stride = gridDim.x * blockDim.x;
+    typename Eng::Storage st;  // per-lane scratch (HW-swizzled)
+    Eng E(st);
+    for (uint32_t trial = gid;; trial += stride) {
+        const bool have = trial < ntrials;
+        if (__ballot(have) == 0) break;
+        if (have) {
+            E.init(&P, cmb::trial_seed(master_seed, trial_base + trial),
+                   (uint32_t)(trial_base + trial));
             Model::setup(E);
-            cur[k] = t2;
-            pth = E.peek_path(until, max_events);
         }
-        E.ctx_save(ctx[k]);
-        return pth;
-    };
-
-    for (int k = 0; k < K; ++k) path[k] = start_trial(k);
-
-    int rr = 0;  // rotation counter: proposal + selection fairness
-    for (;;) {
-        // my proposal: first live slot from rr
-        uint32_t prop = DEAD;
-        for (int k = 0; k < K; ++k) {
-            const uint32_t p = path[(rr + k) % K];
-            if (p != DEAD) {
-                prop = p;
-                break;
+        uint32_t mypath = have ? E.peek_path(until, max_events) : DONE;
+        for (;;) {
+            const uint64_t live = __ballot(mypath != DONE);
+            if (live == 0) break;
+            // poll up to 4 distinct paths; serve the most popular
+            uint64_t rem = live;
+            uint32_t best = 0;
+            int bestv = -1;
+            for (int it = 0; it < 4 && rem; ++it) {
+                const int src = __ffsll((unsigned long long)rem) - 1;
+                const uint32_t cand = __shfl(mypath, src);
+                const uint64_t m = __ballot(mypath == cand);
+                const int v = __popcll(m);
+                if (v > bestv) {
+                    bestv = v;
+                    best = cand;
+                }
+                rem &= ~m;
+            }
+            if (mypath == best) {
+                E.dispatch_one();
+                if (E.ev_dispatched >= max_events)
+                    E.fail(cmb::ST_EVENT_LIMIT);
+                mypath = E.peek_path(until, max_events);
             }
         }
-        const uint64_t live = __ballot(prop != DEAD);
-        if (live == 0) break;  // every slot of every lane exhausted
-
-        // vote: poll up to 4 distinct proposals, pick max coverage
-        uint64_t rem = live;
-        uint32_t best = 0;
-        int bestv = -1;
-        for (int it = 0; it < 4 && rem; ++it) {
-            const int src = __ffsll((unsigned long long)rem) - 1;
-            const uint32_t cand = __shfl(prop, src);
-            bool has = false;
-            for (int k = 0; k < K; ++k) has |= (path[k] == cand);
-            const int v = __popcll(__ballot(has));
-            if (v > bestv) {
-                bestv = v;
-                best = cand;
-            }
-            rem &= ~__ballot(prop == cand);
-        }
-
-        // my serving slot, rotating start so equal-path slots alternate
-        int sel = -1;
-        for (int k = 0; k < K; ++k) {
-            const int kk = (rr + k) % K;
-            if (path[kk] == best) {
-                sel = kk;
-                break;
-            }
-        }
-        rr = (rr + 1) % K;
-        if (sel < 0) continue;  // no match: idle this iteration
-
-        Eng E(st[sel]);
-        E.ctx_load(ctx[sel], &P);
-        E.dispatch_one();
-        if (E.ev_dispatched >= max_events) E.fail(cmb::ST_EVENT_LIMIT);
-        uint32_t pth = E.peek_path(until, max_events);
-        if (pth == DONE) {
+        if (have) {
             if (!E.evq.empty() && E.evq.top().t > until) E.now = until;
-            Model::finish(E, out[cur[sel]]);
-            path[sel] = start_trial(sel);
-        } else {
-            path[sel] = pth;
-            E.ctx_save(ctx[sel]);
+            Model::finish(E, out[trial]);
         }
     }
 }
 
-template <class Model, int K, int MINW>
+template <class Model, int MINW>
 int run_trials_gpu_conv(const typename Model::Params& P, uint64_t ntrials,
                         uint64_t seed, uint64_t trial_base, double until,
                         uint64_t max_events, double* elapsed_ms,
                         typename Model::Result* host_out, uint32_t blocks) {
     using Result = typename Model::Result;
-    if (blocks == 0) {
-        // persistent-style sizing: grid = the blocks actually RESIDENT, so
-        // every lane keeps K live trials to vote among; a bigger grid would
-        // just dilute the trial pool across waiting blocks
-        int dev = 0, nsm = 0, occ = 0;
-        HIP_TRY(hipGetDevice(&dev));
-        HIP_TRY(hipDeviceGetAttribute(
-            &nsm, hipDeviceAttributeMultiprocessorCount, dev));
-        HIP_TRY(hipOccupancyMaxActiveBlocksPerMultiprocessor(
-            &occ, (conv_lane_kernel<Model, K, MINW>), 256, 0));
-        if (occ < 1) occ = 1;
-        blocks = (uint32_t)(nsm * occ);
-    }
-    const uint32_t want = (uint32_t)((ntrials + 256 * K - 1) / (256 * K));
+    if (blocks == 0) blocks = 16384u;
+    const uint32_t want = (uint32_t)((ntrials + 255) / 256);
     const uint32_t grid = want < blocks ? want : blocks;
     if (getenv("CIMBA_CONV_DEBUG"))
-        fprintf(stderr, "[conv] K=%d resident_blocks=%u grid=%u slots=%u ntrials=%llu\n",
-                K, blocks, grid, grid * 256u * (uint32_t)K,
-                (unsigned long long)ntrials);
+        fprintf(stderr, "[conv] grid=%u lanes=%u ntrials=%llu\n", grid,
+                grid * 256u, (unsigned long long)ntrials);
     Result* d_out = nullptr;
-    uint32_t* d_next = nullptr;
     HIP_TRY(hipMalloc(&d_out, sizeof(Result) * ntrials));
-    HIP_TRY(hipMalloc(&d_next, sizeof(uint32_t)));
-    HIP_TRY(hipMemset(d_next, 0, sizeof(uint32_t)));
     hipEvent_t t0, t1;
     HIP_TRY(hipEventCreate(&t0));
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
-    hipLaunchKernelGGL((conv_lane_kernel<Model, K, MINW>), dim3(grid),
+    hipLaunchKernelGGL((conv_lane_kernel<Model, MINW>), dim3(grid),
                        dim3(256), 0, 0, P, seed, trial_base,
-                       (uint32_t)ntrials, until, max_events, d_out, d_next);
+                       (uint32_t)ntrials, until, max_events, d_out);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -321,7 +251,6 @@ int run_trials_gpu_conv(const typename Model::Params& P, uint64_t ntrials,
     HIP_TRY(hipMemcpy(host_out, d_out, sizeof(Result) * ntrials,
                       hipMemcpyDeviceToHost));
     HIP_TRY(hipFree(d_out));
-    HIP_TRY(hipFree(d_next));
     HIP_TRY(hipEventDestroy(t0));
     HIP_TRY(hipEventDestroy(t1));
     return 0;
@@ -395,60 +324,24 @@ int run_trials_gpu_lane(const typename Model::Params& P, uint64_t ntrials,
     return 0;
 }
 
-
-// converged-kernel front end: K (trials per lane) and grid from env, with
-// occupancy-sized grid by default (0 = query residency)
+// converged-kernel front end: grid (CIMBA_CONV_BLOCKS, 0 = one lane per
+// trial capped at 16384 blocks) and register budget (CIMBA_CONV_MINW)
 template <class Model>
 int run_conv_auto(const typename Model::Params& P, uint64_t ntrials,
                   uint64_t seed, uint64_t trial_base, double until,
                   uint64_t max_events, double* elapsed_ms,
                   typename Model::Result* host_out) {
-    // measured (gpurun_out/r2_ab1.log): K=1 wins on M/M/1 — at a fixed
-    // trial pool, higher K cuts the grid K-fold and the lost wave
-    // parallelism outweighs the extra slot choice; K=1 still converges
-    // the wave on ONE voted path per iteration (idle lanes are cheaper
-    // than serialized paths).
-    const char* ke = getenv("CIMBA_CONV_K");
-    const int k = ke ? atoi(ke) : 1;
     const char* be = getenv("CIMBA_CONV_BLOCKS");
     const uint32_t blocks = be ? (uint32_t)atoi(be) : 0u;
     const char* me = getenv("CIMBA_CONV_MINW");
     const int minw = me ? atoi(me) : 1;
-    if (minw >= 4) {
-        if (k >= 8)
-            return run_trials_gpu_conv<Model, 8, 4>(P, ntrials, seed,
-                                                    trial_base, until,
-                                                    max_events, elapsed_ms,
-                                                    host_out, blocks);
-        if (k >= 4)
-            return run_trials_gpu_conv<Model, 4, 4>(P, ntrials, seed,
-                                                    trial_base, until,
-                                                    max_events, elapsed_ms,
-                                                    host_out, blocks);
-        if (k >= 2)
-            return run_trials_gpu_conv<Model, 2, 4>(P, ntrials, seed,
-                                                    trial_base, until,
-                                                    max_events, elapsed_ms,
-                                                    host_out, blocks);
-        return run_trials_gpu_conv<Model, 1, 4>(P, ntrials, seed, trial_base,
-                                                until, max_events, elapsed_ms,
-                                                host_out, blocks);
-    }
-    if (k >= 8)
-        return run_trials_gpu_conv<Model, 8, 1>(P, ntrials, seed, trial_base,
-                                                until, max_events, elapsed_ms,
-                                                host_out, blocks);
-    if (k >= 4)
-        return run_trials_gpu_conv<Model, 4, 1>(P, ntrials, seed, trial_base,
-                                                until, max_events, elapsed_ms,
-                                                host_out, blocks);
-    if (k >= 2)
-        return run_trials_gpu_conv<Model, 2, 1>(P, ntrials, seed, trial_base,
-                                                until, max_events, elapsed_ms,
-                                                host_out, blocks);
-    return run_trials_gpu_conv<Model, 1, 1>(P, ntrials, seed, trial_base,
-                                            until, max_events, elapsed_ms,
-                                            host_out, blocks);
+    if (minw >= 4)
+        return run_trials_gpu_conv<Model, 4>(P, ntrials, seed, trial_base,
+                                             until, max_events, elapsed_ms,
+                                             host_out, blocks);
+    return run_trials_gpu_conv<Model, 1>(P, ntrials, seed, trial_base,
+                                         until, max_events, elapsed_ms,
+                                         host_out, blocks);
 }
 
 // single-trial semantic-parity kernel: runs one Scenario trial on-device;

@@ -1,3 +1,4 @@
+"""Vote-gated conv kernel vs plain lane kernels across batch sizes."""
 import os, sys
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import cimba_amd as ca
@@ -8,17 +9,34 @@ def mm1(tag, n, env):
     for k, v in env.items(): os.environ[k] = v
     try:
         r = ca.mm1_gpu(ntrials=n, num_objects=10000, seed=SEED, device=0)
-        print(f"mm1 {tag:28s} N={n}: {r['total_events']/(r['elapsed_ms']*1e-3)/1e9:.3f} G ev/s ok={r['trials_ok']==n}", flush=True)
+        print(f"mm1 {tag:24s} N={n}: {r['total_events']/(r['elapsed_ms']*1e-3)/1e9:.3f} G ev/s ok={r['trials_ok']==n}", flush=True)
+    finally:
+        for k in env: del os.environ[k]
+
+def mg1(tag, env):
+    for k, v in env.items(): os.environ[k] = v
+    try:
+        r = ca.mg1_gpu(ntrials=262144, num_objects=10000, arr_rate=0.8, srv_mean=1.0,
+                       srv_scv=0.25, dist=3, seed=SEED, device=0)
+        print(f"mg1 {tag}: {r['total_events']/(r['elapsed_ms']*1e-3)/1e9:.3f} G ev/s", flush=True)
+    finally:
+        for k in env: del os.environ[k]
+
+def js(tag, env):
+    for k, v in env.items(): os.environ[k] = v
+    try:
+        r = ca.jobshop_gpu(ntrials=65536, entities=10000, njobs=24, seed=SEED, device=0)
+        print(f"js  {tag}: {r['total_events']/(r['elapsed_ms']*1e-3)/1e9:.3f} G ev/s", flush=True)
     finally:
         for k in env: del os.environ[k]
 
 mm1("warmup", 262144, {})
 for n in (262144, 524288, 1048576):
-    mm1("conv K=1", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_DEBUG": "1"})
-for n in (524288, 1048576):
-    mm1("conv K=1 B=2048", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_BLOCKS": "2048"})
-    mm1("conv K=1 B=4096", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_BLOCKS": "4096"})
-    mm1("conv K=2", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_K": "2", "CIMBA_CONV_DEBUG": "1"})
-    mm1("conv K=2 B=2048", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_K": "2", "CIMBA_CONV_BLOCKS": "2048"})
+    mm1("conv", n, {"CIMBA_MM1_LANE": "3"})
+    mm1("conv MINW=4", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_MINW": "4"})
     mm1("scratch", n, {"CIMBA_MM1_LANE": "2"})
-    mm1("scratch B=4096", n, {"CIMBA_MM1_LANE": "2", "CIMBA_MM1_LANE_BLOCKS": "4096"})
+mg1("scratch ", {"CIMBA_MG1_LANE": "2"})
+mg1("conv    ", {"CIMBA_MG1_LANE": "3"})
+mg1("conv M4 ", {"CIMBA_MG1_LANE": "3", "CIMBA_CONV_MINW": "4"})
+js("hbm ", {"CIMBA_JS_LANE": "1"})
+js("conv", {"CIMBA_JS_LANE": "3"})

@@ -51,26 +51,27 @@ def test_lane_kernel_matches_wave_kernel():
 
 
 def test_conv_kernel_matches_scratch_kernel():
-    """Path-converged K-trials-per-lane kernel: scheduling BETWEEN trials
-    differs, event order WITHIN a trial cannot — totals must be bitwise
-    equal to the plain scratch-lane kernel, for every K."""
+    """Vote-gated converged kernel: scheduling BETWEEN trials differs,
+    event order WITHIN a trial cannot — totals must be bitwise equal to
+    the plain scratch-lane kernel, including when the grid is forced
+    small so lanes loop over multiple trial fills."""
     import os
     os.environ["CIMBA_MM1_LANE"] = "2"
     try:
         ref = ca.mm1_gpu(ntrials=2048, num_objects=2000, seed=99, device=0)
     finally:
         del os.environ["CIMBA_MM1_LANE"]
-    for k in (1, 2, 4, 8):
+    for blocks in ("0", "4"):  # 4 blocks = 2 wave-synchronous fills
         os.environ["CIMBA_MM1_LANE"] = "3"
-        os.environ["CIMBA_CONV_K"] = str(k)
+        os.environ["CIMBA_CONV_BLOCKS"] = blocks
         try:
             c = ca.mm1_gpu(ntrials=2048, num_objects=2000, seed=99, device=0)
         finally:
             del os.environ["CIMBA_MM1_LANE"]
-            del os.environ["CIMBA_CONV_K"]
-        assert c["trials_ok"] == 2048, (k, c)
-        assert c["total_wait"] == ref["total_wait"], k
-        assert c["total_events"] == ref["total_events"], k
+            del os.environ["CIMBA_CONV_BLOCKS"]
+        assert c["trials_ok"] == 2048, (blocks, c)
+        assert c["total_wait"] == ref["total_wait"], blocks
+        assert c["total_events"] == ref["total_events"], blocks
 
 
 def test_conv_kernel_mg1_jobshop_match():
