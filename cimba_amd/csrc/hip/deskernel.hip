@@ -333,8 +333,20 @@ int run_conv_auto(const typename Model::Params& P, uint64_t ntrials,
                   typename Model::Result* host_out) {
     const char* be = getenv("CIMBA_CONV_BLOCKS");
     const uint32_t blocks = be ? (uint32_t)atoi(be) : 0u;
+    // MINW (waves/SIMD floor) trades registers for occupancy: the engine
+    // context spills into hardware-swizzled scratch (L1-friendly) and the
+    // extra resident waves hide the dispatch chain's memory latency — the
+    // dominant term at 131 VGPRs / 3 waves (r2 sweeps).
     const char* me = getenv("CIMBA_CONV_MINW");
-    const int minw = me ? atoi(me) : 1;
+    const int minw = me ? atoi(me) : 8;
+    if (minw >= 8)
+        return run_trials_gpu_conv<Model, 8>(P, ntrials, seed, trial_base,
+                                             until, max_events, elapsed_ms,
+                                             host_out, blocks);
+    if (minw >= 6)
+        return run_trials_gpu_conv<Model, 6>(P, ntrials, seed, trial_base,
+                                             until, max_events, elapsed_ms,
+                                             host_out, blocks);
     if (minw >= 4)
         return run_trials_gpu_conv<Model, 4>(P, ntrials, seed, trial_base,
                                              until, max_events, elapsed_ms,

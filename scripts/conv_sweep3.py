@@ -32,11 +32,14 @@ def js(tag, env):
 
 mm1("warmup", 262144, {})
 for n in (262144, 524288, 1048576):
-    mm1("conv", n, {"CIMBA_MM1_LANE": "3"})
-    mm1("conv MINW=4", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_MINW": "4"})
+    mm1("conv M8", n, {"CIMBA_MM1_LANE": "3"})
+    mm1("conv M6", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_MINW": "6"})
+    mm1("conv M4", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_MINW": "4"})
     mm1("scratch", n, {"CIMBA_MM1_LANE": "2"})
 mg1("scratch ", {"CIMBA_MG1_LANE": "2"})
 mg1("conv    ", {"CIMBA_MG1_LANE": "3"})
+mg1("conv M6 ", {"CIMBA_MG1_LANE": "3", "CIMBA_CONV_MINW": "6"})
 mg1("conv M4 ", {"CIMBA_MG1_LANE": "3", "CIMBA_CONV_MINW": "4"})
 js("hbm ", {"CIMBA_JS_LANE": "1"})
 js("conv", {"CIMBA_JS_LANE": "3"})
+js("conv M6", {"CIMBA_JS_LANE": "3", "CIMBA_CONV_MINW": "6"})
