@@ -324,6 +324,27 @@ int run_trials_gpu_lane(const typename Model::Params& P, uint64_t ntrials,
     return 0;
 }
 
+// scratch-kernel front end: MINW occupancy knob (CIMBA_SCRATCH_MINW)
+template <class Model>
+int run_scratch_auto(const typename Model::Params& P, uint64_t ntrials,
+                     uint64_t seed, uint64_t trial_base, double until,
+                     uint64_t max_events, double* elapsed_ms,
+                     typename Model::Result* host_out, uint32_t blocks) {
+    const char* me = getenv("CIMBA_SCRATCH_MINW");
+    const int minw = me ? atoi(me) : 1;
+    if (minw >= 6)
+        return run_trials_gpu_lane_scratch<Model, 6>(
+            P, ntrials, seed, trial_base, until, max_events, elapsed_ms,
+            host_out, blocks);
+    if (minw >= 4)
+        return run_trials_gpu_lane_scratch<Model, 4>(
+            P, ntrials, seed, trial_base, until, max_events, elapsed_ms,
+            host_out, blocks);
+    return run_trials_gpu_lane_scratch<Model, 1>(
+        P, ntrials, seed, trial_base, until, max_events, elapsed_ms,
+        host_out, blocks);
+}
+
 // converged-kernel front end: grid (CIMBA_CONV_BLOCKS, 0 = one lane per
 // trial capped at 16384 blocks) and register budget (CIMBA_CONV_MINW)
 template <class Model>
@@ -406,10 +427,9 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
                                              max_events, &out->elapsed_ms,
                                              res.data(), blocks);
         else if (lane_mode == 2)  // scratch (HW lane-interleaved)
-            rc = run_trials_gpu_lane_scratch<MM1, 1>(P, ntrials, seed, trial_base, until,
-                                                     max_events,
-                                                     &out->elapsed_ms,
-                                                     res.data(), blocks);
+            rc = run_scratch_auto<MM1>(P, ntrials, seed, trial_base, until,
+                                       max_events, &out->elapsed_ms,
+                                       res.data(), blocks);
         else  // default: path-converged K-trials-per-lane (divergence fix)
             rc = run_conv_auto<MM1>(P, ntrials, seed, trial_base, until,
                                     max_events, &out->elapsed_ms, res.data());
@@ -464,7 +484,7 @@ int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                                       seed, trial_base, 1.0e308,
                                       UINT64_C(0xFFFFFFFFFFFFFFFF),
                                       elapsed_ms, (MG1::Result*)results_out);
-        return run_trials_gpu_lane_scratch<MG1, 1>(
+        return run_scratch_auto<MG1>(
             *(const MG1::Params*)params, ntrials, seed, trial_base, 1.0e308,
             UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
             (MG1::Result*)results_out, 2048u);

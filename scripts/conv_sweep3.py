@@ -36,7 +36,10 @@ for n in (262144, 524288, 1048576):
     mm1("conv M6", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_MINW": "6"})
     mm1("conv M4", n, {"CIMBA_MM1_LANE": "3", "CIMBA_CONV_MINW": "4"})
     mm1("scratch", n, {"CIMBA_MM1_LANE": "2"})
+    mm1("scratch M4", n, {"CIMBA_MM1_LANE": "2", "CIMBA_SCRATCH_MINW": "4"})
+    mm1("scratch M6", n, {"CIMBA_MM1_LANE": "2", "CIMBA_SCRATCH_MINW": "6"})
 mg1("scratch ", {"CIMBA_MG1_LANE": "2"})
+mg1("scr M4  ", {"CIMBA_MG1_LANE": "2", "CIMBA_SCRATCH_MINW": "4"})
 mg1("conv    ", {"CIMBA_MG1_LANE": "3"})
 mg1("conv M6 ", {"CIMBA_MG1_LANE": "3", "CIMBA_CONV_MINW": "6"})
 mg1("conv M4 ", {"CIMBA_MG1_LANE": "3", "CIMBA_CONV_MINW": "4"})
