@@ -359,7 +359,7 @@ int run_conv_auto(const typename Model::Params& P, uint64_t ntrials,
     // extra resident waves hide the dispatch chain's memory latency — the
     // dominant term at 131 VGPRs / 3 waves (r2 sweeps).
     const char* me = getenv("CIMBA_CONV_MINW");
-    const int minw = me ? atoi(me) : 8;
+    const int minw = me ? atoi(me) : 4;  // measured best (r2 sweeps)
     if (minw >= 8)
         return run_trials_gpu_conv<Model, 8>(P, ntrials, seed, trial_base,
                                              until, max_events, elapsed_ms,
@@ -421,7 +421,10 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
     if (use_lane) {
         const char* lb = getenv("CIMBA_MM1_LANE_BLOCKS");
         const uint32_t blocks = lb ? (uint32_t)atoi(lb) : 2048u;
-        const int lane_mode = lane ? atoi(lane) : 3;
+        // measured (gpurun r2 sweeps): plain scratch wins at the bench
+        // batch (4.0 G ev/s at N>=524288); vote-gated conv only matched
+        // the occupancy-tuned scratch on M/M/1, so scratch stays default
+        const int lane_mode = lane ? atoi(lane) : 2;
         if (lane_mode == 1)  // explicit HBM-lane variant
             rc = run_trials_gpu_lane<MM1, 1>(P, ntrials, seed, trial_base, until,
                                              max_events, &out->elapsed_ms,
@@ -478,7 +481,10 @@ int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
     const char* lane = getenv("CIMBA_MG1_LANE");
     const uint64_t nt_ = ntrials;
     if (lane ? atoi(lane) != 0 : nt_ >= 32768) {
-        const int lane_mode = lane ? atoi(lane) : 2;  // conv pending A/B
+        // measured: vote-gated conv at MINW=4 = 4.19 G ev/s vs 3.69 G
+        // scratch — MG1 has a wider path mix (service-distribution
+        // branches), so path convergence pays where it did not for M/M/1
+        const int lane_mode = lane ? atoi(lane) : 3;
         if (lane_mode == 3)
             return run_conv_auto<MG1>(*(const MG1::Params*)params, ntrials,
                                       seed, trial_base, 1.0e308,
