@@ -135,7 +135,13 @@ struct ProcRec {
 struct Guard {
     int16_t head;      // waiter list (proc indices via gnext), -1 = empty
     int16_t observer;  // condition index observing this guard, -1 = none
-    CMB_FORCEINLINE bool empty() const { return head < 0; }
+    uint32_t pad_;
+    // Waiter BITMASK, maintained instead of the list when MAX_PROC <= 64:
+    // guard_front's waiter walk becomes a bit-iterate whose ProcRec loads
+    // are INDEPENDENT (memory-level parallel) instead of a serial
+    // pointer-chase through gnext links — the dominant latency chain in
+    // pool-heavy models (JobShop, profiles/r01_lane_divergence.md).
+    uint64_t wmask;
 };
 
 // FIFO object queue of 64-bit payloads (reference cmb_objectqueue: FIFO of
@@ -328,7 +334,11 @@ struct Engine {
             pr.pnext = -1;
             for (int t = 0; t < Cfg::TIMERS; ++t) pr.timers[t] = 0;
         }
-        for (int i = 0; i < NGUARD; ++i) { guards[i].head = -1; guards[i].observer = -1; }
+        for (int i = 0; i < NGUARD; ++i) {
+            guards[i].head = -1;
+            guards[i].observer = -1;
+            guards[i].wmask = 0;
+        }
         int g = 0;
         for (int i = 0; i < NQ; ++i) {
             ObjQueue<Cfg::QCAP>& q = queues[i];
@@ -359,6 +369,13 @@ struct Engine {
             q.recording = 0; q.len_stats.reset(); q.t_last = now;
         }
         for (int i = 0; i < NC; ++i) conds[i].gid = (int16_t)g++;
+    }
+
+    static constexpr bool GUARD_MASK = (Cfg::MAX_PROC <= 64);
+
+    CMB_FORCEINLINE bool guard_empty(int gid) const {
+        if constexpr (GUARD_MASK) return guards[gid].wmask == 0;
+        else return guards[gid].head < 0;
     }
 
     CMB_FORCEINLINE void fail(int32_t st) {
@@ -644,13 +661,21 @@ struct Engine {
         p.demand_ctx = ctx;
         p.entry_t = now;
         p.wseq = seq++;
-        p.gnext = guards[gid].head;
-        guards[gid].head = (int16_t)pidx_of(&p);
+        if constexpr (GUARD_MASK) {
+            guards[gid].wmask |= (uint64_t)1 << pidx_of(&p);
+        } else {
+            p.gnext = guards[gid].head;
+            guards[gid].head = (int16_t)pidx_of(&p);
+        }
         await_setup(p, AW_GUARD, (uint32_t)p.wseq);  // key is informational
     }
 
     CMB_FORCEINLINE void guard_unlink(ProcT& p) {
         Guard& g = guards[p.gid];
+        if constexpr (GUARD_MASK) {
+            g.wmask &= ~((uint64_t)1 << pidx_of(&p));
+            return;
+        }
         int16_t* link = &g.head;
         const int16_t me = (int16_t)pidx_of(&p);
         while (*link >= 0) {
@@ -667,6 +692,22 @@ struct Engine {
     // (reference cmb_resourceguard.c:66-89 ordering)
     CMB_FORCEINLINE int guard_front(int gid) const {
         int best = -1;
+        if constexpr (GUARD_MASK) {
+            uint64_t m = guards[gid].wmask;
+            while (m) {
+                const int i = __builtin_ctzll(m);
+                m &= m - 1;
+                if (best < 0) { best = i; continue; }
+                const ProcT& a = procs[i];
+                const ProcT& b = procs[best];
+                if (a.priority > b.priority ||
+                    (a.priority == b.priority &&
+                     (a.entry_t < b.entry_t ||
+                      (a.entry_t == b.entry_t && a.wseq < b.wseq))))
+                    best = i;
+            }
+            return best;
+        }
         for (int16_t i = guards[gid].head; i >= 0; i = procs[i].gnext) {
             if (best < 0) { best = i; continue; }
             const ProcT& a = procs[i];
@@ -724,6 +765,20 @@ struct Engine {
     CMB_FORCEINLINE uint64_t condition_signal(int ci) {
         const int gid = conds[ci].gid;
         uint64_t cnt = 0;
+        if constexpr (GUARD_MASK) {
+            uint64_t m = guards[gid].wmask;
+            while (m) {
+                const int i = __builtin_ctzll(m);
+                m &= m - 1;
+                if (eval_demand(procs[i])) {
+                    ProcT& p = procs[i];
+                    schedule(EV_GRANT, (uint16_t)i, (uint32_t)gid, p.wseq,
+                             now, p.priority);
+                    ++cnt;
+                }
+            }
+            return cnt;
+        }
         for (int16_t i = guards[gid].head; i >= 0; i = procs[i].gnext) {
             if (eval_demand(procs[i])) {
                 ProcT& p = procs[i];
@@ -746,7 +801,7 @@ struct Engine {
 
     CMB_FORCEINLINE bool q_try_put(int qi, ProcT& p, uint64_t val) {
         ObjQueue<Cfg::QCAP>& q = queues[qi];
-        const bool may = p.g_granted || guards[q.g_rear].empty();
+        const bool may = p.g_granted || guard_empty(q.g_rear);
         p.g_granted = 0;
         if (!may || q.len >= q.limit) return false;
         if (q.len >= Cfg::QCAP) { fail(ST_QUEUE_FULL); return false; }
@@ -759,7 +814,7 @@ struct Engine {
 
     CMB_FORCEINLINE bool q_try_get(int qi, ProcT& p, uint64_t* out) {
         ObjQueue<Cfg::QCAP>& q = queues[qi];
-        const bool may = p.g_granted || guards[q.g_front].empty();
+        const bool may = p.g_granted || guard_empty(q.g_front);
         p.g_granted = 0;
         if (!may || q.len == 0) return false;
         q_record(q);
@@ -774,7 +829,7 @@ struct Engine {
 
     CMB_FORCEINLINE bool pq_try_put(int qi, ProcT& p, uint64_t val, int priority) {
         PrioQueue<Cfg::PQCAP>& q = pqueues[qi];
-        const bool may = p.g_granted || guards[q.g_rear].empty();
+        const bool may = p.g_granted || guard_empty(q.g_rear);
         p.g_granted = 0;
         if (!may || q.len >= q.limit) return false;
         // binary heap push keyed (pri desc, seq asc)
@@ -796,7 +851,7 @@ struct Engine {
 
     CMB_FORCEINLINE bool pq_try_get(int qi, ProcT& p, uint64_t* out) {
         PrioQueue<Cfg::PQCAP>& q = pqueues[qi];
-        const bool may = p.g_granted || guards[q.g_front].empty();
+        const bool may = p.g_granted || guard_empty(q.g_front);
         p.g_granted = 0;
         if (!may || q.len == 0) return false;
         *out = q.val[0];
@@ -821,7 +876,7 @@ struct Engine {
 
     CMB_FORCEINLINE bool res_try_acquire(int ri, ProcT& p) {
         Resource& r = resources[ri];
-        const bool may = p.g_granted || guards[r.gid].empty();
+        const bool may = p.g_granted || guard_empty(r.gid);
         p.g_granted = 0;
         if (!may || r.holder >= 0) return false;
         if (r.recording) {
@@ -845,7 +900,7 @@ struct Engine {
 
     CMB_FORCEINLINE int32_t pool_try_take(int pi, ProcT& p, int32_t want) {
         Pool& pl = pools[pi];
-        const bool may = p.g_granted || guards[pl.gid].empty();
+        const bool may = p.g_granted || guard_empty(pl.gid);
         p.g_granted = 0;
         if (!may) return 0;
         const int32_t free_units = pl.capacity - pl.in_use;
@@ -917,7 +972,7 @@ struct Engine {
     // loop for multi-unit requests; companion of CMB_POOL_ACQUIRE_ALL)
     CMB_FORCEINLINE bool pool_try_take_all(int pi, ProcT& p, int32_t want) {
         Pool& pl = pools[pi];
-        const bool may = p.g_granted || guards[pl.gid].empty();
+        const bool may = p.g_granted || guard_empty(pl.gid);
         p.g_granted = 0;
         if (!may || pl.capacity - pl.in_use < want) return false;
         if (pl.recording) {
@@ -948,7 +1003,7 @@ struct Engine {
 
     CMB_FORCEINLINE bool buf_try_get(int bi, ProcT& p, int64_t amount) {
         Buffer& b = buffers[bi];
-        const bool may = p.g_granted || guards[b.g_get].empty();
+        const bool may = p.g_granted || guard_empty(b.g_get);
         p.g_granted = 0;
         if (!may || b.level < amount) return false;
         if (b.recording) {
@@ -962,7 +1017,7 @@ struct Engine {
 
     CMB_FORCEINLINE bool buf_try_put(int bi, ProcT& p, int64_t amount) {
         Buffer& b = buffers[bi];
-        const bool may = p.g_granted || guards[b.g_put].empty();
+        const bool may = p.g_granted || guard_empty(b.g_put);
         p.g_granted = 0;
         if (!may || b.capacity - b.level < amount) return false;
         if (b.recording) {
