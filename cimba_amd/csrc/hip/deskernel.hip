@@ -36,13 +36,16 @@ template <class Model, int WPB, int MINW = 1>
 __global__ __launch_bounds__(WPB * 64, MINW) __attribute__((flatten)) void trial_kernel(
     typename Model::Params P, uint64_t master_seed, uint64_t trial_base,
     uint32_t ntrials, double until, uint64_t max_events,
-    typename Model::Result* __restrict__ out) {
+    typename Model::Result* __restrict__ out,
+    typename Engine<Model>::Spill* __restrict__ sp_arena,
+    int32_t* __restrict__ sp_cur, int32_t sp_cap) {
     // arrays in LDS; the Engine context (clock, seq, handles, status, RNG
     // state, heap size) is a per-lane local -> register-resident
     __shared__ typename Engine<Model>::Storage st[WPB];
     const int w = (int)(threadIdx.x >> 6);
     if ((threadIdx.x & 63) != 0) return;  // lane 0 of each wave drives
     Engine<Model> E(st[w]);
+    E.set_spill_pool(sp_arena, sp_cur, sp_cap);
     const uint32_t stride = gridDim.x * WPB;
     for (uint32_t trial = blockIdx.x * WPB + (uint32_t)w; trial < ntrials;
          trial += stride) {
@@ -66,6 +69,41 @@ static_assert(sizeof(Engine<JobShop>::Storage) * 4 < 64 * 1024,
         if (err_ != hipSuccess) return (int)err_;     \
     } while (0)
 
+// Device spill pool: slabs claimed lazily by trials that overflow their
+// fast tier (Engine::claim_spill).  Sized far above the measured overflow
+// rate (~1 trial per 5e5 at MG1 lognormal SCV=4, rho=0.8) and slabs stay
+// claimed by a lane across its later trials, so 4096 covers any batch.
+template <class Model>
+struct DevSpillPool {
+    using Spill = typename cmb::Engine<Model>::Spill;
+    Spill* arena = nullptr;
+    int32_t* cursor = nullptr;
+    int32_t cap = 0;
+    int alloc(uint64_t ntrials) {
+        if constexpr (!cmb::Engine<Model>::NEEDS_SPILL) {
+            (void)ntrials;
+            return 0;
+        } else {
+            const char* e = getenv("CIMBA_SPILL_SLABS");
+            uint64_t want = e ? (uint64_t)atoll(e) : 4096;
+            if (want > ntrials) want = ntrials;
+            if (want == 0) want = 1;
+            cap = (int32_t)want;
+            hipError_t err = hipMalloc(&arena, sizeof(Spill) * want);
+            if (err != hipSuccess) return (int)err;
+            err = hipMalloc(&cursor, sizeof(int32_t));
+            if (err != hipSuccess) return (int)err;
+            return (int)hipMemset(cursor, 0, sizeof(int32_t));
+        }
+    }
+    void release() {
+        if (arena) hipFree(arena);
+        if (cursor) hipFree(cursor);
+        arena = nullptr;
+        cursor = nullptr;
+    }
+};
+
 // host-side launcher: upload params, launch, copy per-trial results back
 template <class Model, int WPB, int MINW = 1>
 int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
@@ -77,6 +115,8 @@ int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
     HIP_TRY(hipMalloc(&d_out, sizeof(Result) * ntrials));
     const uint32_t want_blocks = (uint32_t)((ntrials + WPB - 1) / WPB);
     const uint32_t grid = want_blocks < 16384u ? want_blocks : 16384u;
+    DevSpillPool<Model> pool;
+    HIP_TRY((hipError_t)pool.alloc(ntrials));
 
     hipEvent_t t0, t1;
     HIP_TRY(hipEventCreate(&t0));
@@ -84,7 +124,8 @@ int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
     HIP_TRY(hipEventRecord(t0));
     hipLaunchKernelGGL((trial_kernel<Model, WPB, MINW>), dim3(grid),
                        dim3(WPB * 64), 0, 0, P, seed, trial_base,
-                       (uint32_t)ntrials, until, max_events, d_out);
+                       (uint32_t)ntrials, until, max_events, d_out,
+                       pool.arena, pool.cursor, pool.cap);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -94,6 +135,7 @@ int run_trials_gpu(const typename Model::Params& P, uint64_t ntrials,
     HIP_TRY(hipMemcpy(host_out, d_out, sizeof(Result) * ntrials,
                       hipMemcpyDeviceToHost));
     HIP_TRY(hipFree(d_out));
+    pool.release();
     HIP_TRY(hipEventDestroy(t0));
     HIP_TRY(hipEventDestroy(t1));
     return 0;
@@ -110,10 +152,13 @@ __global__ __launch_bounds__(256, MINW) __attribute__((flatten)) void lane_trial
     typename Model::Params P, uint64_t master_seed, uint64_t trial_base,
     uint32_t ntrials, double until, uint64_t max_events,
     typename Model::Result* __restrict__ out,
-    typename Engine<Model>::Storage* __restrict__ stores) {
+    typename Engine<Model>::Storage* __restrict__ stores,
+    typename Engine<Model>::Spill* __restrict__ sp_arena,
+    int32_t* __restrict__ sp_cur, int32_t sp_cap) {
     const uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
     const uint32_t stride = gridDim.x * blockDim.x;
     Engine<Model> E(stores[gid]);
+    E.set_spill_pool(sp_arena, sp_cur, sp_cap);
     for (uint32_t trial = gid; trial < ntrials; trial += stride) {
         E.init(&P, cmb::trial_seed(master_seed, trial_base + trial),
                (uint32_t)(trial_base + trial));
@@ -131,11 +176,14 @@ template <class Model, int MINW = 1>
 __global__ __launch_bounds__(256, MINW) void lane_scratch_kernel(
     typename Model::Params P, uint64_t master_seed, uint64_t trial_base,
     uint32_t ntrials, double until, uint64_t max_events,
-    typename Model::Result* __restrict__ out) {
+    typename Model::Result* __restrict__ out,
+    typename Engine<Model>::Spill* __restrict__ sp_arena,
+    int32_t* __restrict__ sp_cur, int32_t sp_cap) {
     const uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
     const uint32_t stride = gridDim.x * blockDim.x;
     typename Engine<Model>::Storage st;  // per-lane scratch (HW-swizzled)
     Engine<Model> E(st);
+    E.set_spill_pool(sp_arena, sp_cur, sp_cap);
     for (uint32_t trial = gid; trial < ntrials; trial += stride) {
         E.init(&P, cmb::trial_seed(master_seed, trial_base + trial),
                (uint32_t)(trial_base + trial));
@@ -173,13 +221,16 @@ template <class Model, int MINW = 1>
 __global__ __launch_bounds__(256, MINW) void conv_lane_kernel(
     typename Model::Params P, uint64_t master_seed, uint64_t trial_base,
     uint32_t ntrials, double until, uint64_t max_events,
-    typename Model::Result* __restrict__ out) {
+    typename Model::Result* __restrict__ out,
+    typename Engine<Model>::Spill* __restrict__ sp_arena,
+    int32_t* __restrict__ sp_cur, int32_t sp_cap) {
     using Eng = Engine<Model>;
     constexpr uint32_t DONE = Eng::PATH_DONE;
     const uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
     const uint32_t stride = gridDim.x * blockDim.x;
     typename Eng::Storage st;  // per-lane scratch (HW-swizzled)
     Eng E(st);
+    E.set_spill_pool(sp_arena, sp_cur, sp_cap);
     for (uint32_t trial = gid;; trial += stride) {
         const bool have = trial < ntrials;
         if (__ballot(have) == 0) break;
@@ -235,13 +286,16 @@ int run_trials_gpu_conv(const typename Model::Params& P, uint64_t ntrials,
                 grid * 256u, (unsigned long long)ntrials);
     Result* d_out = nullptr;
     HIP_TRY(hipMalloc(&d_out, sizeof(Result) * ntrials));
+    DevSpillPool<Model> pool;
+    HIP_TRY((hipError_t)pool.alloc(ntrials));
     hipEvent_t t0, t1;
     HIP_TRY(hipEventCreate(&t0));
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
     hipLaunchKernelGGL((conv_lane_kernel<Model, MINW>), dim3(grid),
                        dim3(256), 0, 0, P, seed, trial_base,
-                       (uint32_t)ntrials, until, max_events, d_out);
+                       (uint32_t)ntrials, until, max_events, d_out,
+                       pool.arena, pool.cursor, pool.cap);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -251,6 +305,7 @@ int run_trials_gpu_conv(const typename Model::Params& P, uint64_t ntrials,
     HIP_TRY(hipMemcpy(host_out, d_out, sizeof(Result) * ntrials,
                       hipMemcpyDeviceToHost));
     HIP_TRY(hipFree(d_out));
+    pool.release();
     HIP_TRY(hipEventDestroy(t0));
     HIP_TRY(hipEventDestroy(t1));
     return 0;
@@ -268,13 +323,16 @@ int run_trials_gpu_lane_scratch(const typename Model::Params& P,
     const uint32_t grid = want < blocks ? want : blocks;
     Result* d_out = nullptr;
     HIP_TRY(hipMalloc(&d_out, sizeof(Result) * ntrials));
+    DevSpillPool<Model> pool;
+    HIP_TRY((hipError_t)pool.alloc(ntrials));
     hipEvent_t t0, t1;
     HIP_TRY(hipEventCreate(&t0));
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
     hipLaunchKernelGGL((lane_scratch_kernel<Model, MINW>), dim3(grid),
                        dim3(256), 0, 0, P, seed, trial_base,
-                       (uint32_t)ntrials, until, max_events, d_out);
+                       (uint32_t)ntrials, until, max_events, d_out,
+                       pool.arena, pool.cursor, pool.cap);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -284,6 +342,7 @@ int run_trials_gpu_lane_scratch(const typename Model::Params& P,
     HIP_TRY(hipMemcpy(host_out, d_out, sizeof(Result) * ntrials,
                       hipMemcpyDeviceToHost));
     HIP_TRY(hipFree(d_out));
+    pool.release();
     HIP_TRY(hipEventDestroy(t0));
     HIP_TRY(hipEventDestroy(t1));
     return 0;
@@ -302,13 +361,16 @@ int run_trials_gpu_lane(const typename Model::Params& P, uint64_t ntrials,
     St* d_st = nullptr;
     HIP_TRY(hipMalloc(&d_out, sizeof(Result) * ntrials));
     HIP_TRY(hipMalloc(&d_st, sizeof(St) * (size_t)grid * 256));
+    DevSpillPool<Model> pool;
+    HIP_TRY((hipError_t)pool.alloc(ntrials));
     hipEvent_t t0, t1;
     HIP_TRY(hipEventCreate(&t0));
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
     hipLaunchKernelGGL((lane_trial_kernel<Model, MINW>), dim3(grid),
                        dim3(256), 0, 0, P, seed, trial_base,
-                       (uint32_t)ntrials, until, max_events, d_out, d_st);
+                       (uint32_t)ntrials, until, max_events, d_out, d_st,
+                       pool.arena, pool.cursor, pool.cap);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
@@ -319,6 +381,7 @@ int run_trials_gpu_lane(const typename Model::Params& P, uint64_t ntrials,
                       hipMemcpyDeviceToHost));
     HIP_TRY(hipFree(d_out));
     HIP_TRY(hipFree(d_st));
+    pool.release();
     HIP_TRY(hipEventDestroy(t0));
     HIP_TRY(hipEventDestroy(t1));
     return 0;

@@ -33,9 +33,11 @@ struct CModel : ModelBase {
     struct Cfg {
         static constexpr int MAX_PROC = C_MAX_PROC;
         static constexpr int MAX_EV = 16384;
+        static constexpr int SPILL_EV = 49152;   // events grow to 64K total
         static constexpr int TIMERS = 4;
         static constexpr int NUM_QUEUES = 16;
         static constexpr int QCAP = 8192;
+        static constexpr int SPILL_Q = 24576;    // queues grow to 32K total
         static constexpr int NUM_RES = 16;
         static constexpr int NUM_POOLS = 16;
         static constexpr int NUM_BUFS = 16;
@@ -235,6 +237,8 @@ uint64_t cimba_run(void* experiment, uint64_t n, size_t size,
         if (g_thread_init) g_thread_init(widx);
         auto store = std::make_unique<CStorage>();
         auto eng = std::make_unique<CEngine>(*store);
+        auto slab = std::make_unique<CEngine::Spill>();  // HBM-analog tier
+        eng->set_spill(slab.get());
         cmb_sim sim;
         sim.E = eng.get();
         sim.params.sim = &sim;
@@ -333,8 +337,8 @@ uint64_t cmb_event_pattern_count(cmb_sim* s, cmb_event_func* action,
     uint64_t cnt = 0;
     auto& q = s->E->evq;
     for (int32_t i = 0; i < q.n; ++i) {
-        if (q.e[i].kind != EV_USER) continue;
-        const auto& u = s->E->globals.uev[(int)q.e[i].b];
+        if (q.at(i).kind != EV_USER) continue;
+        const auto& u = s->E->globals.uev[(int)q.at(i).b];
         if ((action == CMB_ANY_ACTION || u.fn == action) &&
             (subject == CMB_ANY_SUBJECT || u.subj == subject) &&
             (object == CMB_ANY_OBJECT || u.obj == object))
@@ -354,8 +358,8 @@ uint64_t cmb_event_pattern_cancel(cmb_sim* s, cmb_event_func* action,
     for (;;) {
         int32_t hit = -1;
         for (int32_t i = 0; i < q.n; ++i) {
-            if (q.e[i].kind != EV_USER) continue;
-            const auto& u = s->E->globals.uev[(int)q.e[i].b];
+            if (q.at(i).kind != EV_USER) continue;
+            const auto& u = s->E->globals.uev[(int)q.at(i).b];
             if ((action == CMB_ANY_ACTION || u.fn == action) &&
                 (subject == CMB_ANY_SUBJECT || u.subj == subject) &&
                 (object == CMB_ANY_OBJECT || u.obj == object)) {
@@ -364,8 +368,8 @@ uint64_t cmb_event_pattern_cancel(cmb_sim* s, cmb_event_func* action,
             }
         }
         if (hit < 0) return cnt;
-        const uint32_t h = q.e[hit].handle;
-        uev_free_slot(*s->E, (int)q.e[hit].b);
+        const uint32_t h = q.at(hit).handle;
+        uev_free_slot(*s->E, (int)q.at(hit).b);
         q.remove_at(hit);
         if (s->E->n_event_waiters)
             s->E->wake_event_waiters(h, SIG_CANCELLED);
@@ -384,13 +388,13 @@ bool cmb_event_queue_is_empty(const cmb_sim* s) { return s->E->evq.empty(); }
 void cmb_event_queue_clear(cmb_sim* s) {
     auto& q = s->E->evq;
     for (int32_t i = 0; i < q.n; ++i)
-        if (q.e[i].kind == EV_USER) uev_free_slot(*s->E, (int)q.e[i].b);
+        if (q.at(i).kind == EV_USER) uev_free_slot(*s->E, (int)q.at(i).b);
     q.reset();
 }
 static int32_t find_slot_(const cmb_sim* s, uint64_t handle) {
     const auto& q = s->E->evq;
     for (int32_t i = 0; i < q.n; ++i)
-        if (q.e[i].handle == (uint32_t)handle) return i;
+        if (q.at(i).handle == (uint32_t)handle) return i;
     return -1;
 }
 bool cmb_event_is_scheduled(const cmb_sim* s, uint64_t handle) {
@@ -398,29 +402,29 @@ bool cmb_event_is_scheduled(const cmb_sim* s, uint64_t handle) {
 }
 double cmb_event_time(const cmb_sim* s, uint64_t handle) {
     const int32_t i = find_slot_(s, handle);
-    return i >= 0 ? s->E->evq.e[i].t : -1.0;
+    return i >= 0 ? s->E->evq.at(i).t : -1.0;
 }
 int cmb_event_priority(const cmb_sim* s, uint64_t handle) {
     const int32_t i = find_slot_(s, handle);
     if (i < 0) return 0;
-    return 32767 - (int)(s->E->evq.e[i].pseq >> 48);
+    return 32767 - (int)(s->E->evq.at(i).pseq >> 48);
 }
 bool cmb_event_reprioritize(cmb_sim* s, uint64_t handle, int priority) {
     const int32_t i = find_slot_(s, handle);
     if (i < 0) return false;
-    const double t = s->E->evq.e[i].t;
+    const double t = s->E->evq.at(i).t;
     return s->E->event_reschedule((uint32_t)handle, t, priority);
 }
 uint64_t cmb_event_pattern_find(cmb_sim* s, cmb_event_func* action,
                                 void* subject, void* object) {
     const auto& q = s->E->evq;
     for (int32_t i = 0; i < q.n; ++i) {
-        if (q.e[i].kind != EV_USER) continue;
-        const auto& u = s->E->globals.uev[(int)q.e[i].b];
+        if (q.at(i).kind != EV_USER) continue;
+        const auto& u = s->E->globals.uev[(int)q.at(i).b];
         if ((action == CMB_ANY_ACTION || u.fn == action) &&
             (subject == CMB_ANY_SUBJECT || u.subj == subject) &&
             (object == CMB_ANY_OBJECT || u.obj == object))
-            return q.e[i].handle;
+            return q.at(i).handle;
     }
     return 0;
 }
@@ -1207,7 +1211,7 @@ void cmb_event_queue_print_formatted(cmb_sim* s, FILE* out,
     auto& q = s->E->evq;
     fprintf(out, "event queue @ t=%.6f: %d pending\n", s->E->now, q.n);
     for (int32_t i = 0; i < q.n; ++i) {
-        const auto& e = q.e[i];
+        const auto& e = q.at(i);
         const char* label = NULL;
         if (epf && e.kind == EV_USER) {
             const auto& u = s->E->globals.uev[(int)e.b];

@@ -28,6 +28,8 @@
 // trial-abort flag of SURVEY.md §5.3) rather than growing.
 #pragma once
 
+#include <type_traits>
+
 #include "config.hpp"
 #include "hashheap.hpp"
 #include "rng.hpp"
@@ -150,12 +152,15 @@ template <int CAP>
 struct ObjQueue {
     uint64_t ring[CAP];
     int32_t head, len;
-    int32_t limit;       // runtime capacity (<= CAP); CMB_UNLIMITED -> CAP
+    int32_t limit;       // runtime capacity; CMB_UNLIMITED -> CAP (+spill)
     int16_t g_front;     // getters wait here
     int16_t g_rear;      // putters wait here
     uint8_t recording;
     WtdSummary len_stats;  // time-weighted queue length
     double t_last;
+    // spill-tier FIFO segment (logical order: ring entries, then spill
+    // entries); active only when the model declares Cfg::SPILL_Q
+    int32_t sp_head, sp_len;
 };
 
 // one-holder resource (reference cmb_resource: binary semaphore with
@@ -228,6 +233,28 @@ struct ArrOf {
     static constexpr int n = (N > 0) ? N : 1;
 };
 
+// Optional Cfg spill capacities (default 0 = no spill tier, zero cost).
+// Models opt in by declaring SPILL_EV / SPILL_Q in their Cfg — the
+// MI355X answer to the reference's unbounded doubling growth
+// (cmi_hashheap.c:380-432): bounded fast tier in LDS/scratch, bulk
+// overflow in an HBM slab claimed on first use from a shared pool.
+template <class C, class = void>
+struct SpillEvOf {
+    static constexpr int v = 0;
+};
+template <class C>
+struct SpillEvOf<C, std::void_t<decltype(C::SPILL_EV)>> {
+    static constexpr int v = C::SPILL_EV;
+};
+template <class C, class = void>
+struct SpillQOf {
+    static constexpr int v = 0;
+};
+template <class C>
+struct SpillQOf<C, std::void_t<decltype(C::SPILL_Q)>> {
+    static constexpr int v = C::SPILL_Q;
+};
+
 // ---------------------------------------------------------------------------
 // The engine
 // ---------------------------------------------------------------------------
@@ -245,6 +272,9 @@ struct Engine {
     static constexpr int NB = Cfg::NUM_BUFS;
     static constexpr int NPQ = Cfg::NUM_PQ;
     static constexpr int NC = Cfg::NUM_COND;
+    static constexpr int SPILL_EV = SpillEvOf<Cfg>::v;
+    static constexpr int SPILL_Q = SpillQOf<Cfg>::v;
+    static constexpr bool NEEDS_SPILL = (SPILL_EV > 0) || (SPILL_Q > 0);
     static constexpr int NGUARD =
         2 * NQ + NR + NP + 2 * NB + 2 * NPQ + NC > 0
             ? 2 * NQ + NR + NP + 2 * NB + 2 * NPQ + NC
@@ -282,7 +312,18 @@ struct Engine {
     uint32_t trial_index;
     const Params* params;
     Rng rng;
-    HashHeap<Cfg::MAX_EV> evq;  // view: entries in Storage, size here
+    HashHeap<Cfg::MAX_EV, SPILL_EV> evq;  // entries in Storage, size here
+
+    // ---- spill tier (one slab per trial, claimed from a shared pool on
+    // first overflow; slabs stay claimed across a lane's later trials) --
+    struct Spill {
+        EvEntry ev[ArrOf<SPILL_EV>::n];
+        uint64_t q[ArrOf<NQ>::n][ArrOf<SPILL_Q>::n];
+    };
+    Spill* spill = nullptr;        // this trial's slab (null = none yet)
+    Spill* spill_arena = nullptr;  // pool base
+    int32_t* spill_cursor = nullptr;
+    int32_t spill_pool_cap = 0;
 
     // ---- references into storage (existing field syntax keeps working) ----
     typename Model::Globals& globals;
@@ -342,10 +383,12 @@ struct Engine {
         int g = 0;
         for (int i = 0; i < NQ; ++i) {
             ObjQueue<Cfg::QCAP>& q = queues[i];
-            q.head = 0; q.len = 0; q.limit = Cfg::QCAP;
+            q.head = 0; q.len = 0; q.limit = Cfg::QCAP + SPILL_Q;
             q.g_front = (int16_t)g++; q.g_rear = (int16_t)g++;
             q.recording = 0; q.len_stats.reset(); q.t_last = now;
+            q.sp_head = 0; q.sp_len = 0;
         }
+        if (spill) evq.attach_spill(spill->ev, SPILL_EV);  // keep held slab
         for (int i = 0; i < NR; ++i) {
             resources[i].holder = -1; resources[i].gid = (int16_t)g++;
             resources[i].recording = 0; resources[i].busy.reset();
@@ -373,6 +416,34 @@ struct Engine {
 
     static constexpr bool GUARD_MASK = (Cfg::MAX_PROC <= 64);
 
+    // direct slab attach (host runner: one slab per worker engine)
+    CMB_FORCEINLINE void set_spill(Spill* s_) {
+        spill = s_;
+        if (s_) evq.attach_spill(s_->ev, SPILL_EV);
+    }
+
+    // shared-pool attach (device launchers): claim lazily on overflow
+    CMB_FORCEINLINE void set_spill_pool(Spill* arena, int32_t* cursor,
+                                        int32_t cap) {
+        spill_arena = arena;
+        spill_cursor = cursor;
+        spill_pool_cap = cap;
+    }
+
+    CMB_FORCEINLINE bool claim_spill() {
+        if (spill) return true;
+        if (!spill_arena) return false;
+        int32_t idx;
+#ifdef __HIP_DEVICE_COMPILE__
+        idx = atomicAdd(spill_cursor, 1);
+#else
+        idx = __atomic_fetch_add(spill_cursor, 1, __ATOMIC_RELAXED);
+#endif
+        if (idx >= spill_pool_cap) return false;  // pool dry: abort path
+        set_spill(&spill_arena[idx]);
+        return true;
+    }
+
     CMB_FORCEINLINE bool guard_empty(int gid) const {
         if constexpr (GUARD_MASK) return guards[gid].wmask == 0;
         else return guards[gid].head < 0;
@@ -398,6 +469,10 @@ struct Engine {
         ev.a = a;
         ev.c = c;
         if (!evq.push(ev)) {
+            if constexpr (SPILL_EV > 0) {
+                // first overflow: claim an HBM slab and retry
+                if (claim_spill() && evq.push(ev)) return ev.handle;
+            }
             fail(ST_HEAP_FULL);
             return 0;
         }
@@ -724,8 +799,9 @@ struct Engine {
     CMB_FORCEINLINE bool eval_demand(const ProcT& p) {
         const uint32_t ctx = p.demand_ctx;
         switch (p.demand_kind) {
-            case DEM_QSPACE: return queues[ctx].len < queues[ctx].limit;
-            case DEM_QOBJ: return queues[ctx].len > 0;
+            case DEM_QSPACE:
+                return queues[ctx].len + queues[ctx].sp_len < queues[ctx].limit;
+            case DEM_QOBJ: return queues[ctx].len + queues[ctx].sp_len > 0;
             case DEM_RES: return resources[ctx].holder < 0;
             case DEM_POOL: return pools[ctx].in_use < pools[ctx].capacity;
             case DEM_BUF_GE:
@@ -794,7 +870,7 @@ struct Engine {
 
     CMB_FORCEINLINE void q_record(ObjQueue<Cfg::QCAP>& q) {
         if (q.recording) {
-            q.len_stats.add((double)q.len, now - q.t_last);
+            q.len_stats.add((double)(q.len + q.sp_len), now - q.t_last);
             q.t_last = now;
         }
     }
@@ -803,8 +879,25 @@ struct Engine {
         ObjQueue<Cfg::QCAP>& q = queues[qi];
         const bool may = p.g_granted || guard_empty(q.g_rear);
         p.g_granted = 0;
-        if (!may || q.len >= q.limit) return false;
-        if (q.len >= Cfg::QCAP) { fail(ST_QUEUE_FULL); return false; }
+        if (!may || q.len + q.sp_len >= q.limit) return false;
+        if (q.len >= Cfg::QCAP || q.sp_len > 0) {
+            // ring full, or spill already active (FIFO: the spill segment
+            // follows the ring, and gets refill the ring from its head)
+            if constexpr (SPILL_Q > 0) {
+                if (!claim_spill() || q.sp_len >= SPILL_Q) {
+                    fail(ST_QUEUE_FULL);
+                    return false;
+                }
+                q_record(q);
+                spill->q[qi][(q.sp_head + q.sp_len) % SPILL_Q] = val;
+                ++q.sp_len;
+                guard_signal(q.g_front);
+                return true;
+            } else {
+                fail(ST_QUEUE_FULL);
+                return false;
+            }
+        }
         q_record(q);
         q.ring[(q.head + q.len) % Cfg::QCAP] = val;
         ++q.len;
@@ -816,16 +909,26 @@ struct Engine {
         ObjQueue<Cfg::QCAP>& q = queues[qi];
         const bool may = p.g_granted || guard_empty(q.g_front);
         p.g_granted = 0;
-        if (!may || q.len == 0) return false;
+        if (!may || q.len + q.sp_len == 0) return false;
         q_record(q);
         *out = q.ring[q.head];
         q.head = (q.head + 1) % Cfg::QCAP;
         --q.len;
+        if constexpr (SPILL_Q > 0) {
+            if (q.sp_len > 0) {  // refill the ring tail from the spill head
+                q.ring[(q.head + q.len) % Cfg::QCAP] = spill->q[qi][q.sp_head];
+                q.sp_head = (q.sp_head + 1) % SPILL_Q;
+                --q.sp_len;
+                ++q.len;
+            }
+        }
         guard_signal(q.g_rear);
         return true;
     }
 
-    CMB_FORCEINLINE int64_t q_length(int qi) const { return queues[qi].len; }
+    CMB_FORCEINLINE int64_t q_length(int qi) const {
+        return (int64_t)queues[qi].len + queues[qi].sp_len;
+    }
 
     CMB_FORCEINLINE bool pq_try_put(int qi, ProcT& p, uint64_t val, int priority) {
         PrioQueue<Cfg::PQCAP>& q = pqueues[qi];

@@ -48,46 +48,83 @@ CMB_FORCEINLINE bool ev_less(const EvEntry& x, const EvEntry& y) {
 // HBM) while the size `n` lives in the view itself — on the device the
 // view is part of the register-resident Engine context, so the hot-loop
 // size checks and the count never round-trip through LDS.
-template <int CAP>
+//
+// SCAP > 0 enables a SECOND TIER: entries [CAP, CAP+cap2) live in a
+// spill buffer (HBM slab on the device, heap memory on the host) that
+// the engine attaches on first overflow — the MI355X counterpart of the
+// reference's unbounded doubling growth (cmi_hashheap.c:380-432).  The
+// heap is index-contiguous across the tiers, so deep (cold) entries
+// land in the slow tier naturally while the hot top stays fast.  With
+// SCAP == 0 every access compiles to the plain fast-tier load — zero
+// cost for models that never spill.
+template <int CAP, int SCAP = 0>
 struct HashHeap {
     EvEntry (&e)[CAP];
+    EvEntry* e2;   // spill tier base (null until attached)
+    int32_t cap2;  // usable spill entries (0 until attached)
     int32_t n;
 
-    CMB_FORCEINLINE explicit HashHeap(EvEntry (&buf)[CAP]) : e(buf), n(0) {}
+    CMB_FORCEINLINE explicit HashHeap(EvEntry (&buf)[CAP])
+        : e(buf), e2(nullptr), cap2(0), n(0) {}
+
+    CMB_FORCEINLINE EvEntry& at(int32_t i) {
+        if constexpr (SCAP > 0) {
+            return i < CAP ? e[i] : e2[i - CAP];
+        } else {
+            return e[i];
+        }
+    }
+    CMB_FORCEINLINE const EvEntry& at(int32_t i) const {
+        if constexpr (SCAP > 0) {
+            return i < CAP ? e[i] : e2[i - CAP];
+        } else {
+            return e[i];
+        }
+    }
+
+    CMB_FORCEINLINE void attach_spill(EvEntry* buf, int32_t cap) {
+        e2 = buf;
+        cap2 = cap < SCAP ? cap : SCAP;
+    }
+
+    CMB_FORCEINLINE int32_t capacity() const {
+        if constexpr (SCAP > 0) return CAP + cap2;
+        else return CAP;
+    }
 
     CMB_FORCEINLINE void reset() { n = 0; }
     CMB_FORCEINLINE bool empty() const { return n == 0; }
-    CMB_FORCEINLINE bool full() const { return n == CAP; }
+    CMB_FORCEINLINE bool full() const { return n == capacity(); }
     CMB_FORCEINLINE const EvEntry& top() const { return e[0]; }
 
     CMB_FORCEINLINE void sift_up(int32_t i) {
-        EvEntry tmp = e[i];
+        EvEntry tmp = at(i);
         while (i > 0) {
             const int32_t p = (i - 1) >> 1;
-            if (!ev_less(tmp, e[p])) break;
-            e[i] = e[p];
+            if (!ev_less(tmp, at(p))) break;
+            at(i) = at(p);
             i = p;
         }
-        e[i] = tmp;
+        at(i) = tmp;
     }
 
     CMB_FORCEINLINE void sift_down(int32_t i) {
-        EvEntry tmp = e[i];
+        EvEntry tmp = at(i);
         for (;;) {
             int32_t c = 2 * i + 1;
             if (c >= n) break;
-            if (c + 1 < n && ev_less(e[c + 1], e[c])) ++c;
-            if (!ev_less(e[c], tmp)) break;
-            e[i] = e[c];
+            if (c + 1 < n && ev_less(at(c + 1), at(c))) ++c;
+            if (!ev_less(at(c), tmp)) break;
+            at(i) = at(c);
             i = c;
         }
-        e[i] = tmp;
+        at(i) = tmp;
     }
 
-    // returns false when full (caller converts to a trial abort)
+    // returns false when full (caller attaches spill or aborts the trial)
     CMB_FORCEINLINE bool push(const EvEntry& ev) {
-        if (n == CAP) return false;
-        e[n] = ev;
+        if (n == capacity()) return false;
+        at(n) = ev;
         sift_up(n);
         ++n;
         return true;
@@ -97,7 +134,7 @@ struct HashHeap {
         EvEntry out = e[0];
         --n;
         if (n > 0) {
-            e[0] = e[n];
+            e[0] = at(n);
             sift_down(0);
         }
         return out;
@@ -106,8 +143,8 @@ struct HashHeap {
     // cancel by handle; O(n) scan over a small heap (see header comment)
     CMB_FORCEINLINE bool cancel(uint32_t handle, EvEntry* out = nullptr) {
         for (int32_t i = 0; i < n; ++i) {
-            if (e[i].handle == handle) {
-                if (out) *out = e[i];
+            if (at(i).handle == handle) {
+                if (out) *out = at(i);
                 remove_at(i);
                 return true;
             }
@@ -118,7 +155,7 @@ struct HashHeap {
     CMB_FORCEINLINE void remove_at(int32_t i) {
         --n;
         if (i == n) return;
-        e[i] = e[n];
+        at(i) = at(n);
         sift_down(i);
         sift_up(i);
     }
@@ -126,9 +163,9 @@ struct HashHeap {
     // reschedule (reference cmb_event_reschedule): new time, keep payload
     CMB_FORCEINLINE bool reschedule(uint32_t handle, double t, uint64_t pseq) {
         for (int32_t i = 0; i < n; ++i) {
-            if (e[i].handle == handle) {
-                e[i].t = t;
-                e[i].pseq = pseq;
+            if (at(i).handle == handle) {
+                at(i).t = t;
+                at(i).pseq = pseq;
                 sift_down(i);
                 sift_up(i);
                 return true;
@@ -145,8 +182,8 @@ struct HashHeap {
                                           uint64_t b) const {
         int32_t cnt = 0;
         for (int32_t i = 0; i < n; ++i) {
-            if ((kind == 0xFFFF || e[i].kind == kind) &&
-                (a == 0xFFFF || e[i].a == a) && (!match_b || e[i].b == b))
+            if ((kind == 0xFFFF || at(i).kind == kind) &&
+                (a == 0xFFFF || at(i).a == a) && (!match_b || at(i).b == b))
                 ++cnt;
         }
         return cnt;
@@ -162,9 +199,9 @@ struct HashHeap {
         for (;;) {
             int32_t hit = -1;
             for (int32_t i = 0; i < n; ++i) {
-                if ((kind == 0xFFFF || e[i].kind == kind) &&
-                    (a == 0xFFFF || e[i].a == a) &&
-                    (!match_b || e[i].b == b)) {
+                if ((kind == 0xFFFF || at(i).kind == kind) &&
+                    (a == 0xFFFF || at(i).a == a) &&
+                    (!match_b || at(i).b == b)) {
                     hit = i;
                     break;
                 }
@@ -178,9 +215,9 @@ struct HashHeap {
     CMB_FORCEINLINE uint32_t pattern_find(uint16_t kind, uint16_t a, bool match_b,
                                           uint64_t b) const {
         for (int32_t i = 0; i < n; ++i) {
-            if ((kind == 0xFFFF || e[i].kind == kind) &&
-                (a == 0xFFFF || e[i].a == a) && (!match_b || e[i].b == b))
-                return e[i].handle;
+            if ((kind == 0xFFFF || at(i).kind == kind) &&
+                (a == 0xFFFF || at(i).a == a) && (!match_b || at(i).b == b))
+                return at(i).handle;
         }
         return 0;  // 0 = no handle
     }

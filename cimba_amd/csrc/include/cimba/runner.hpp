@@ -71,6 +71,11 @@ RunReport run_host(const typename Model::Params& params, uint64_t master_seed,
         // event queue reset between trials, cimba.c:332)
         auto store = std::make_unique<typename Engine<Model>::Storage>();
         auto eng = std::make_unique<Engine<Model>>(*store);
+        std::unique_ptr<typename Engine<Model>::Spill> slab;
+        if constexpr (Engine<Model>::NEEDS_SPILL) {
+            slab = std::make_unique<typename Engine<Model>::Spill>();
+            eng->set_spill(slab.get());
+        }
         for (;;) {
             const uint64_t t = next.fetch_add(1, std::memory_order_relaxed);
             if (t >= ntrials) break;
