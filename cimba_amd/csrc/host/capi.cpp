@@ -34,6 +34,7 @@ struct CModel : ModelBase {
         static constexpr int MAX_PROC = C_MAX_PROC;
         static constexpr int MAX_EV = 16384;
         static constexpr int SPILL_EV = 49152;   // events grow to 64K total
+        static constexpr bool EV_MAP = true;     // O(1) cancel/reschedule
         static constexpr int TIMERS = 4;
         static constexpr int NUM_QUEUES = 16;
         static constexpr int QCAP = 8192;
@@ -392,10 +393,7 @@ void cmb_event_queue_clear(cmb_sim* s) {
     q.reset();
 }
 static int32_t find_slot_(const cmb_sim* s, uint64_t handle) {
-    const auto& q = s->E->evq;
-    for (int32_t i = 0; i < q.n; ++i)
-        if (q.at(i).handle == (uint32_t)handle) return i;
-    return -1;
+    return s->E->evq.find_index((uint32_t)handle);
 }
 bool cmb_event_is_scheduled(const cmb_sim* s, uint64_t handle) {
     return find_slot_(s, handle) >= 0;

@@ -254,6 +254,15 @@ template <class C>
 struct SpillQOf<C, std::void_t<decltype(C::SPILL_Q)>> {
     static constexpr int v = C::SPILL_Q;
 };
+// optional Cfg::EV_MAP: handle->index back-map for host-scale heaps
+template <class C, class = void>
+struct EvMapOf {
+    static constexpr bool v = false;
+};
+template <class C>
+struct EvMapOf<C, std::void_t<decltype(C::EV_MAP)>> {
+    static constexpr bool v = C::EV_MAP;
+};
 
 // ---------------------------------------------------------------------------
 // The engine
@@ -275,6 +284,8 @@ struct Engine {
     static constexpr int SPILL_EV = SpillEvOf<Cfg>::v;
     static constexpr int SPILL_Q = SpillQOf<Cfg>::v;
     static constexpr bool NEEDS_SPILL = (SPILL_EV > 0) || (SPILL_Q > 0);
+    static constexpr bool EV_MAP = EvMapOf<Cfg>::v;
+    using EvHeap = HashHeap<Cfg::MAX_EV, SPILL_EV, EV_MAP>;
     static constexpr int NGUARD =
         2 * NQ + NR + NP + 2 * NB + 2 * NPQ + NC > 0
             ? 2 * NQ + NR + NP + 2 * NB + 2 * NPQ + NC
@@ -285,6 +296,9 @@ struct Engine {
     // heap on the host.  Hot scalars live in the Engine context below.
     struct Storage {
         EvEntry evbuf[Cfg::MAX_EV];
+        EvMapSlot evmap[EV_MAP ? HashHeap<Cfg::MAX_EV, SpillEvOf<Cfg>::v,
+                                          EvMapOf<Cfg>::v>::MSIZE
+                               : 1];
         typename Model::Globals globals;
         ProcT procs[Cfg::MAX_PROC];
         Frame frames[Cfg::MAX_PROC];
@@ -312,7 +326,7 @@ struct Engine {
     uint32_t trial_index;
     const Params* params;
     Rng rng;
-    HashHeap<Cfg::MAX_EV, SPILL_EV> evq;  // entries in Storage, size here
+    EvHeap evq;  // entries in Storage, size here
 
     // ---- spill tier (one slab per trial, claimed from a shared pool on
     // first overflow; slabs stay claimed across a lane's later trials) --
@@ -342,7 +356,9 @@ struct Engine {
         : evq(s.evbuf), globals(s.globals), procs(s.procs), frames(s.frames),
           guards(s.guards), queues(s.queues), resources(s.resources),
           pools(s.pools), pool_held(s.pool_held), buffers(s.buffers),
-          pqueues(s.pqueues), conds(s.conds) {}
+          pqueues(s.pqueues), conds(s.conds) {
+        if constexpr (EV_MAP) evq.map = s.evmap;
+    }
 
     // ---- lifecycle --------------------------------------------------------
 

@@ -57,15 +57,101 @@ CMB_FORCEINLINE bool ev_less(const EvEntry& x, const EvEntry& y) {
 // land in the slow tier naturally while the hot top stays fast.  With
 // SCAP == 0 every access compiles to the plain fast-tier load — zero
 // cost for models that never spill.
-template <int CAP, int SCAP = 0>
+// MAP enables the handle->index BACK-MAP (reference cmi_hashheap.c
+// open-addressing contract, :89-120): cancel/reschedule/find become an
+// O(1) hash lookup + O(log n) sift instead of an O(n) scan.  Meant for
+// the host C API configuration (MAX_EV 16384+); device models keep
+// MAP=false and the O(small-n) scan, which wins at LDS heap sizes.
+// Open addressing, linear probing, backward-shift deletion, fixed <=50%
+// load (map is 2x total heap capacity, power of two).
+struct EvMapSlot {
+    uint32_t key;  // event handle; 0 = empty (handles start at 1)
+    int32_t idx;   // heap index
+};
+
+constexpr int32_t cmb_pow2_atleast(int32_t v) {
+    int32_t p = 2;
+    while (p < v) p <<= 1;
+    return p;
+}
+
+template <int CAP, int SCAP = 0, bool MAP = false>
 struct HashHeap {
+    static constexpr int32_t MSIZE =
+        MAP ? cmb_pow2_atleast(2 * (CAP + SCAP)) : 2;
+
     EvEntry (&e)[CAP];
-    EvEntry* e2;   // spill tier base (null until attached)
-    int32_t cap2;  // usable spill entries (0 until attached)
+    EvEntry* e2;     // spill tier base (null until attached)
+    EvMapSlot* map;  // back-map slots (storage-owned when MAP)
+    int32_t cap2;    // usable spill entries (0 until attached)
     int32_t n;
 
     CMB_FORCEINLINE explicit HashHeap(EvEntry (&buf)[CAP])
-        : e(buf), e2(nullptr), cap2(0), n(0) {}
+        : e(buf), e2(nullptr), map(nullptr), cap2(0), n(0) {}
+
+    // ---- back-map primitives (compiled out when !MAP) ----
+    static CMB_FORCEINLINE uint32_t mhash(uint32_t h) {
+        if constexpr (!MAP) return 0;
+        return (h * 2654435769u) >> (32 - __builtin_ctz((uint32_t)MSIZE));
+    }
+    CMB_FORCEINLINE void map_clear() {
+        if constexpr (MAP) {
+            for (int32_t i = 0; i < MSIZE; ++i) map[i].key = 0;
+        }
+    }
+    CMB_FORCEINLINE void map_set(uint32_t key, int32_t idx) {
+        if constexpr (MAP) {
+            uint32_t i = mhash(key);
+            for (;;) {
+                if (map[i].key == key || map[i].key == 0) {
+                    map[i].key = key;
+                    map[i].idx = idx;
+                    return;
+                }
+                i = (i + 1) & (MSIZE - 1);
+            }
+        }
+    }
+    CMB_FORCEINLINE int32_t map_get(uint32_t key) const {
+        if constexpr (!MAP) return -1;
+        uint32_t i = mhash(key);
+        for (;;) {
+            if (map[i].key == key) return map[i].idx;
+            if (map[i].key == 0) return -1;
+            i = (i + 1) & (MSIZE - 1);
+        }
+    }
+    CMB_FORCEINLINE void map_erase(uint32_t key) {
+        if constexpr (MAP) {
+            uint32_t i = mhash(key);
+            for (;;) {
+                if (map[i].key == key) break;
+                if (map[i].key == 0) return;
+                i = (i + 1) & (MSIZE - 1);
+            }
+            // backward-shift deletion keeps probe chains intact with no
+            // tombstones, so no compaction pass is ever needed
+            uint32_t j = i;
+            for (;;) {
+                map[i].key = 0;
+                for (;;) {
+                    j = (j + 1) & (MSIZE - 1);
+                    if (map[j].key == 0) return;
+                    const uint32_t h = mhash(map[j].key);
+                    if (i <= j ? (h <= i || h > j) : (h <= i && h > j))
+                        break;
+                }
+                map[i] = map[j];
+                i = j;
+            }
+        }
+    }
+
+    // placement: every heap move goes through here so the map tracks it
+    CMB_FORCEINLINE void place(int32_t i, const EvEntry& ev) {
+        at(i) = ev;
+        map_set(ev.handle, i);
+    }
 
     CMB_FORCEINLINE EvEntry& at(int32_t i) {
         if constexpr (SCAP > 0) {
@@ -92,7 +178,10 @@ struct HashHeap {
         else return CAP;
     }
 
-    CMB_FORCEINLINE void reset() { n = 0; }
+    CMB_FORCEINLINE void reset() {
+        n = 0;
+        if (map) map_clear();
+    }
     CMB_FORCEINLINE bool empty() const { return n == 0; }
     CMB_FORCEINLINE bool full() const { return n == capacity(); }
     CMB_FORCEINLINE const EvEntry& top() const { return e[0]; }
@@ -102,10 +191,10 @@ struct HashHeap {
         while (i > 0) {
             const int32_t p = (i - 1) >> 1;
             if (!ev_less(tmp, at(p))) break;
-            at(i) = at(p);
+            place(i, at(p));
             i = p;
         }
-        at(i) = tmp;
+        place(i, tmp);
     }
 
     CMB_FORCEINLINE void sift_down(int32_t i) {
@@ -115,10 +204,10 @@ struct HashHeap {
             if (c >= n) break;
             if (c + 1 < n && ev_less(at(c + 1), at(c))) ++c;
             if (!ev_less(at(c), tmp)) break;
-            at(i) = at(c);
+            place(i, at(c));
             i = c;
         }
-        at(i) = tmp;
+        place(i, tmp);
     }
 
     // returns false when full (caller attaches spill or aborts the trial)
@@ -132,46 +221,52 @@ struct HashHeap {
 
     CMB_FORCEINLINE EvEntry pop() {
         EvEntry out = e[0];
+        map_erase(out.handle);
         --n;
         if (n > 0) {
-            e[0] = at(n);
+            place(0, at(n));
             sift_down(0);
         }
         return out;
     }
 
-    // cancel by handle; O(n) scan over a small heap (see header comment)
-    CMB_FORCEINLINE bool cancel(uint32_t handle, EvEntry* out = nullptr) {
-        for (int32_t i = 0; i < n; ++i) {
-            if (at(i).handle == handle) {
-                if (out) *out = at(i);
-                remove_at(i);
-                return true;
-            }
+    // find heap index by handle: O(1) with the back-map, O(n) scan else
+    CMB_FORCEINLINE int32_t find_index(uint32_t handle) const {
+        if constexpr (MAP) {
+            const int32_t i = map_get(handle);
+            return (i >= 0 && i < n) ? i : -1;
         }
-        return false;
+        for (int32_t i = 0; i < n; ++i)
+            if (at(i).handle == handle) return i;
+        return -1;
+    }
+
+    CMB_FORCEINLINE bool cancel(uint32_t handle, EvEntry* out = nullptr) {
+        const int32_t i = find_index(handle);
+        if (i < 0) return false;
+        if (out) *out = at(i);
+        remove_at(i);
+        return true;
     }
 
     CMB_FORCEINLINE void remove_at(int32_t i) {
+        map_erase(at(i).handle);
         --n;
         if (i == n) return;
-        at(i) = at(n);
+        place(i, at(n));
         sift_down(i);
         sift_up(i);
     }
 
     // reschedule (reference cmb_event_reschedule): new time, keep payload
     CMB_FORCEINLINE bool reschedule(uint32_t handle, double t, uint64_t pseq) {
-        for (int32_t i = 0; i < n; ++i) {
-            if (at(i).handle == handle) {
-                at(i).t = t;
-                at(i).pseq = pseq;
-                sift_down(i);
-                sift_up(i);
-                return true;
-            }
-        }
-        return false;
+        const int32_t i = find_index(handle);
+        if (i < 0) return false;
+        at(i).t = t;
+        at(i).pseq = pseq;
+        sift_down(i);
+        sift_up(i);
+        return true;
     }
 
     // wildcard pattern ops over (kind, a, b) — reference
