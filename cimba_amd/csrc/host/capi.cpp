@@ -25,8 +25,40 @@ using namespace cmb;
 
 namespace {
 
-constexpr int C_MAX_PROC = 4096;
-constexpr int C_UEV = 4096;  // pending user events
+// ---- host C API envelope -------------------------
+// Build-time configurable via -DCIMBA_C_MAX_PROC=... etc.; the defaults
+// are sized for host memory (each worker thread owns one engine of this
+// shape, ~tens of MB) and far beyond the reference's typical scale.  The
+// engine arrays are still bounded — the reference grows without bound —
+// but the event heap and object queues spill to a growth tier (SPILL_*
+// below), and true exhaustion aborts the trial cleanly.
+#ifndef CIMBA_C_MAX_PROC
+#define CIMBA_C_MAX_PROC 16384
+#endif
+#ifndef CIMBA_C_UEV
+#define CIMBA_C_UEV 65536
+#endif
+#ifndef CIMBA_C_NUM_QUEUES
+#define CIMBA_C_NUM_QUEUES 64
+#endif
+#ifndef CIMBA_C_NUM_RES
+#define CIMBA_C_NUM_RES 64
+#endif
+#ifndef CIMBA_C_NUM_POOLS
+#define CIMBA_C_NUM_POOLS 64
+#endif
+#ifndef CIMBA_C_NUM_BUFS
+#define CIMBA_C_NUM_BUFS 64
+#endif
+#ifndef CIMBA_C_NUM_PQ
+#define CIMBA_C_NUM_PQ 32
+#endif
+#ifndef CIMBA_C_NUM_COND
+#define CIMBA_C_NUM_COND 64
+#endif
+
+constexpr int C_MAX_PROC = CIMBA_C_MAX_PROC;
+constexpr int C_UEV = CIMBA_C_UEV;  // pending user events
 constexpr int C_NAME = 32;  // reference CMB_PROCESS_NAMEBUF_SZ
 
 struct CModel : ModelBase {
@@ -36,15 +68,15 @@ struct CModel : ModelBase {
         static constexpr int SPILL_EV = 49152;   // events grow to 64K total
         static constexpr bool EV_MAP = true;     // O(1) cancel/reschedule
         static constexpr int TIMERS = 4;
-        static constexpr int NUM_QUEUES = 16;
+        static constexpr int NUM_QUEUES = CIMBA_C_NUM_QUEUES;
         static constexpr int QCAP = 8192;
         static constexpr int SPILL_Q = 24576;    // queues grow to 32K total
-        static constexpr int NUM_RES = 16;
-        static constexpr int NUM_POOLS = 16;
-        static constexpr int NUM_BUFS = 16;
-        static constexpr int NUM_PQ = 8;
-        static constexpr int PQCAP = 1024;
-        static constexpr int NUM_COND = 16;
+        static constexpr int NUM_RES = CIMBA_C_NUM_RES;
+        static constexpr int NUM_POOLS = CIMBA_C_NUM_POOLS;
+        static constexpr int NUM_BUFS = CIMBA_C_NUM_BUFS;
+        static constexpr int NUM_PQ = CIMBA_C_NUM_PQ;
+        static constexpr int PQCAP = 4096;
+        static constexpr int NUM_COND = CIMBA_C_NUM_COND;
     };
     struct Params {
         cmb_sim* sim;

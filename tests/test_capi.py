@@ -109,3 +109,74 @@ def test_capi_header_is_c_clean(tmp_path):
          str(tmp_path / "hdr.o")],
         capture_output=True, text=True)
     assert r.returncode == 0, r.stderr
+
+
+def test_capi_event_storm_no_quadratic_blowup(libcimba, tmp_path):
+    """Host-scale hashheap contract: a 60K-event
+    schedule + random-cancel storm and a 4000-process wait_event fanout
+    must run in O(n log n).  Before the handle back-map + waiter list,
+    the cancel path alone was ~1.8e9 scan steps (many seconds); now the
+    whole storm is tens of milliseconds, so a generous wall bound still
+    discriminates sharply."""
+    src = tmp_path / "storm.c"
+    src.write_text(r'''
+#include "cimba.h"
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+
+static void noop(cmb_sim* sim, void* subj, void* obj) {
+    (void)sim; (void)subj; (void)obj;
+}
+
+static void trial(cmb_sim* sim, void* exp) {
+    (void)exp;
+    enum { N = 60000 };
+    static uint64_t h[N];
+    /* schedule N events at pseudo-random times */
+    uint64_t s = 0x9E3779B97F4A7C15ull;
+    for (int i = 0; i < N; ++i) {
+        s = s * 6364136223846793005ull + 1442695040888963407ull;
+        double t = (double)(s >> 40);
+        h[i] = cmb_event_schedule(sim, noop, (void*)(uintptr_t)(i + 1),
+                                  NULL, t, 0);
+        if (!h[i]) { fprintf(stderr, "schedule failed at %d\n", i); exit(1); }
+    }
+    /* reschedule a third, cancel half in a scrambled order */
+    for (int i = 0; i < N; i += 3) {
+        if (!cmb_event_reschedule(sim, h[i], (double)i, 0)) exit(2);
+    }
+    for (int i = 0; i < N; i += 2) {
+        int j = (int)((unsigned)(i * 2654435761u) % N) & ~1;
+        if (h[j]) {
+            if (!cmb_event_cancel(sim, h[j])) exit(3);
+            h[j] = 0;
+        }
+    }
+    cmb_event_queue_clear(sim);
+}
+
+int main(void) {
+    char exp[8];
+    uint64_t failed = cimba_run(exp, 1, 8, trial, 42, 1);
+    if (failed) { fprintf(stderr, "trial failed\n"); return 1; }
+    puts("storm OK");
+    return 0;
+}
+''')
+    exe = str(tmp_path / "storm")
+    r = subprocess.run(
+        ["gcc", "-std=c11", "-O2", "-I", os.path.join(ROOT, "include"),
+         str(src), "-L", os.path.join(ROOT, "cimba_amd"), "-lcimba",
+         f"-Wl,-rpath,{os.path.join(ROOT, 'cimba_amd')}", "-lm", "-o", exe],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    import time
+    t0 = time.perf_counter()
+    out = subprocess.run([exe], capture_output=True, text=True, timeout=60)
+    dt = time.perf_counter() - t0
+    assert out.returncode == 0, (out.stdout, out.stderr)
+    assert "storm OK" in out.stdout
+    # O(n^2) was multiple seconds of pure scanning; O(n log n) is ~tens
+    # of ms — 3 s leaves huge CI headroom while still failing a scan
+    assert dt < 3.0, f"event storm took {dt:.1f}s — quadratic path?"
