@@ -33,7 +33,9 @@ constexpr float PI_F = 3.14159265358979f;
 // One 64-target tile of complex beamforming on MFMA.
 // Input per lane: its A-operand row is target (base + sub*16 + (lane&15)),
 // K-group lane>>4; output per lane: D rows (lane>>4)*4+r, beam col lane&15
-// (the standard 16x16x4 C/D map).  Accumulates detections into det/pow.
+// (the standard 16x16x4 C/D map).  Writes the RAW best-beam power into
+// g.bf[t]; composition (RCS, path loss, clutter pipeline) happens in the
+// separate detection phases so the same tile serves both modes.
 __device__ void beamform_tile(GlA& g, uint32_t trial, uint32_t dwl,
                               double snr_ref, int base, int lane,
                               unsigned long long* det_local,
@@ -72,19 +74,11 @@ __device__ void beamform_tile(GlA& g, uint32_t trial, uint32_t dwl,
             if (col == 0) {
                 const int t = base + sub * 16 + kgrp * 4 + r;
                 if (t < g.nt) {
-                    const float r2 =
-                        g.x[t] * g.x[t] + g.y[t] * g.y[t] + 1.0f;
-                    const float power = p * g.rcs[t] / (r2 * r2);
+                    g.bf[t] = p;  // raw best-beam power
                     if (pow_out) {
-                        pow_out[t] = power;
-                    }
-                    if (snr_ref > 0.0) {
-                        if (AWACS::detect_draw(trial, dwl, (uint32_t)t,
-                                               power, snr_ref)) {
-                            g.det_cnt[t] += 1u;
-                            *det_local += 1ull;
-                        }
-                        *pow_local += (double)power;
+                        const float r2 =
+                            g.x[t] * g.x[t] + g.y[t] * g.y[t] + 1.0f;
+                        pow_out[t] = p * g.rcs[t] / (r2 * r2);
                     }
                 }
             }
@@ -92,12 +86,22 @@ __device__ void beamform_tile(GlA& g, uint32_t trial, uint32_t dwl,
     }
 }
 
+// wave-uniform fold to lane 0's value (the host tree_sum64 order), then
+// broadcast — deterministic and identical to the host reference bitwise
+__device__ __forceinline__ float wave_fold_sum(float v) {
+    for (int w = 32; w >= 1; w >>= 1) v += __shfl_xor(v, w);
+    return __int_as_float(
+        __builtin_amdgcn_readfirstlane(__float_as_int(v)));
+}
+
 // dt/nt/area/trial/dwl/snr are wave-uniform: computed on lane 0 and
 // broadcast via readfirstlane so no lane reads the engine context
-__device__ void dwell_physics_wave(GlA& g, uint32_t trial, uint32_t dwl,
+__device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
+                                   uint32_t trial, uint32_t dwl,
                                    double snr_ref, int lane, float dt,
                                    int nt, float area, double now,
-                                   float* pow_out) {
+                                   float* pow_out, int* surv) {
+    // ---- kinematics (lane-parallel; terrain-following altitude) ----
     for (int t = lane; t < nt; t += 64) {
         g.x[t] += g.vx[t] * dt;
         g.y[t] += g.vy[t] * dt;
@@ -105,21 +109,144 @@ __device__ void dwell_physics_wave(GlA& g, uint32_t trial, uint32_t dwl,
         if (g.x[t] < -area) g.x[t] += 2.0f * area;
         if (g.y[t] > area) g.y[t] -= 2.0f * area;
         if (g.y[t] < -area) g.y[t] += 2.0f * area;
+        if (P.use_terrain)
+            g.alt[t] = cmb::th_sample(P.terrain, P.tdesc, g.x[t], g.y[t]) +
+                       (float)P.target_height;
     }
-    // (no barrier needed: one wave, lockstep)
+    // ---- MFMA beamforming: raw best-beam power into g.bf[] ----
     unsigned long long det_local = 0;
     double pow_local = 0.0;
     for (int base = 0; base < nt; base += 64)
         beamform_tile(g, trial, dwl, snr_ref, base, lane, &det_local,
                       &pow_local, pow_out);
-    // reduce the per-lane accumulators (only col==0 lanes are nonzero)
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+
+    if (!P.use_terrain) {
+        // legacy free-space mode: lane-parallel compose + draw
+        for (int t = lane; t < nt; t += 64) {
+            const float r2 = g.x[t] * g.x[t] + g.y[t] * g.y[t] + 1.0f;
+            const float power = g.bf[t] * g.rcs[t] / (r2 * r2);
+            if (snr_ref > 0.0 &&
+                AWACS::detect_draw(trial, dwl, (uint32_t)t, power,
+                                   snr_ref)) {
+                g.det_cnt[t] += 1u;
+                det_local += 1ull;
+            }
+            pow_local += (double)power;
+        }
+        for (int w = 32; w >= 1; w >>= 1) {
+            det_local += __shfl_xor((unsigned long long)det_local, w);
+            pow_local += __shfl_xor(pow_local, w);
+        }
+        if (lane == 0) {
+            g.detections += det_local;
+            g.sum_power += pow_local;
+            g.last_t = now;
+            g.dwells += 1u;
+        }
+        return;
+    }
+
+    // ---- full pipeline: triage -> LOS -> clutter/CFAR -> draw ----
+    const float bdir = AWACS::beam_dir_at(P, now);
+    const float halfgate =
+        0.5f * (float)(P.beamwidth + P.rot_rate * P.dwell);
+    // triage with ORDER-PRESERVING ballot compaction (ascending t, the
+    // host iteration order): illuminated -> horizon -> survivor list
+    int nsurv = 0;
+    unsigned long long illum_local = 0;
+    for (int base = 0; base < nt; base += 64) {
+        const int t = base + lane;
+        bool pass = false;
+        bool illum = false;
+        if (t < nt) {
+            // diagnostic sum_power accumulation (free-space compose)
+            const float r2 = g.x[t] * g.x[t] + g.y[t] * g.y[t] + 1.0f;
+            pow_local += (double)(g.bf[t] * g.rcs[t] / (r2 * r2));
+            const float az = atan2f(g.y[t], g.x[t]);
+            if (AWACS::in_beam(az, bdir, halfgate)) {
+                illum = true;
+                const float r2d =
+                    sqrtf(g.x[t] * g.x[t] + g.y[t] * g.y[t]);
+                const float terr_t = g.alt[t] - (float)P.target_height;
+                pass = !AWACS::beyond_horizon(
+                    r2d, (float)P.sensor_alt - terr_t,
+                    (float)P.target_height);
+            }
+        }
+        illum_local += illum ? 1ull : 0ull;
+        const unsigned long long m = __ballot(pass);
+        const int rank = __popcll(m & ((1ull << lane) - 1ull));
+        if (pass) surv[nsurv + rank] = t;
+        nsurv += __popcll(m);
+    }
+
+    // survivor loop: whole wave cooperates on one target at a time
+    unsigned long long shield_local = 0;
+    double clut_local = 0.0;
+    for (int si = 0; si < nsurv; ++si) {
+        const int t = __builtin_amdgcn_readfirstlane(surv[si]);
+        const float tx = g.x[t], ty = g.y[t], ta = g.alt[t];
+        const float r2d = sqrtf(tx * tx + ty * ty);
+        // LOS march, samples strided across the 64 lanes
+        const int nst = AWACS::los_steps(P, r2d);
+        bool blocked = false;
+        for (int k = lane; k < nst && !blocked; k += 64)
+            blocked = cmb::th_los_blocked_at(P.terrain, P.tdesc, 0.0f,
+                                             0.0f, (float)P.sensor_alt,
+                                             tx, ty, ta, nst, k);
+        if (__any(blocked)) {
+            shield_local += 1ull;
+            continue;
+        }
+        // clutter: test cell + CA-CFAR reference cells, each cell one
+        // wave-parallel 64-sample integral folded in host tree order
+        const float e_c =
+            wave_fold_sum(AWACS::clutter_partial(P, r2d, bdir, lane));
+        const float dr = (float)P.range_res;
+        float sum = 0.0f;
+        int used = 0;
+        for (int k = P.cfar_nguard + 1;
+             k <= P.cfar_nguard + P.cfar_nref; ++k) {
+            const float rlo = r2d - (float)k * dr;
+            const float rhi = r2d + (float)k * dr;
+            if (rlo > dr) {
+                sum += wave_fold_sum(
+                    AWACS::clutter_partial(P, rlo, bdir, lane));
+                ++used;
+            }
+            sum += wave_fold_sum(
+                AWACS::clutter_partial(P, rhi, bdir, lane));
+            ++used;
+        }
+        const float mean = used > 0 ? sum / (float)used : 0.0f;
+        const float thr =
+            (float)P.cfar_alpha * (mean + (float)P.noise_floor);
+        // multipath + energy + draw: uniform scalar math on every lane
+        const float mp = AWACS::multipath_gain(P, tx, ty, ta, r2d);
+        const float e_t =
+            AWACS::target_energy(P, g.bf[t], g.rcs[t], r2d, ta, mp);
+        const float pd =
+            AWACS::detect_pd(e_t, e_c, (float)P.noise_floor, thr);
+        clut_local += (double)e_c;
+        if (lane == 0) {
+            if (AWACS::draw_u01(trial, dwl, (uint32_t)t) < pd) {
+                g.det_cnt[t] += 1u;
+                g.detections += 1u;
+            }
+        }
+    }
+    // fold diagnostics; counters from lane-local accumulators
     for (int w = 32; w >= 1; w >>= 1) {
-        det_local += __shfl_xor((unsigned long long)det_local, w);
         pow_local += __shfl_xor(pow_local, w);
+        illum_local += __shfl_xor((unsigned long long)illum_local, w);
     }
     if (lane == 0) {
-        g.detections += det_local;
         g.sum_power += pow_local;
+        g.illuminated += illum_local;
+        g.shielded += shield_local;
+        g.sum_clutter += clut_local;
         g.last_t = now;
         g.dwells += 1u;
     }
@@ -165,6 +292,7 @@ __global__ __launch_bounds__(256) void awacs_kernel(
     const AWACS::Params* __restrict__ dP, uint64_t master_seed,
     uint64_t trial_base, uint32_t ntrials, AWACS::Result* __restrict__ out,
     StA* __restrict__ stores, float* __restrict__ dbg, int scalar_phys) {
+    __shared__ int surv_lds[4][AWACS::MAX_T];  // per-wave survivor lists
     const int lane = (int)(threadIdx.x & 63);
     const uint32_t wslot =
         blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
@@ -197,24 +325,32 @@ __global__ __launch_bounds__(256) void awacs_kernel(
             if (!req) break;
             int mynt = 0;
             float mydt = 0.0f;
-            float mynow = 0.0f;
+            unsigned long long mynow = 0;
             if (lane == 0) {
                 mynt = E.globals.nt;
                 mydt = (float)(E.now - E.globals.last_t);
-                mynow = (float)E.now;
+                mynow = (unsigned long long)__double_as_longlong(E.now);
             }
             const int nt = __builtin_amdgcn_readfirstlane(mynt);
             const float dt = __int_as_float(
                 __builtin_amdgcn_readfirstlane(__float_as_int(mydt)));
-            const double now_b = (double)__int_as_float(
-                __builtin_amdgcn_readfirstlane(__float_as_int(mynow)));
+            // broadcast the FULL double clock (beam direction must match
+            // the host's fmod(rot_rate * now) bitwise)
+            const int now_lo =
+                __builtin_amdgcn_readfirstlane((int)(mynow & 0xFFFFFFFFull));
+            const int now_hi =
+                __builtin_amdgcn_readfirstlane((int)(mynow >> 32));
+            const double now_b = __longlong_as_double(
+                ((long long)now_hi << 32) | (unsigned int)now_lo);
             float* dbg_now = (dbg && trial == 0 && dwl == 0) ? dbg : nullptr;
             if (scalar_phys) {
                 if (lane == 0) AWACS::physics_all(E);
             } else {
-                dwell_physics_wave(stores[wslot].globals, trial, dwl,
+                dwell_physics_wave(*dP, stores[wslot].globals,
+                                   (uint32_t)(trial_base + trial), dwl,
                                    dP->snr_ref, lane, dt, nt,
-                                   (float)dP->area, now_b, dbg_now);
+                                   (float)dP->area, now_b, dbg_now,
+                                   surv_lds[threadIdx.x >> 6]);
             }
             WAVE_FENCE();
             engine_resume_phase(E, lane);
@@ -261,6 +397,18 @@ __global__ __launch_bounds__(64) void awacs_power_kernel_devinit(
         hipError_t err_ = (x);                        \
         if (err_ != hipSuccess) return (int)err_;     \
     } while (0)
+
+// shared read-only heightmap for the trial batch (one per launch; the
+// fixed seed makes it identical to the host cache in bindings.cpp)
+__global__ __launch_bounds__(256) void awacs_terrain_fill(
+    float* __restrict__ h, cmb::TerrainDesc T) {
+    const size_t n = (size_t)T.cols * T.rows;
+    const size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        h[i] = cmb::th_texel_height(T, (int32_t)(i % T.cols),
+                                    (int32_t)(i / T.cols));
+}
 
 __global__ __launch_bounds__(64) void xlane_repro_kernel(int* buf,
                                                           int iters) {
@@ -314,7 +462,16 @@ int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                         uint64_t trial_base, int device, double* elapsed_ms,
                         void* results_out) {
     HIP_TRY(hipSetDevice(device));
-    const AWACS::Params& P = *(const AWACS::Params*)params;
+    AWACS::Params P = *(const AWACS::Params*)params;  // local: terrain ptr
+    float* d_terr = nullptr;
+    if (P.use_terrain) {
+        const size_t n = (size_t)P.tdesc.cols * P.tdesc.rows;
+        HIP_TRY(hipMalloc(&d_terr, n * sizeof(float)));
+        hipLaunchKernelGGL(awacs_terrain_fill, dim3(2048), dim3(256), 0, 0,
+                           d_terr, P.tdesc);
+        HIP_TRY(hipGetLastError());
+        P.terrain = d_terr;
+    }
     const uint32_t want_waves = (uint32_t)ntrials;
     const uint32_t nwaves = want_waves < 8192u ? want_waves : 8192u;
     const uint32_t blocks = (nwaves + 3) / 4;
@@ -346,6 +503,7 @@ int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
     HIP_TRY(hipFree(d_P));
     HIP_TRY(hipFree(d_out));
     HIP_TRY(hipFree(d_eng));
+    if (d_terr) HIP_TRY(hipFree(d_terr));
     HIP_TRY(hipEventDestroy(t0));
     HIP_TRY(hipEventDestroy(t1));
     return 0;

@@ -24,6 +24,7 @@
 #pragma once
 
 #include "../include/cimba/engine.hpp"
+#include "../include/cimba/terrain.hpp"
 
 #include <math.h>
 
@@ -68,10 +69,30 @@ struct AWACS : cmb::ModelBase {
         double dwell;          // dwell period (reference: 0.04 s)
         double maneuver_mean;  // mean time between target maneuvers
         int32_t ntargets;
-        int32_t pad_;
+        int32_t use_terrain;   // 1 = full pipeline (triage/LOS/clutter/CFAR)
         double area;           // half-width of the surveillance box [m]
         double speed;          // target speed scale [m/s]
-        double snr_ref;        // SNR at reference range
+        double snr_ref;        // SNR scale at reference range
+        // ---- radar pipeline (reference tut_5_2.cu capability set:
+        // rotating beam triage, horizon, terrain LOS, constant-gamma
+        // clutter, CA-CFAR, specular multipath) ----
+        double sensor_alt;     // platform altitude [m]
+        double rot_rate;       // beam rotation [rad/s]
+        double beamwidth;      // [rad]
+        double range_res;      // range cell depth [m]
+        double cfar_alpha;     // threshold multiplier
+        int32_t cfar_nref;     // reference cells per side
+        int32_t cfar_nguard;   // guard cells per side
+        double noise_floor;    // normalized
+        double gamma0;         // constant-gamma clutter scale
+        double rough_m;        // surface roughness sigma_h [m]
+        double wavelength;     // [m]
+        double target_height;  // target height above terrain [m]
+        // shared read-only heightmap (device ptr on GPU, host ptr on CPU;
+        // built once per device from the SAME seed — trials share terrain
+        // exactly as the reference keeps one terrain per device)
+        const float* terrain;
+        cmb::TerrainDesc tdesc;
     };
 
     struct Result {
@@ -79,7 +100,10 @@ struct AWACS : cmb::ModelBase {
         uint64_t dwells;
         uint64_t maneuvers;
         uint64_t events;
-        double sum_power;  // accumulated best-beam power (diagnostic)
+        double sum_power;     // accumulated best-beam power (diagnostic)
+        uint64_t illuminated; // targets passing the beam gate (triage)
+        uint64_t shielded;    // LOS-blocked among illuminated
+        double sum_clutter;   // accumulated test-cell clutter (diagnostic)
         int32_t status;
         int32_t pad_;
     };
@@ -93,6 +117,8 @@ struct AWACS : cmb::ModelBase {
     struct Globals {
         // target SoA (f32 for the MFMA path)
         float x[MAX_T], y[MAX_T], vx[MAX_T], vy[MAX_T], rcs[MAX_T];
+        float alt[MAX_T];  // terrain height + target_height at (x, y)
+        float bf[MAX_T];   // raw best-beam beamforming power (per dwell)
         uint32_t det_cnt[MAX_T];
         // beam steering weights w[e][b], complex
         float wr[ELEM][BEAMS], wi[ELEM][BEAMS];
@@ -101,7 +127,10 @@ struct AWACS : cmb::ModelBase {
         uint64_t detections;
         uint64_t dwells;
         uint64_t maneuvers;
+        uint64_t illuminated;
+        uint64_t shielded;
         double sum_power;
+        double sum_clutter;
         double last_t;  // time of previous dwell (kinematics dt)
     };
 
@@ -118,10 +147,10 @@ struct AWACS : cmb::ModelBase {
         return (double)(h >> 11) * 0x1.0p-53;
     }
 
-    // scalar single-target dwell physics (host path AND the numerics
-    // reference for the device MFMA path) — f32 math to match MFMA
-    CMB_FORCEINLINE static float target_power(const Globals& g, int t) {
-        const float r2 = g.x[t] * g.x[t] + g.y[t] * g.y[t] + 1.0f;
+    // scalar single-target beamforming (host path AND the numerics
+    // reference for the device MFMA path) — f32 math to match MFMA.
+    // Returns the RAW best-beam power |sum_e a_e conj(w_eb)|^2.
+    CMB_FORCEINLINE static float target_bf(const Globals& g, int t) {
         const float az = atan2f(g.y[t], g.x[t]);
         const float s = sinf(az);
         float ar[ELEM], ai[ELEM];
@@ -141,8 +170,18 @@ struct AWACS : cmb::ModelBase {
             const float p = re * re + im * im;
             best = p > best ? p : best;
         }
-        // normalized beamforming gain x RCS / r^4 path loss
+        return best;
+    }
+
+    // legacy free-space composition: gain x RCS / r^4 path loss
+    CMB_FORCEINLINE static float compose_power(const Globals& g, int t,
+                                               float best) {
+        const float r2 = g.x[t] * g.x[t] + g.y[t] * g.y[t] + 1.0f;
         return best * g.rcs[t] / (r2 * r2);
+    }
+
+    CMB_FORCEINLINE static float target_power(const Globals& g, int t) {
+        return compose_power(g, t, target_bf(g, t));
     }
 
     template <class E_>
@@ -157,6 +196,10 @@ struct AWACS : cmb::ModelBase {
         if (g.x[t] < -a) g.x[t] += 2.0f * a;
         if (g.y[t] > a) g.y[t] -= 2.0f * a;
         if (g.y[t] < -a) g.y[t] += 2.0f * a;
+        if (E.params->use_terrain)  // ground target rides the terrain
+            g.alt[t] = cmb::th_sample(E.params->terrain, E.params->tdesc,
+                                      g.x[t], g.y[t]) +
+                       (float)E.params->target_height;
     }
 
     // one target's detection draw for this dwell (shared by the scalar
@@ -170,22 +213,265 @@ struct AWACS : cmb::ModelBase {
         return draw_u01(trial, dwell, t) < pd;
     }
 
+
+    // =====================================================================
+    // Radar detection pipeline (reference capability set, tut_5_2.cu
+    // triage_kernel :1198 / raymarch_kernel :1304 / integrate_cell_clutter
+    // :810 / cfar_threshold :894 / multipath_gain_dev :966 — behaviors
+    // reproduced with our own geometry, biome and sampling design).
+    // Every function is f32, CMB_FORCEINLINE, host+device: the host path
+    // is the bitwise numerics reference (clutter reductions use the same
+    // 64-partial tree as the device wave reduce).
+    // =====================================================================
+
+    static constexpr float PI_ = 3.14159265358979f;
+    static constexpr float KE_RE = 4.0f / 3.0f * 6371000.0f;  // eff. Earth
+
+    CMB_FORCEINLINE static float beam_dir_at(const Params& P, double now) {
+        return (float)fmod(P.rot_rate * now, 2.0 * 3.14159265358979323846);
+    }
+
+    // beam gate: target azimuth within the sector the beam sweeps during
+    // this dwell (reference target_in_beam semantics)
+    CMB_FORCEINLINE static bool in_beam(float az, float bdir,
+                                        float halfgate) {
+        float d = az - bdir;
+        while (d > PI_) d -= 2.0f * PI_;
+        while (d < -PI_) d += 2.0f * PI_;
+        return fabsf(d) <= halfgate;
+    }
+
+    // 4/3-Earth radar horizon from the two effective heights
+    CMB_FORCEINLINE static bool beyond_horizon(float r2d, float hs,
+                                               float ht) {
+        const float dh = sqrtf(fmaxf(hs, 0.0f) * 2.0f * KE_RE) +
+                         sqrtf(fmaxf(ht, 0.0f) * 2.0f * KE_RE);
+        return r2d > dh;
+    }
+
+    // Gaussian two-way antenna pattern at angular offset from beam center
+    CMB_FORCEINLINE static float antenna_gain_sq(float off, float bw) {
+        const float a = 2.7725887f * (off * off) / (bw * bw);  // 4 ln 2
+        const float g = expf(-a);
+        return g * g;
+    }
+
+    // constant-gamma clutter coefficient by terrain altitude (our biome
+    // parameterization: low ground scatters more than high rock)
+    CMB_FORCEINLINE static float biome_gamma(const Params& P, float alt) {
+        const float hi = fmaxf((float)P.tdesc.base + (float)P.tdesc.amp,
+                               1.0f);
+        float t = alt / hi;
+        t = t < 0.0f ? 0.0f : (t > 1.0f ? 1.0f : t);
+        return (float)P.gamma0 * (1.2f - 0.8f * t);
+    }
+
+    // clutter cell sampling grid: 64 samples (one per lane on the device)
+    static constexpr int CL_NR = 4;   // range samples
+    static constexpr int CL_NC = 16;  // cross-range samples
+
+    // one lane's partial of the clutter integral over an annular cell
+    // centered at (center_range, bdir): samples s = lane, lane+64, ...
+    // with s -> (i = s / CL_NC, j = s % CL_NC).  sigma0*G^2*dA / R^4 per
+    // sample (reference integrate_cell_clutter model).
+    CMB_FORCEINLINE static float clutter_partial(const Params& P,
+                                                 float center_range,
+                                                 float bdir, int lane) {
+        const float dr = (float)P.range_res;
+        const float bw = (float)P.beamwidth;
+        const float sa = (float)P.sensor_alt;
+        const float dA_unit = (dr * bw) / (float)(CL_NR * CL_NC);
+        float acc = 0.0f;
+        for (int sidx = lane; sidx < CL_NR * CL_NC; sidx += 64) {
+            const int i = sidx / CL_NC, j = sidx % CL_NC;
+            const float tr = ((float)i + 0.5f) / (float)CL_NR;
+            const float R = fmaxf(1.0f, center_range - 0.5f * dr + tr * dr);
+            const float inv_R4 = 1.0f / (R * R * R * R);
+            const float tc = ((float)j + 0.5f) / (float)CL_NC;
+            const float off = -0.5f * bw + tc * bw;
+            const float az = bdir + off;
+            const float wx = R * cosf(az);
+            const float wy = R * sinf(az);
+            const float terr = cmb::th_sample(P.terrain, P.tdesc, wx, wy);
+            // curvature drop lifts the sample above the sensor's tangent
+            const float drop = (R * R) / (2.0f * KE_RE);
+            const float h_eff = sa - terr - drop;
+            if (h_eff <= 0.0f) continue;  // below local horizon
+            const float sin_graze = fminf(1.0f, h_eff / R);
+            const float sigma0 = biome_gamma(P, terr) * sin_graze;
+            const float g2 = antenna_gain_sq(off, bw);
+            acc += sigma0 * g2 * (R * dA_unit) * inv_R4;
+        }
+        return acc;
+    }
+
+    // deterministic 64-partial binary-tree fold — the EXACT order of the
+    // device's __shfl_xor wave reduction, so host == device bitwise
+    CMB_FORCEINLINE static float tree_sum64(float* v) {
+        for (int w = 32; w >= 1; w >>= 1)
+            for (int l = 0; l < w; ++l) v[l] += v[l + w];
+        return v[0];
+    }
+
+    CMB_FORCEINLINE static float clutter_cell_host(const Params& P,
+                                                   float center_range,
+                                                   float bdir) {
+        float part[64];
+        for (int l = 0; l < 64; ++l)
+            part[l] = clutter_partial(P, center_range, bdir, l);
+        return tree_sum64(part);
+    }
+
+    // CA-CFAR: mean clutter over 2*n_ref reference cells (skipping
+    // n_guard each side of the test cell), times alpha, plus noise
+    // (reference cfar_threshold contract).  Host form; the device
+    // integrates each cell wave-parallel and folds identically.
+    CMB_FORCEINLINE static float cfar_threshold_host(const Params& P,
+                                                     float target_range,
+                                                     float bdir) {
+        const float dr = (float)P.range_res;
+        float sum = 0.0f;
+        int used = 0;
+        for (int k = P.cfar_nguard + 1;
+             k <= P.cfar_nguard + P.cfar_nref; ++k) {
+            const float rlo = target_range - (float)k * dr;
+            const float rhi = target_range + (float)k * dr;
+            if (rlo > dr) {
+                sum += clutter_cell_host(P, rlo, bdir);
+                ++used;
+            }
+            sum += clutter_cell_host(P, rhi, bdir);
+            ++used;
+        }
+        const float mean = used > 0 ? sum / (float)used : 0.0f;
+        return (float)P.cfar_alpha * (mean + (float)P.noise_floor);
+    }
+
+    // specular two-way multipath factor (reference multipath_gain_dev
+    // capability): one-iteration specular point refinement, Rayleigh
+    // roughness suppression, pi phase flip on reflection
+    CMB_FORCEINLINE static float multipath_gain(const Params& P, float tx,
+                                                float ty, float ta,
+                                                float r2d) {
+        const float sa = (float)P.sensor_alt;
+        const float t0 = sa / fmaxf(sa + ta, 1.0f);
+        float gx = tx * t0, gy = ty * t0;
+        float galt = cmb::th_sample(P.terrain, P.tdesc, gx, gy);
+        const float hs0 = sa - galt, ht0 = ta - galt;
+        if (hs0 <= 1.0f || ht0 <= 1.0f) return 1.0f;
+        const float t1 = hs0 / (hs0 + ht0);
+        gx = tx * t1;
+        gy = ty * t1;
+        galt = cmb::th_sample(P.terrain, P.tdesc, gx, gy);
+        const float hs = sa - galt, ht = ta - galt;
+        if (hs <= 1.0f || ht <= 1.0f) return 1.0f;
+        const float d_dir = sqrtf(r2d * r2d + (hs - ht) * (hs - ht));
+        const float b1 = sqrtf(t1 * r2d * (t1 * r2d) + hs * hs);
+        const float b2 = sqrtf((1.0f - t1) * r2d * ((1.0f - t1) * r2d) +
+                               ht * ht);
+        const float delta = b1 + b2 - d_dir;
+        const float sin_graze = hs / fmaxf(1.0f, b1);
+        // Rayleigh roughness: smooth low ground reflects, rough high rock
+        // does not (same biome scale as the clutter gamma)
+        const float arg = (4.0f * PI_ * (float)P.rough_m * sin_graze) /
+                          (float)P.wavelength;
+        const float rho = 0.9f * expf(-arg * arg);
+        if (rho < 0.01f) return 1.0f;
+        const float phase =
+            2.0f * PI_ * delta / (float)P.wavelength + PI_;
+        const float one_way = 1.0f + rho * rho + 2.0f * rho * cosf(phase);
+        const float two_way = one_way * one_way;
+        return fmaxf(two_way, 1.0e-3f);
+    }
+
+    // LOS march step count: <= half a terrain cell per step, capped so a
+    // dwell's march is bounded (identical host/device => identical masks)
+    CMB_FORCEINLINE static int los_steps(const Params& P, float r2d) {
+        const float step = 0.5f * fminf(P.tdesc.dx, P.tdesc.dy);
+        int n = (int)(r2d / step);
+        return n < 2048 ? n : 2048;
+    }
+
+    // per-dwell, per-illuminated-target detection decision given the
+    // already-computed pieces; returns pd (the draw itself stays with the
+    // caller so device lane 0 and host share draw_u01 exactly)
+    CMB_FORCEINLINE static float detect_pd(float e_target, float e_clutter,
+                                           float noise, float threshold) {
+        const float cell = e_target + e_clutter + noise;
+        const float m = cell / fmaxf(threshold, 1.0e-30f);
+        // logistic on the CFAR margin (reference's probabilistic step)
+        return 1.0f / (1.0f + expf(-6.0f * (m - 1.0f)));
+    }
+
+    // target signal energy: MFMA/scalar beamforming best-beam power
+    // (normalized), Swerling RCS, R^-4, multipath
+    CMB_FORCEINLINE static float target_energy(const Params& P, float bf,
+                                               float rcs, float r2d,
+                                               float ta, float mp) {
+        const float dz = (float)P.sensor_alt - ta;
+        const float r2 = r2d * r2d + dz * dz;
+        const float bfn = bf * (1.0f / (float)(ELEM * ELEM));
+        return (float)P.snr_ref * bfn * bfn * rcs * mp / (r2 * r2);
+    }
+
     // host-path dwell: scalar over all targets (device: awacs_kernel.hip
     // runs the same math wave-parallel with MFMA beamforming)
     template <class E_>
     CMB_FORCEINLINE static void physics_all(E_& E) {
         Globals& g = E.globals;
+        const Params& P = *E.params;
         const float dt = (float)(E.now - g.last_t);
         g.last_t = E.now;
+        if (!P.use_terrain) {  // legacy free-space mode (r01 behavior)
+            for (int t = 0; t < g.nt; ++t) {
+                advance_target(E, t, dt);
+                const float p = target_power(g, t);
+                if (detect_draw(E.trial_index, g.dwells, (uint32_t)t, p,
+                                P.snr_ref)) {
+                    g.det_cnt[t] += 1u;
+                    g.detections += 1u;
+                }
+                g.sum_power += (double)p;
+            }
+            g.dwells += 1u;
+            return;
+        }
+        // ---- full pipeline: triage -> LOS -> clutter/CFAR -> draw ----
+        const float bdir = beam_dir_at(P, E.now);
+        const float halfgate =
+            0.5f * (float)(P.beamwidth + P.rot_rate * P.dwell);
         for (int t = 0; t < g.nt; ++t) {
             advance_target(E, t, dt);
-            const float p = target_power(g, t);
-            if (detect_draw(E.trial_index, g.dwells, (uint32_t)t, p,
-                            E.params->snr_ref)) {
+            const float bf = target_bf(g, t);
+            const float p = compose_power(g, t, bf);  // diagnostic
+            g.sum_power += (double)p;
+            const float az = atan2f(g.y[t], g.x[t]);
+            if (!in_beam(az, bdir, halfgate)) continue;  // triage: beam
+            g.illuminated += 1u;
+            const float r2d = sqrtf(g.x[t] * g.x[t] + g.y[t] * g.y[t]);
+            const float terr_t = g.alt[t] - (float)P.target_height;
+            if (beyond_horizon(r2d, (float)P.sensor_alt - terr_t,
+                               (float)P.target_height))
+                continue;  // triage: horizon
+            const int nst = los_steps(P, r2d);
+            if (!cmb::th_los_clear(P.terrain, P.tdesc, 0.0f, 0.0f,
+                                   (float)P.sensor_alt, g.x[t], g.y[t],
+                                   g.alt[t], nst)) {
+                g.shielded += 1u;
+                continue;  // terrain masked
+            }
+            const float mp = multipath_gain(P, g.x[t], g.y[t], g.alt[t],
+                                            r2d);
+            const float e_t =
+                target_energy(P, bf, g.rcs[t], r2d, g.alt[t], mp);
+            const float e_c = clutter_cell_host(P, r2d, bdir);
+            const float thr = cfar_threshold_host(P, r2d, bdir);
+            g.sum_clutter += (double)e_c;
+            const float pd = detect_pd(e_t, e_c, (float)P.noise_floor, thr);
+            if (draw_u01(E.trial_index, g.dwells, (uint32_t)t) < pd) {
                 g.det_cnt[t] += 1u;
                 g.detections += 1u;
             }
-            g.sum_power += (double)p;
         }
         g.dwells += 1u;
     }
@@ -234,7 +520,10 @@ struct AWACS : cmb::ModelBase {
         g.detections = 0;
         g.dwells = 0;
         g.maneuvers = 0;
+        g.illuminated = 0;
+        g.shielded = 0;
         g.sum_power = 0.0;
+        g.sum_clutter = 0.0;
         g.last_t = 0.0;
         // beams uniformly over sin-space; w[e][b] = exp(i*pi*e*sin_b)
         for (int b = 0; b < BEAMS; ++b) {
@@ -253,6 +542,12 @@ struct AWACS : cmb::ModelBase {
             g.vx[t] = sp * (float)cos(ang);
             g.vy[t] = sp * (float)sin(ang);
             g.rcs[t] = (float)E.rng.exponential(1.0);  // Swerling-1 RCS
+            g.alt[t] = P.use_terrain
+                           ? cmb::th_sample(P.terrain, P.tdesc, g.x[t],
+                                            g.y[t]) +
+                                 (float)P.target_height
+                           : 0.0f;
+            g.bf[t] = 0.0f;
             g.det_cnt[t] = 0;
             E.schedule(EV_MANEUVER, (uint16_t)t, 0, 0,
                        E.rng.exponential(P.maneuver_mean), 0);
@@ -268,6 +563,9 @@ struct AWACS : cmb::ModelBase {
         r.maneuvers = E.globals.maneuvers;
         r.events = E.ev_dispatched;
         r.sum_power = E.globals.sum_power;
+        r.illuminated = E.globals.illuminated;
+        r.shielded = E.globals.shielded;
+        r.sum_clutter = E.globals.sum_clutter;
         r.status = E.status;
     }
 };

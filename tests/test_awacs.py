@@ -62,3 +62,66 @@ def test_awacs_mfma_beamforming_numerics():
     assert np.abs(f32 - f64).max() / scale < 5e-5
     # and the MFMA path is not silently the host path: exact zeros match
     assert np.abs(dev - f32).max() / scale < 5e-5
+
+
+def test_awacs_pipeline_host_stats():
+    """Full radar pipeline (terrain masking, clutter, CA-CFAR, multipath):
+    triage keeps ~beamwidth/2pi of targets per dwell, mountainous terrain
+    shields a large fraction, and CFAR discrimination keeps det/clear off
+    the saturated 1.0 (the r01 gap: no clutter/CFAR/multipath at all)."""
+    r = ca._C.awacs_host(ntrials=4, duration=10.0, ntargets=1000, seed=42,
+                         threads=4)
+    assert r["trials_ok"] == 4
+    illum = r["total_illuminated"]
+    shield = r["total_shielded"]
+    det = r["total_detections"]
+    clear = illum - shield
+    # beam gate: ~(beamwidth + sweep)/2pi * nt per dwell ~ 8
+    per_dwell = illum / r["total_dwells"]
+    assert 3.0 < per_dwell < 20.0, per_dwell
+    assert 0.30 < shield / illum < 0.95  # terrain really masks
+    assert 0.10 < det / clear < 0.95     # CFAR really discriminates
+    assert r["sum_clutter"] > 0.0
+    assert det <= clear
+
+
+def test_awacs_terrain_off_legacy_mode():
+    """terrain=0 keeps the r01 free-space behavior (no triage counters)."""
+    r = ca._C.awacs_host(ntrials=2, duration=5.0, ntargets=200, seed=7,
+                         threads=2, terrain=0)
+    assert r["trials_ok"] == 2
+    assert r["total_illuminated"] == 0
+    assert r["total_shielded"] == 0
+    assert r["total_detections"] > 0
+
+
+def test_awacs_pipeline_deterministic():
+    a = ca._C.awacs_host(ntrials=2, duration=4.0, ntargets=500, seed=9,
+                         threads=1)
+    b = ca._C.awacs_host(ntrials=2, duration=4.0, ntargets=500, seed=9,
+                         threads=2)
+    for k in ("total_detections", "total_illuminated", "total_shielded",
+              "sum_clutter", "sum_power"):
+        assert a[k] == b[k], k
+
+
+@pytest.mark.gpu
+def test_awacs_pipeline_gpu_matches_host():
+    """Device pipeline (wave-cooperative LOS + lane-parallel clutter with
+    host-tree-order folds) vs the host scalar reference.  libm-vs-OCML
+    transcendentals differ by ulps, so boundary decisions (beam gate, LOS,
+    draws) may flip rarely — tight relative tolerances, not bitwise."""
+    g = ca._C.awacs_gpu(ntrials=8, duration=8.0, ntargets=1000, seed=21,
+                        device=0)
+    h = ca._C.awacs_host(ntrials=8, duration=8.0, ntargets=1000, seed=21,
+                         threads=0)
+    assert g["trials_ok"] == 8
+    assert g["total_dwells"] == h["total_dwells"]
+    assert g["total_events"] == h["total_events"]
+    assert g["total_illuminated"] == h["total_illuminated"]  # geometry
+    rel = lambda a, b: abs(a - b) / max(b, 1)
+    assert rel(g["total_shielded"], h["total_shielded"]) < 0.01
+    assert rel(g["total_detections"], h["total_detections"]) < 0.03
+    assert abs(g["sum_clutter"] - h["sum_clutter"]) < 1e-6 * max(
+        h["sum_clutter"], 1e-30)
+    assert g["total_shielded"] > 0
