@@ -38,10 +38,18 @@ def build():
     g["jobshop"] = {"events": r["total_events"],
                     "makespan": dhex(r["mean_makespan"])}
     r = ca._C.awacs_host(ntrials=2, duration=2.0, ntargets=128, seed=SEED,
-                         threads=2)
+                         threads=2, terrain=0)
     g["awacs"] = {"events": r["total_events"],
                   "detections": r["total_detections"],
                   "power": dhex(r["sum_power"])}
+    r = ca._C.awacs_host(ntrials=2, duration=2.0, ntargets=128, seed=SEED,
+                         threads=2, terrain=1)
+    g["awacs_terrain"] = {"events": r["total_events"],
+                          "detections": r["total_detections"],
+                          "illuminated": r["total_illuminated"],
+                          "shielded": r["total_shielded"],
+                          "clutter": dhex(r["sum_clutter"]),
+                          "power": dhex(r["sum_power"])}
     g["scenarios"] = {str(w): ca._C.scenario_host(w)["trace"]
                       for w in list(range(1, 14)) + [15, 16, 17, 18, 19]}
     t = ca._C.terrain_host(64, 48, base=50.0, amp=400.0, octaves=5,
