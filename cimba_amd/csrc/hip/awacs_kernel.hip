@@ -109,9 +109,7 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
         if (g.x[t] < -area) g.x[t] += 2.0f * area;
         if (g.y[t] > area) g.y[t] -= 2.0f * area;
         if (g.y[t] < -area) g.y[t] += 2.0f * area;
-        if (P.use_terrain)
-            g.alt[t] = cmb::th_sample(P.terrain, P.tdesc, g.x[t], g.y[t]) +
-                       (float)P.target_height;
+        // altitude sampled lazily at triage (illuminated targets only)
     }
     // ---- MFMA beamforming: raw best-beam power into g.bf[] ----
     unsigned long long det_local = 0;
@@ -167,6 +165,9 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
             const float az = atan2f(g.y[t], g.x[t]);
             if (AWACS::in_beam(az, bdir, halfgate)) {
                 illum = true;
+                g.alt[t] =
+                    cmb::th_sample(P.terrain, P.tdesc, g.x[t], g.y[t]) +
+                    (float)P.target_height;
                 const float r2d =
                     sqrtf(g.x[t] * g.x[t] + g.y[t] * g.y[t]);
                 const float terr_t = g.alt[t] - (float)P.target_height;
@@ -189,14 +190,31 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
         const int t = __builtin_amdgcn_readfirstlane(surv[si]);
         const float tx = g.x[t], ty = g.y[t], ta = g.alt[t];
         const float r2d = sqrtf(tx * tx + ty * ty);
-        // LOS march, samples strided across the 64 lanes
+        // LOS march from the TARGET end in 64-lane rounds; the chord
+        // altitude rises monotonically toward the sensor, so a round
+        // whose every live sample clears the terrain ceiling proves the
+        // rest clear (exact — see AWACS::los_clear_fast)
         const int nst = AWACS::los_steps(P, r2d);
-        bool blocked = false;
-        for (int k = lane; k < nst && !blocked; k += 64)
-            blocked = cmb::th_los_blocked_at(P.terrain, P.tdesc, 0.0f,
-                                             0.0f, (float)P.sensor_alt,
-                                             tx, ty, ta, nst, k);
-        if (__any(blocked)) {
+        const float zmax = P.tdesc.base + P.tdesc.amp;
+        bool shielded_t = false;
+        for (int base = nst - 1; base >= 0; base -= 64) {
+            const int k = base - lane;
+            bool blocked = false;
+            bool below = false;
+            if (k >= 0) {
+                below = AWACS::los_z_at(P, ta, nst, k) <= zmax;
+                if (below)
+                    blocked = cmb::th_los_blocked_at(
+                        P.terrain, P.tdesc, 0.0f, 0.0f,
+                        (float)P.sensor_alt, tx, ty, ta, nst, k);
+            }
+            if (__any(blocked)) {
+                shielded_t = true;
+                break;
+            }
+            if (__ballot(below) == 0) break;  // rest of the ray is higher
+        }
+        if (shielded_t) {
             shield_local += 1ull;
             continue;
         }

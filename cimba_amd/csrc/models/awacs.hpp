@@ -196,10 +196,9 @@ struct AWACS : cmb::ModelBase {
         if (g.x[t] < -a) g.x[t] += 2.0f * a;
         if (g.y[t] > a) g.y[t] -= 2.0f * a;
         if (g.y[t] < -a) g.y[t] += 2.0f * a;
-        if (E.params->use_terrain)  // ground target rides the terrain
-            g.alt[t] = cmb::th_sample(E.params->terrain, E.params->tdesc,
-                                      g.x[t], g.y[t]) +
-                       (float)E.params->target_height;
+        // altitude is sampled LAZILY at triage time for illuminated
+        // targets only (the only consumers) — saves a terrain bilinear
+        // per target per dwell on both paths
     }
 
     // one target's detection draw for this dwell (shared by the scalar
@@ -392,6 +391,34 @@ struct AWACS : cmb::ModelBase {
         return n < 2048 ? n : 2048;
     }
 
+    // ray altitude at sample k of the sensor->target chord
+    CMB_FORCEINLINE static float los_z_at(const Params& P, float ta,
+                                          int nsteps, int k) {
+        const float t = (float)(k + 1) / (float)(nsteps + 1);
+        return (float)P.sensor_alt + (ta - (float)P.sensor_alt) * t;
+    }
+
+    // Exact-result fast LOS: march from the TARGET end (high k, low ray)
+    // toward the sensor; the chord altitude rises monotonically, so once
+    // a sample clears the terrain ceiling (base + amp bounds th_sample)
+    // every remaining sample is clear.  A sample above the ceiling can
+    // never be blocked, so the early-outs change nothing in the result —
+    // they only skip provably-clear work.  The device kernel does the
+    // same in 64-lane rounds with ballot votes.
+    CMB_FORCEINLINE static bool los_clear_fast(const Params& P, float tx,
+                                               float ty, float ta,
+                                               int nsteps) {
+        const float zmax = P.tdesc.base + P.tdesc.amp;
+        for (int k = nsteps - 1; k >= 0; --k) {
+            if (los_z_at(P, ta, nsteps, k) > zmax) return true;
+            if (cmb::th_los_blocked_at(P.terrain, P.tdesc, 0.0f, 0.0f,
+                                       (float)P.sensor_alt, tx, ty, ta,
+                                       nsteps, k))
+                return false;
+        }
+        return true;
+    }
+
     // per-dwell, per-illuminated-target detection decision given the
     // already-computed pieces; returns pd (the draw itself stays with the
     // caller so device lane 0 and host share draw_u01 exactly)
@@ -448,15 +475,16 @@ struct AWACS : cmb::ModelBase {
             const float az = atan2f(g.y[t], g.x[t]);
             if (!in_beam(az, bdir, halfgate)) continue;  // triage: beam
             g.illuminated += 1u;
+            // lazy altitude: only illuminated targets sample the terrain
+            g.alt[t] = cmb::th_sample(P.terrain, P.tdesc, g.x[t], g.y[t]) +
+                       (float)P.target_height;
             const float r2d = sqrtf(g.x[t] * g.x[t] + g.y[t] * g.y[t]);
             const float terr_t = g.alt[t] - (float)P.target_height;
             if (beyond_horizon(r2d, (float)P.sensor_alt - terr_t,
                                (float)P.target_height))
                 continue;  // triage: horizon
             const int nst = los_steps(P, r2d);
-            if (!cmb::th_los_clear(P.terrain, P.tdesc, 0.0f, 0.0f,
-                                   (float)P.sensor_alt, g.x[t], g.y[t],
-                                   g.alt[t], nst)) {
+            if (!los_clear_fast(P, g.x[t], g.y[t], g.alt[t], nst)) {
                 g.shielded += 1u;
                 continue;  // terrain masked
             }
