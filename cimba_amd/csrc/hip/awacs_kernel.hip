@@ -100,7 +100,7 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
                                    uint32_t trial, uint32_t dwl,
                                    double snr_ref, int lane, float dt,
                                    int nt, float area, double now,
-                                   float* pow_out, int* surv) {
+                                   float* pow_out, int* surv, int probe) {
     // ---- kinematics (lane-parallel; terrain-following altitude) ----
     for (int t = lane; t < nt; t += 64) {
         g.x[t] += g.vx[t] * dt;
@@ -114,9 +114,10 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
     // ---- MFMA beamforming: raw best-beam power into g.bf[] ----
     unsigned long long det_local = 0;
     double pow_local = 0.0;
-    for (int base = 0; base < nt; base += 64)
-        beamform_tile(g, trial, dwl, snr_ref, base, lane, &det_local,
-                      &pow_local, pow_out);
+    if (!(probe & 8))
+        for (int base = 0; base < nt; base += 64)
+            beamform_tile(g, trial, dwl, snr_ref, base, lane, &det_local,
+                          &pow_local, pow_out);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
 
@@ -186,6 +187,7 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
     // survivor loop: whole wave cooperates on one target at a time
     unsigned long long shield_local = 0;
     double clut_local = 0.0;
+    if (probe & 1) nsurv = 0;  // probe: skip the survivor loop entirely
     for (int si = 0; si < nsurv; ++si) {
         const int t = __builtin_amdgcn_readfirstlane(surv[si]);
         const float tx = g.x[t], ty = g.y[t], ta = g.alt[t];
@@ -197,6 +199,7 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
         const int nst = AWACS::los_steps(P, r2d);
         const float zmax = P.tdesc.base + P.tdesc.amp;
         bool shielded_t = false;
+        if (probe & 4) goto after_los;  // probe: skip LOS
         for (int base = nst - 1; base >= 0; base -= 64) {
             const int k = base - lane;
             bool blocked = false;
@@ -214,6 +217,7 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
             }
             if (__ballot(below) == 0) break;  // rest of the ray is higher
         }
+    after_los:
         if (shielded_t) {
             shield_local += 1ull;
             continue;
@@ -221,11 +225,13 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
         // clutter: test cell + CA-CFAR reference cells, each cell one
         // wave-parallel 64-sample integral folded in host tree order
         const float e_c =
-            wave_fold_sum(AWACS::clutter_partial(P, r2d, bdir, lane));
+            (probe & 2) ? 0.0f
+                        : wave_fold_sum(
+                              AWACS::clutter_partial(P, r2d, bdir, lane));
         const float dr = (float)P.range_res;
         float sum = 0.0f;
         int used = 0;
-        for (int k = P.cfar_nguard + 1;
+        for (int k = (probe & 2) ? 999 : P.cfar_nguard + 1;
              k <= P.cfar_nguard + P.cfar_nref; ++k) {
             const float rlo = r2d - (float)k * dr;
             const float rhi = r2d + (float)k * dr;
@@ -368,7 +374,8 @@ __global__ __launch_bounds__(256) void awacs_kernel(
                                    (uint32_t)(trial_base + trial), dwl,
                                    dP->snr_ref, lane, dt, nt,
                                    (float)dP->area, now_b, dbg_now,
-                                   surv_lds[threadIdx.x >> 6]);
+                                   surv_lds[threadIdx.x >> 6],
+                                   scalar_phys >> 8);
             }
             WAVE_FENCE();
             engine_resume_phase(E, lane);
@@ -507,9 +514,11 @@ int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
     const char* sp = getenv("CIMBA_AWACS_SCALAR");
+    const char* pr = getenv("CIMBA_AWACS_PROBE");  // perf phase toggles
     hipLaunchKernelGGL(awacs_kernel, dim3(blocks), dim3(256), 0, 0, d_P,
                        seed, trial_base, (uint32_t)ntrials, d_out, d_eng,
-                       (float*)nullptr, sp ? atoi(sp) : 0);
+                       (float*)nullptr,
+                       (sp ? atoi(sp) : 0) | ((pr ? atoi(pr) : 0) << 8));
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
