@@ -367,7 +367,7 @@ __global__ __launch_bounds__(256) void awacs_kernel(
             const double now_b = __longlong_as_double(
                 ((long long)now_hi << 32) | (unsigned int)now_lo);
             float* dbg_now = (dbg && trial == 0 && dwl == 0) ? dbg : nullptr;
-            if (scalar_phys) {
+            if (scalar_phys & 0xFF) {
                 if (lane == 0) AWACS::physics_all(E);
             } else {
                 dwell_physics_wave(*dP, stores[wslot].globals,

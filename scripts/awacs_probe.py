@@ -5,7 +5,7 @@ def run(tag, probe):
     os.environ["CIMBA_AWACS_PROBE"] = str(probe)
     try:
         t0 = time.perf_counter()
-        r = ca._C.awacs_gpu(ntrials=300, duration=600.0, ntargets=1000, seed=5, device=0)
+        r = ca._C.awacs_gpu(ntrials=300, duration=240.0, ntargets=1000, seed=5, device=0)
         print(f"{tag:26s}: {time.perf_counter()-t0:6.2f}s det={r['total_detections']}", flush=True)
     finally:
         del os.environ["CIMBA_AWACS_PROBE"]
