@@ -86,6 +86,46 @@ __device__ void beamform_tile(GlA& g, uint32_t trial, uint32_t dwl,
     }
 }
 
+// beamform_tile over an INDEX LIST (the LOS-clear survivors): same MFMA
+// structure, rows gathered through idx[]; writes g.bf[idx[row]]
+__device__ void beamform_tile_idx(GlA& g, const int* __restrict__ idx,
+                                  int cnt, int base, int lane) {
+    const int col = lane & 15;
+    const int kgrp = lane >> 4;
+    for (int sub = 0; sub < 4; ++sub) {
+        const int row = base + sub * 16 + col;
+        const bool valid_row = row < cnt;
+        const int trow = valid_row ? idx[row] : 0;
+        float saz = 0.0f;
+        if (valid_row) saz = sinf(atan2f(g.y[trow], g.x[trow]));
+        f32x4 acc_re = {0.f, 0.f, 0.f, 0.f};
+        f32x4 acc_im = {0.f, 0.f, 0.f, 0.f};
+        for (int sgrp = 0; sgrp < 4; ++sgrp) {
+            const int e = sgrp * 4 + kgrp;
+            const float ph = PI_F * (float)e * saz;
+            const float ar = valid_row ? cosf(ph) : 0.0f;
+            const float ai = valid_row ? sinf(ph) : 0.0f;
+            const float bwr = g.wr[e][col];
+            const float bwi = g.wi[e][col];
+            acc_re = __builtin_amdgcn_mfma_f32_16x16x4f32(ar, bwr, acc_re, 0, 0, 0);
+            acc_re = __builtin_amdgcn_mfma_f32_16x16x4f32(ai, bwi, acc_re, 0, 0, 0);
+            acc_im = __builtin_amdgcn_mfma_f32_16x16x4f32(ai, bwr, acc_im, 0, 0, 0);
+            acc_im = __builtin_amdgcn_mfma_f32_16x16x4f32(-ar, bwi, acc_im, 0, 0, 0);
+        }
+        for (int r = 0; r < 4; ++r) {
+            float p = acc_re[r] * acc_re[r] + acc_im[r] * acc_im[r];
+            for (int w = 8; w >= 1; w >>= 1) {
+                const float o = __shfl_xor(p, w, 16);
+                p = o > p ? o : p;
+            }
+            if (col == 0) {
+                const int rrow = base + sub * 16 + kgrp * 4 + r;
+                if (rrow < cnt) g.bf[idx[rrow]] = p;
+            }
+        }
+    }
+}
+
 // wave-uniform fold to lane 0's value (the host tree_sum64 order), then
 // broadcast — deterministic and identical to the host reference bitwise
 __device__ __forceinline__ float wave_fold_sum(float v) {
@@ -147,12 +187,13 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
         return;
     }
 
-    // ---- full pipeline: triage -> LOS -> clutter/CFAR -> draw ----
+    // ---- full pipeline: fused kinematics+triage -> LOS -> MFMA bf on
+    // clear survivors -> clutter/CFAR -> draw (expensive work follows
+    // the triage funnel; all-target beamforming measured 48% of the
+    // dwell before this restructure) ----
     const float bdir = AWACS::beam_dir_at(P, now);
     const float halfgate =
         0.5f * (float)(P.beamwidth + P.rot_rate * P.dwell);
-    // triage with ORDER-PRESERVING ballot compaction (ascending t, the
-    // host iteration order): illuminated -> horizon -> survivor list
     int nsurv = 0;
     unsigned long long illum_local = 0;
     for (int base = 0; base < nt; base += 64) {
@@ -160,9 +201,13 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
         bool pass = false;
         bool illum = false;
         if (t < nt) {
-            // diagnostic sum_power accumulation (free-space compose)
-            const float r2 = g.x[t] * g.x[t] + g.y[t] * g.y[t] + 1.0f;
-            pow_local += (double)(g.bf[t] * g.rcs[t] / (r2 * r2));
+            // kinematics fused into the triage pass
+            g.x[t] += g.vx[t] * dt;
+            g.y[t] += g.vy[t] * dt;
+            if (g.x[t] > area) g.x[t] -= 2.0f * area;
+            if (g.x[t] < -area) g.x[t] += 2.0f * area;
+            if (g.y[t] > area) g.y[t] -= 2.0f * area;
+            if (g.y[t] < -area) g.y[t] += 2.0f * area;
             const float az = atan2f(g.y[t], g.x[t]);
             if (AWACS::in_beam(az, bdir, halfgate)) {
                 illum = true;
@@ -184,18 +229,16 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
         nsurv += __popcll(m);
     }
 
-    // survivor loop: whole wave cooperates on one target at a time
+    // LOS per survivor; compact the CLEAR list in place (ascending order
+    // preserved — the host iteration order)
     unsigned long long shield_local = 0;
     double clut_local = 0.0;
     if (probe & 1) nsurv = 0;  // probe: skip the survivor loop entirely
+    int nclear = 0;
     for (int si = 0; si < nsurv; ++si) {
         const int t = __builtin_amdgcn_readfirstlane(surv[si]);
         const float tx = g.x[t], ty = g.y[t], ta = g.alt[t];
         const float r2d = sqrtf(tx * tx + ty * ty);
-        // LOS march from the TARGET end in 64-lane rounds; the chord
-        // altitude rises monotonically toward the sensor, so a round
-        // whose every live sample clears the terrain ceiling proves the
-        // rest clear (exact — see AWACS::los_clear_fast)
         const int nst = AWACS::los_steps(P, r2d);
         const float zmax = P.tdesc.base + P.tdesc.amp;
         bool shielded_t = false;
@@ -220,10 +263,25 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
     after_los:
         if (shielded_t) {
             shield_local += 1ull;
-            continue;
+        } else {
+            if (lane == 0) surv[nclear] = t;
+            ++nclear;
         }
-        // clutter: test cell + CA-CFAR reference cells, each cell one
-        // wave-parallel 64-sample integral folded in host tree order
+    }
+
+    // MFMA beamforming batched over the clear survivors
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    for (int base = 0; base < nclear; base += 64)
+        beamform_tile_idx(g, surv, nclear, base, lane);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+
+    // clutter / CFAR / multipath / draw per clear survivor
+    for (int si = 0; si < nclear; ++si) {
+        const int t = __builtin_amdgcn_readfirstlane(surv[si]);
+        const float tx = g.x[t], ty = g.y[t], ta = g.alt[t];
+        const float r2d = sqrtf(tx * tx + ty * ty);
         const float e_c =
             (probe & 2) ? 0.0f
                         : wave_fold_sum(
@@ -247,27 +305,27 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
         const float mean = used > 0 ? sum / (float)used : 0.0f;
         const float thr =
             (float)P.cfar_alpha * (mean + (float)P.noise_floor);
-        // multipath + energy + draw: uniform scalar math on every lane
         const float mp = AWACS::multipath_gain(P, tx, ty, ta, r2d);
+        const float bf = g.bf[t];
         const float e_t =
-            AWACS::target_energy(P, g.bf[t], g.rcs[t], r2d, ta, mp);
+            AWACS::target_energy(P, bf, g.rcs[t], r2d, ta, mp);
         const float pd =
             AWACS::detect_pd(e_t, e_c, (float)P.noise_floor, thr);
         clut_local += (double)e_c;
         if (lane == 0) {
+            const float r2 = tx * tx + ty * ty + 1.0f;
+            g.sum_power += (double)(bf * g.rcs[t] / (r2 * r2));
             if (AWACS::draw_u01(trial, dwl, (uint32_t)t) < pd) {
                 g.det_cnt[t] += 1u;
                 g.detections += 1u;
             }
         }
     }
-    // fold diagnostics; counters from lane-local accumulators
-    for (int w = 32; w >= 1; w >>= 1) {
-        pow_local += __shfl_xor(pow_local, w);
+    // fold the lane-strided triage counter; per-survivor accumulators
+    // were already uniform / lane-0-owned
+    for (int w = 32; w >= 1; w >>= 1)
         illum_local += __shfl_xor((unsigned long long)illum_local, w);
-    }
     if (lane == 0) {
-        g.sum_power += pow_local;
         g.illuminated += illum_local;
         g.shielded += shield_local;
         g.sum_clutter += clut_local;

@@ -463,15 +463,16 @@ struct AWACS : cmb::ModelBase {
             g.dwells += 1u;
             return;
         }
-        // ---- full pipeline: triage -> LOS -> clutter/CFAR -> draw ----
+        // ---- full pipeline: triage -> LOS -> MFMA bf -> clutter/draw.
+        // Beamforming and the power diagnostic run only for LOS-CLEAR
+        // survivors: the expensive work follows the triage funnel
+        // exactly as the reference's triage->raymarch shape intends
+        // (measured: all-target beamforming was 48% of the dwell).
         const float bdir = beam_dir_at(P, E.now);
         const float halfgate =
             0.5f * (float)(P.beamwidth + P.rot_rate * P.dwell);
         for (int t = 0; t < g.nt; ++t) {
             advance_target(E, t, dt);
-            const float bf = target_bf(g, t);
-            const float p = compose_power(g, t, bf);  // diagnostic
-            g.sum_power += (double)p;
             const float az = atan2f(g.y[t], g.x[t]);
             if (!in_beam(az, bdir, halfgate)) continue;  // triage: beam
             g.illuminated += 1u;
@@ -488,6 +489,8 @@ struct AWACS : cmb::ModelBase {
                 g.shielded += 1u;
                 continue;  // terrain masked
             }
+            const float bf = target_bf(g, t);  // clear survivors only
+            g.sum_power += (double)compose_power(g, t, bf);
             const float mp = multipath_gain(P, g.x[t], g.y[t], g.alt[t],
                                             r2d);
             const float e_t =
