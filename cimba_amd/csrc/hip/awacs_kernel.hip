@@ -141,28 +141,25 @@ __device__ void dwell_physics_wave(const AWACS::Params& P, GlA& g,
                                    double snr_ref, int lane, float dt,
                                    int nt, float area, double now,
                                    float* pow_out, int* surv, int probe) {
-    // ---- kinematics (lane-parallel; terrain-following altitude) ----
-    for (int t = lane; t < nt; t += 64) {
-        g.x[t] += g.vx[t] * dt;
-        g.y[t] += g.vy[t] * dt;
-        if (g.x[t] > area) g.x[t] -= 2.0f * area;
-        if (g.x[t] < -area) g.x[t] += 2.0f * area;
-        if (g.y[t] > area) g.y[t] -= 2.0f * area;
-        if (g.y[t] < -area) g.y[t] += 2.0f * area;
-        // altitude sampled lazily at triage (illuminated targets only)
-    }
-    // ---- MFMA beamforming: raw best-beam power into g.bf[] ----
     unsigned long long det_local = 0;
     double pow_local = 0.0;
-    if (!(probe & 8))
-        for (int base = 0; base < nt; base += 64)
-            beamform_tile(g, trial, dwl, snr_ref, base, lane, &det_local,
-                          &pow_local, pow_out);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-
     if (!P.use_terrain) {
-        // legacy free-space mode: lane-parallel compose + draw
+        // ---- legacy free-space mode: kinematics, all-target MFMA
+        // beamforming, compose + draw ----
+        for (int t = lane; t < nt; t += 64) {
+            g.x[t] += g.vx[t] * dt;
+            g.y[t] += g.vy[t] * dt;
+            if (g.x[t] > area) g.x[t] -= 2.0f * area;
+            if (g.x[t] < -area) g.x[t] += 2.0f * area;
+            if (g.y[t] > area) g.y[t] -= 2.0f * area;
+            if (g.y[t] < -area) g.y[t] += 2.0f * area;
+        }
+        if (!(probe & 8))
+            for (int base = 0; base < nt; base += 64)
+                beamform_tile(g, trial, dwl, snr_ref, base, lane,
+                              &det_local, &pow_local, pow_out);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
         for (int t = lane; t < nt; t += 64) {
             const float r2 = g.x[t] * g.x[t] + g.y[t] * g.y[t] + 1.0f;
             const float power = g.bf[t] * g.rcs[t] / (r2 * r2);
