@@ -439,6 +439,244 @@ __global__ __launch_bounds__(256) void awacs_kernel(
     }
 }
 
+
+// ---------------------------------------------------------------------------
+// Block-cooperative AWACS: ONE TRIAL PER BLOCK (the north star's literal
+// mapping), 4 waves sharing the dwell physics.  In the published-point
+// regime (300 trials) the wave-per-trial kernel leaves ~96% of the chip
+// idle and per-dwell LATENCY is the whole metric; splitting the triage
+// stripes and the survivor pipeline across the block's waves cuts that
+// latency ~nw-fold.  Wave 0 lane 0 still owns the sequential DES engine.
+// Per-survivor math is wave-local and identical to the wave kernel (host
+// tree-order folds), so detections match the host to gate-ulps; the
+// double diagnostics combine per-wave partials in wave order.
+// ---------------------------------------------------------------------------
+__device__ void dwell_physics_block(
+    const AWACS::Params& P, GlA& g, uint32_t trial, uint32_t dwl, int lane,
+    int wv, int nw, float dt, int nt, float area, double now,
+    int* __restrict__ surv, int* __restrict__ dst,
+    int* __restrict__ cnts, unsigned long long* __restrict__ ull_acc,
+    double* __restrict__ dbl_acc) {
+    const float bdir = AWACS::beam_dir_at(P, now);
+    const float halfgate =
+        0.5f * (float)(P.beamwidth + P.rot_rate * P.dwell);
+    // ---- triage over contiguous per-wave chunks (ascending order) ----
+    const int chunk = ((nt + nw * 64 - 1) / (nw * 64)) * 64;
+    const int lo = wv * chunk;
+    const int hi = lo + chunk < nt ? lo + chunk : nt;
+    int mycnt = 0;
+    unsigned long long illum_local = 0;
+    for (int base = lo; base < hi; base += 64) {
+        const int t = base + lane;
+        bool pass = false;
+        bool illum = false;
+        if (t < hi) {
+            g.x[t] += g.vx[t] * dt;
+            g.y[t] += g.vy[t] * dt;
+            if (g.x[t] > area) g.x[t] -= 2.0f * area;
+            if (g.x[t] < -area) g.x[t] += 2.0f * area;
+            if (g.y[t] > area) g.y[t] -= 2.0f * area;
+            if (g.y[t] < -area) g.y[t] += 2.0f * area;
+            const float az = atan2f(g.y[t], g.x[t]);
+            if (AWACS::in_beam(az, bdir, halfgate)) {
+                illum = true;
+                g.alt[t] =
+                    cmb::th_sample(P.terrain, P.tdesc, g.x[t], g.y[t]) +
+                    (float)P.target_height;
+                const float r2d =
+                    sqrtf(g.x[t] * g.x[t] + g.y[t] * g.y[t]);
+                const float terr_t = g.alt[t] - (float)P.target_height;
+                pass = !AWACS::beyond_horizon(
+                    r2d, (float)P.sensor_alt - terr_t,
+                    (float)P.target_height);
+            }
+        }
+        illum_local += illum ? 1ull : 0ull;
+        const unsigned long long m = __ballot(pass);
+        const int rank = __popcll(m & ((1ull << lane) - 1ull));
+        if (pass) surv[lo + mycnt + rank] = t;
+        mycnt += __popcll(m);
+    }
+    for (int w = 32; w >= 1; w >>= 1)
+        illum_local += __shfl_xor((unsigned long long)illum_local, w);
+    if (lane == 0) {
+        cnts[wv] = mycnt;
+        ull_acc[wv] = illum_local;
+    }
+    __syncthreads();
+    // concatenate the per-chunk lists (ascending t preserved)
+    int off = 0, nsurv = 0;
+    for (int w = 0; w < nw; ++w) {
+        if (w == wv) off = nsurv;
+        nsurv += cnts[w];
+    }
+    for (int i = lane; i < cnts[wv]; i += 64) dst[off + i] = surv[lo + i];
+    __syncthreads();
+
+    // ---- survivor pipeline, one survivor per WAVE at a time ----
+    unsigned long long shield_local = 0, det_local = 0;
+    double clut_local = 0.0, pow_local = 0.0;
+    for (int si = wv; si < nsurv; si += nw) {
+        const int t = __builtin_amdgcn_readfirstlane(dst[si]);
+        const float tx = g.x[t], ty = g.y[t], ta = g.alt[t];
+        const float r2d = sqrtf(tx * tx + ty * ty);
+        const int nst = AWACS::los_steps(P, r2d);
+        const float zmax = P.tdesc.base + P.tdesc.amp;
+        bool shielded_t = false;
+        for (int base = nst - 1; base >= 0; base -= 64) {
+            const int k = base - lane;
+            bool blocked = false;
+            bool below = false;
+            if (k >= 0) {
+                below = AWACS::los_z_at(P, ta, nst, k) <= zmax;
+                if (below)
+                    blocked = cmb::th_los_blocked_at(
+                        P.terrain, P.tdesc, 0.0f, 0.0f,
+                        (float)P.sensor_alt, tx, ty, ta, nst, k);
+            }
+            if (__any(blocked)) {
+                shielded_t = true;
+                break;
+            }
+            if (__ballot(below) == 0) break;
+        }
+        if (shielded_t) {
+            shield_local += 1ull;
+            continue;
+        }
+        // per-survivor MFMA beamforming (this wave only)
+        beamform_tile_idx(g, dst + si, 1, 0, lane);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        const float bf = g.bf[t];
+        const float e_c = wave_fold_sum(
+            AWACS::clutter_partial(P, r2d, bdir, lane));
+        const float dr = (float)P.range_res;
+        float sum = 0.0f;
+        int used = 0;
+        for (int k = P.cfar_nguard + 1;
+             k <= P.cfar_nguard + P.cfar_nref; ++k) {
+            const float rlo = r2d - (float)k * dr;
+            const float rhi = r2d + (float)k * dr;
+            if (rlo > dr) {
+                sum += wave_fold_sum(
+                    AWACS::clutter_partial(P, rlo, bdir, lane));
+                ++used;
+            }
+            sum += wave_fold_sum(
+                AWACS::clutter_partial(P, rhi, bdir, lane));
+            ++used;
+        }
+        const float mean = used > 0 ? sum / (float)used : 0.0f;
+        const float thr =
+            (float)P.cfar_alpha * (mean + (float)P.noise_floor);
+        const float mp = AWACS::multipath_gain(P, tx, ty, ta, r2d);
+        const float e_t =
+            AWACS::target_energy(P, bf, g.rcs[t], r2d, ta, mp);
+        const float pd =
+            AWACS::detect_pd(e_t, e_c, (float)P.noise_floor, thr);
+        clut_local += (double)e_c;
+        if (lane == 0) {
+            const float r2 = tx * tx + ty * ty + 1.0f;
+            pow_local += (double)(bf * g.rcs[t] / (r2 * r2));
+            if (AWACS::draw_u01(trial, dwl, (uint32_t)t) < pd) {
+                g.det_cnt[t] += 1u;  // unique t per wave: no race
+                det_local += 1ull;
+            }
+        }
+    }
+    // per-wave partials -> LDS; wave 0 lane 0 combines in wave order
+    if (lane == 0) {
+        ull_acc[nw + wv] = shield_local;
+        ull_acc[2 * nw + wv] = det_local;
+        dbl_acc[wv] = clut_local;
+        dbl_acc[nw + wv] = pow_local;
+    }
+    __syncthreads();
+    if (wv == 0 && lane == 0) {
+        unsigned long long ilm = 0, shd = 0, det = 0;
+        double clt = 0.0, pwr = 0.0;
+        for (int w = 0; w < nw; ++w) {
+            ilm += ull_acc[w];
+            shd += ull_acc[nw + w];
+            det += ull_acc[2 * nw + w];
+            clt += dbl_acc[w];
+            pwr += dbl_acc[nw + w];
+        }
+        g.illuminated += ilm;
+        g.shielded += shd;
+        g.detections += det;
+        g.sum_clutter += clt;
+        g.sum_power += pwr;
+        g.last_t = now;
+        g.dwells += 1u;
+    }
+}
+
+__global__ __launch_bounds__(256) void awacs_block_kernel(
+    const AWACS::Params* __restrict__ dP, uint64_t master_seed,
+    uint64_t trial_base, uint32_t ntrials, AWACS::Result* __restrict__ out,
+    StA* __restrict__ stores) {
+    __shared__ int surv_lds[AWACS::MAX_T];
+    __shared__ int dst_lds[AWACS::MAX_T];
+    __shared__ int cnts[8];
+    __shared__ int bcast[4];
+    __shared__ float fdt[1];
+    __shared__ unsigned long long ull_acc[24];
+    __shared__ double dbl_acc[16];
+    const int lane = (int)(threadIdx.x & 63);
+    const int wv = (int)(threadIdx.x >> 6);
+    const int nw = (int)(blockDim.x >> 6);
+#define BWAVE_FENCE() asm volatile("s_waitcnt vmcnt(0)" ::: "memory")
+    EngA E(stores[blockIdx.x]);
+    for (uint32_t trial = blockIdx.x; trial < ntrials; trial += gridDim.x) {
+        if (wv == 0) {
+            engine_init_phase(E, lane, dP, master_seed, trial_base, trial);
+            BWAVE_FENCE();
+        }
+        __syncthreads();
+        for (;;) {
+            if (wv == 0) {
+                const int pw =
+                    __builtin_amdgcn_readfirstlane(engine_phase(E, lane));
+                BWAVE_FENCE();
+                if (lane == 0) {
+                    bcast[0] = pw;
+                    bcast[1] = E.globals.nt;
+                    fdt[0] = (float)(E.now - E.globals.last_t);
+                    const unsigned long long nb =
+                        (unsigned long long)__double_as_longlong(E.now);
+                    bcast[2] = (int)(nb & 0xFFFFFFFFull);
+                    bcast[3] = (int)(nb >> 32);
+                }
+            }
+            __syncthreads();
+            // every wave drops stale L1 lines before touching the
+            // engine-updated globals (guide §6 G16)
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+            const int pw = bcast[0];
+            if (!(pw & 1)) break;
+            const uint32_t dwl = (uint32_t)pw >> 1;
+            const double now_b = __longlong_as_double(
+                ((long long)bcast[3] << 32) | (unsigned int)bcast[2]);
+            dwell_physics_block(*dP, stores[blockIdx.x].globals,
+                                (uint32_t)(trial_base + trial), dwl, lane,
+                                wv, nw, fdt[0], bcast[1], (float)dP->area,
+                                now_b, surv_lds, dst_lds, cnts, ull_acc,
+                                dbl_acc);
+            __syncthreads();
+            if (wv == 0) {
+                BWAVE_FENCE();
+                __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+                engine_resume_phase(E, lane);
+            }
+        }
+        if (wv == 0) engine_finish_phase(E, lane, &out[trial]);
+        __syncthreads();
+    }
+#undef BWAVE_FENCE
+}
+
 // numerics-test kernel: one dwell's beamforming powers for a preloaded
 // engine state (host compares against the scalar fp32/fp64 reference)
 __global__ __launch_bounds__(64) void awacs_power_kernel(
@@ -552,9 +790,19 @@ int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
         HIP_TRY(hipGetLastError());
         P.terrain = d_terr;
     }
+    const char* sp = getenv("CIMBA_AWACS_SCALAR");
+    const char* pr = getenv("CIMBA_AWACS_PROBE");  // perf phase toggles
+    const char* wk = getenv("CIMBA_AWACS_WAVE");   // force wave kernel
+    // pipeline mode: trial-per-BLOCK (4 cooperating waves — per-dwell
+    // latency is the metric at small trial counts); free-space mode and
+    // the A/B override keep the wave-per-trial kernel
+    const bool use_block = P.use_terrain && !(wk && atoi(wk)) &&
+                           !(sp && atoi(sp));
     const uint32_t want_waves = (uint32_t)ntrials;
     const uint32_t nwaves = want_waves < 8192u ? want_waves : 8192u;
-    const uint32_t blocks = (nwaves + 3) / 4;
+    const uint32_t blocks = use_block
+                                ? (uint32_t)(ntrials < 2048 ? ntrials : 2048)
+                                : (nwaves + 3) / 4;
 
     AWACS::Params* d_P = nullptr;
     AWACS::Result* d_out = nullptr;
@@ -562,18 +810,22 @@ int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
     HIP_TRY(hipMalloc(&d_P, sizeof(P)));
     HIP_TRY(hipMemcpy(d_P, &P, sizeof(P), hipMemcpyHostToDevice));
     HIP_TRY(hipMalloc(&d_out, sizeof(AWACS::Result) * ntrials));
-    HIP_TRY(hipMalloc(&d_eng, sizeof(StA) * blocks * 4));
+    HIP_TRY(hipMalloc(&d_eng, sizeof(StA) * (use_block ? blocks
+                                                       : blocks * 4)));
 
     hipEvent_t t0, t1;
     HIP_TRY(hipEventCreate(&t0));
     HIP_TRY(hipEventCreate(&t1));
     HIP_TRY(hipEventRecord(t0));
-    const char* sp = getenv("CIMBA_AWACS_SCALAR");
-    const char* pr = getenv("CIMBA_AWACS_PROBE");  // perf phase toggles
-    hipLaunchKernelGGL(awacs_kernel, dim3(blocks), dim3(256), 0, 0, d_P,
-                       seed, trial_base, (uint32_t)ntrials, d_out, d_eng,
-                       (float*)nullptr,
-                       (sp ? atoi(sp) : 0) | ((pr ? atoi(pr) : 0) << 8));
+    if (use_block)
+        hipLaunchKernelGGL(awacs_block_kernel, dim3(blocks), dim3(256), 0,
+                           0, d_P, seed, trial_base, (uint32_t)ntrials,
+                           d_out, d_eng);
+    else
+        hipLaunchKernelGGL(awacs_kernel, dim3(blocks), dim3(256), 0, 0,
+                           d_P, seed, trial_base, (uint32_t)ntrials, d_out,
+                           d_eng, (float*)nullptr,
+                           (sp ? atoi(sp) : 0) | ((pr ? atoi(pr) : 0) << 8));
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(t1));
     HIP_TRY(hipEventSynchronize(t1));
