@@ -54,6 +54,9 @@ int cimba_sample_gpu(int dist, double p0, uint64_t n, uint64_t seed,
                      int device, double* host_out, double* elapsed_ms);
 int cimba_sample_moments_gpu(int dist, double p0, uint64_t n, uint64_t seed,
                              int device, double* out7, double* elapsed_ms);
+int cimba_sample_moments_gpu2(int dist, double p0, uint64_t n,
+                              uint64_t seed, int device, int use_mfma,
+                              double* out7, double* elapsed_ms);
 int cimba_awacs_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                         uint64_t trial_base, int device, double* elapsed_ms,
                         void* results_out);
@@ -313,7 +316,7 @@ static py::dict rng_sample_gpu(const std::string& dist, double p0, uint64_t n,
 }
 
 static py::dict rng_moments_gpu(const std::string& dist, double p0, uint64_t n,
-                                uint64_t seed, int device) {
+                                uint64_t seed, int device, int mfma) {
     auto it = GPU_DISTS.find(dist);
     if (it == GPU_DISTS.end())
         throw std::invalid_argument("unknown gpu distribution: " + dist);
@@ -322,7 +325,8 @@ static py::dict rng_moments_gpu(const std::string& dist, double p0, uint64_t n,
     int rc;
     {
         py::gil_scoped_release nogil;
-        rc = cimba_sample_moments_gpu(it->second, p0, n, seed, device, o, &ms);
+        rc = cimba_sample_moments_gpu2(it->second, p0, n, seed, device, mfma,
+                                       o, &ms);
     }
     if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
     py::dict d;
@@ -962,7 +966,7 @@ PYBIND11_MODULE(_C, m) {
           py::arg("device") = 0);
     m.def("rng_moments_gpu", &rng_moments_gpu, py::arg("dist"),
           py::arg("p0") = 0.0, py::arg("n") = 1 << 26, py::arg("seed") = 1ULL,
-          py::arg("device") = 0);
+          py::arg("device") = 0, py::arg("mfma") = 0);
     m.def("fmix64", &py_fmix64);
     m.def("terrain_host", &terrain_host, py::arg("cols"), py::arg("rows"),
           py::arg("base") = 0.0, py::arg("amp") = 1000.0,
