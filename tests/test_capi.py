@@ -180,3 +180,41 @@ int main(void) {
     # O(n^2) was multiple seconds of pure scanning; O(n log n) is ~tens
     # of ms — 3 s leaves huge CI headroom while still failing a scan
     assert dt < 3.0, f"event storm took {dt:.1f}s — quadratic path?"
+
+
+def _build_and_run(tmp_path, src_name, exe_name, timeout=600, args=()):
+    exe = str(tmp_path / exe_name)
+    r = subprocess.run(
+        ["gcc", "-std=c11", "-O2", "-Wall", "-I", os.path.join(ROOT, "include"),
+         os.path.join(ROOT, "tutorial", src_name),
+         "-L", os.path.join(ROOT, "cimba_amd"), "-lcimba",
+         f"-Wl,-rpath,{os.path.join(ROOT, 'cimba_amd')}", "-lm", "-o", exe],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr[-3000:]
+    out = subprocess.run([exe, *args], capture_output=True, text=True,
+                         timeout=timeout)
+    return out
+
+
+def test_capi_harbor_tutorial(libcimba, tmp_path):
+    """Harbor-with-abandoned-trials (reference tut_4_3 counterpart):
+    condition waits, multi-unit pool acquires, logger_error -> trial
+    abandon with the cleanup hook, multithreaded scenario grid."""
+    out = _build_and_run(tmp_path, "harbor_capi.c", "harbor")
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+    assert "abandoned trials:" in out.stdout
+    # the cleanup-hook count must equal the failed count (checked in C,
+    # nonzero exit otherwise); determinism: same seed -> same output
+    out2 = _build_and_run(tmp_path, "harbor_capi.c", "harbor")
+    assert out2.stdout == out.stdout
+
+
+def test_capi_awacs_tutorial(libcimba, tmp_path):
+    """CPU AWACS (reference tut_5_1 counterpart): 1000 target processes,
+    per-dwell radar scanning, a condition watcher — the C API at the
+    reference tutorial's process scale."""
+    out = _build_and_run(tmp_path, "awacs_capi.c", "awacsc")
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+    assert "reached the 200-track goal" in out.stdout
+    last = out.stdout.strip().splitlines()[-1]
+    assert "(0 failed)" in last
