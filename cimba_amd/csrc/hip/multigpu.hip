@@ -60,15 +60,16 @@ __global__ __launch_bounds__(256) void mm1_lane_kernel_mg(
 }
 
 // per-device reduction kernel: per-trial avg system times -> raw moment
-// sums (n, Sx, Sx2, Sx3, Sx4, min, max) + event total, on-device so only
-// 8 doubles cross xGMI per GPU
+// sums (n, Sx, Sx2, Sx3, Sx4, events, min, max), ALL on-device so only
+// 8 doubles cross xGMI per GPU (the event total included — no per-trial
+// Result round-trip to the host)
 __global__ __launch_bounds__(256) void mm1_summarize_kernel(
     const MM1::Result* __restrict__ res, uint32_t n,
     double* __restrict__ out8) {
-    __shared__ double acc[5][256];
+    __shared__ double acc[6][256];
     __shared__ double amin[256], amax[256];
     const uint32_t tid = threadIdx.x;
-    double s[5] = {0, 0, 0, 0, 0};
+    double s[6] = {0, 0, 0, 0, 0, 0};
     double mn = 1e308, mx = -1e308;
     for (uint32_t i = blockIdx.x * blockDim.x + tid; i < n;
          i += gridDim.x * blockDim.x) {
@@ -80,24 +81,25 @@ __global__ __launch_bounds__(256) void mm1_summarize_kernel(
         s[2] += x2;
         s[3] += x2 * x;
         s[4] += x2 * x2;
+        s[5] += (double)res[i].events;
         mn = x < mn ? x : mn;
         mx = x > mx ? x : mx;
     }
-    for (int k = 0; k < 5; ++k) acc[k][tid] = s[k];
+    for (int k = 0; k < 6; ++k) acc[k][tid] = s[k];
     amin[tid] = mn;
     amax[tid] = mx;
     __syncthreads();
     for (uint32_t w = 128; w > 0; w >>= 1) {
         if (tid < w) {
-            for (int k = 0; k < 5; ++k) acc[k][tid] += acc[k][tid + w];
+            for (int k = 0; k < 6; ++k) acc[k][tid] += acc[k][tid + w];
             amin[tid] = amin[tid + w] < amin[tid] ? amin[tid + w] : amin[tid];
             amax[tid] = amax[tid + w] > amax[tid] ? amax[tid + w] : amax[tid];
         }
         __syncthreads();
     }
     if (tid == 0) {
-        for (int k = 0; k < 5; ++k) atomicAdd(&out8[k], acc[k][0]);
-        // min/max via CAS loops (few blocks); slot 5 = events (host-set)
+        for (int k = 0; k < 6; ++k) atomicAdd(&out8[k], acc[k][0]);
+        // min/max via CAS loops (few blocks)
         unsigned long long* pmn = (unsigned long long*)&out8[6];
         unsigned long long old = *pmn, assumed;
         do {
@@ -124,7 +126,6 @@ struct DevCtx {
     double* d_sums;   // [8]: n, Sx, Sx2, Sx3, Sx4, min, max, events
     double* d_red;    // reduced [8]
     uint32_t lo, hi;
-    uint64_t events;
     int rc;
 };
 
@@ -203,21 +204,8 @@ int cimba_mm1_multigpu_rccl(uint64_t ntrials, double arr_mean,
         hipLaunchKernelGGL(mm1_summarize_kernel, dim3(64), dim3(256), 0,
                            c.stream, c.d_res, mine, c.d_sums);
         if (!TRY(hipGetLastError())) return;
-        // events total: host-side accumulate (small copy)
-        std::vector<MM1::Result> host_res(mine);
-        if (!TRY(hipMemcpyAsync(host_res.data(), c.d_res,
-                                sizeof(MM1::Result) * mine,
-                                hipMemcpyDeviceToHost, c.stream)))
-            return;
-        if (!TRY(hipStreamSynchronize(c.stream))) return;
-        c.events = 0;
-        for (auto& r : host_res) c.events += r.events;
-        double ev_d = (double)c.events;
-        if (!TRY(hipMemcpyAsync(c.d_sums + 5, &ev_d, sizeof(double),
-                                hipMemcpyHostToDevice, c.stream)))
-            return;
         // the north-star reduce: statistics over xGMI
-        // (sums: moments + event total; then min / max)
+        // (sums: moments + event total, all summed on-device; min / max)
         if (!NTRY(ncclAllReduce(c.d_sums, c.d_red, 6, ncclDouble, ncclSum,
                                 c.comm, c.stream)))
             return;
