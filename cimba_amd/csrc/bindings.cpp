@@ -41,6 +41,11 @@ int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
                       uint64_t num_objects, uint64_t seed,
                       uint64_t trial_base, int device, double until,
                       uint64_t max_events, Mm1GpuOut* out);
+int cimba_mm1_gpu_run_pt(uint64_t ntrials, double arr_mean, double srv_mean,
+                         uint64_t num_objects, uint64_t seed,
+                         uint64_t trial_base, int device, double until,
+                         uint64_t max_events, Mm1GpuOut* out,
+                         double* per_trial_avg_out);
 int cimba_gpu_device_count(int* n);
 int cimba_gpu_sync(void);
 int cimba_scenario_gpu_run(int which, void* result_out);

@@ -467,10 +467,28 @@ struct Mm1GpuOut {
     int32_t pad_;
 };
 
+// per_trial_avg_out (nullable): per-trial average system times, for the
+// host-vs-device divergence-count test (tests/test_gpu.py)
+int cimba_mm1_gpu_run_pt(uint64_t ntrials, double arr_mean, double srv_mean,
+                         uint64_t num_objects, uint64_t seed,
+                         uint64_t trial_base, int device, double until,
+                         uint64_t max_events, Mm1GpuOut* out,
+                         double* per_trial_avg_out);
+
 int cimba_mm1_gpu_run(uint64_t ntrials, double arr_mean, double srv_mean,
                       uint64_t num_objects, uint64_t seed,
                       uint64_t trial_base, int device, double until,
                       uint64_t max_events, Mm1GpuOut* out) {
+    return cimba_mm1_gpu_run_pt(ntrials, arr_mean, srv_mean, num_objects,
+                                seed, trial_base, device, until, max_events,
+                                out, nullptr);
+}
+
+int cimba_mm1_gpu_run_pt(uint64_t ntrials, double arr_mean, double srv_mean,
+                         uint64_t num_objects, uint64_t seed,
+                         uint64_t trial_base, int device, double until,
+                         uint64_t max_events, Mm1GpuOut* out,
+                         double* per_trial_avg_out) {
     HIP_TRY(hipSetDevice(device));
     MM1::Params P{arr_mean, srv_mean, num_objects};
     std::vector<MM1::Result> res(ntrials);
@@ -528,6 +546,10 @@ aggregate:
         out->total_events += res[i].events;
         out->total_objs += res[i].obj_cnt;
         out->total_wait += res[i].sum_wait;
+        if (per_trial_avg_out)
+            per_trial_avg_out[i] =
+                res[i].obj_cnt ? res[i].sum_wait / (double)res[i].obj_cnt
+                               : 0.0;
         if (res[i].status == 0)
             ++out->trials_ok;
         else if (out->first_bad_status == 0)
