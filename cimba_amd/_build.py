@@ -17,6 +17,9 @@ SOURCES = [
     os.path.join(CSRC, "host", "support.cpp"),
     os.path.join(CSRC, "bindings.cpp"),
     os.path.join(CSRC, "hip", "deskernel.hip"),
+    os.path.join(CSRC, "hip", "deskernel_mg1.hip"),
+    os.path.join(CSRC, "hip", "deskernel_jobshop.hip"),
+    os.path.join(CSRC, "hip", "deskernel_scenario.hip"),
     os.path.join(CSRC, "hip", "rng_kernel.hip"),
     os.path.join(CSRC, "hip", "awacs_kernel.hip"),
     os.path.join(CSRC, "hip", "multigpu.hip"),
@@ -64,15 +67,18 @@ def needs_build():
     return os.path.getmtime(so) < _sources_mtime()
 
 
-def build(verbose=True, force=False):
-    if not force and not needs_build():
+def build(verbose=True, force=False, out_dir=None):
+    """Build the native extension.  out_dir redirects every artifact
+    (objects, _C*.so, libcimba.so) for from-scratch rebuild checks
+    without touching the in-tree binaries."""
+    if out_dir is None and not force and not needs_build():
         return ext_path()
     import pybind11
 
     hipcc = _hipcc()
     py_inc = sysconfig.get_paths()["include"]
     pb_inc = pybind11.get_include()
-    objdir = os.path.join(ROOT, "_objs")
+    objdir = os.path.join(out_dir or ROOT, "_objs")
     os.makedirs(objdir, exist_ok=True)
 
     cflags = [
@@ -100,7 +106,8 @@ def build(verbose=True, force=False):
     with concurrent.futures.ThreadPoolExecutor(max_workers=len(SOURCES)) as ex:
         objs = list(ex.map(compile_one, SOURCES))
 
-    so = ext_path()
+    so = (os.path.join(out_dir, os.path.basename(ext_path()))
+          if out_dir else ext_path())
     cmd = [hipcc, "-shared", "-fPIC", "-o", so] + objs + ["-L/opt/rocm/lib", "-lrccl"]
     if verbose:
         print("[cimba_amd build]", " ".join(cmd), flush=True)
@@ -114,7 +121,7 @@ def build(verbose=True, force=False):
     if verbose:
         print("[cimba_amd build]", " ".join(cmd), flush=True)
     subprocess.run(cmd, check=True)
-    libcimba = os.path.join(ROOT, "libcimba.so")
+    libcimba = os.path.join(out_dir or ROOT, "libcimba.so")
     cmd = [hipcc, "-shared", "-fPIC", "-o", libcimba, capi_obj,
            os.path.join(objdir, "support.cpp.o")]
     if verbose:
