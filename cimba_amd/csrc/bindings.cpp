@@ -585,11 +585,13 @@ static py::dict mm1_gpu(uint64_t ntrials, uint64_t num_objects, double arr_rate,
                         uint64_t trial_base) {
     Mm1GpuOut o;
     int rc;
+    std::vector<double> pt(ntrials);
     {
         py::gil_scoped_release nogil;
-        rc = cimba_mm1_gpu_run(ntrials, 1.0 / arr_rate, 1.0 / srv_rate,
-                               num_objects, seed, trial_base, device, 1.0e308,
-                               UINT64_C(0xFFFFFFFFFFFFFFFF), &o);
+        rc = cimba_mm1_gpu_run_pt(ntrials, 1.0 / arr_rate, 1.0 / srv_rate,
+                                  num_objects, seed, trial_base, device,
+                                  1.0e308, UINT64_C(0xFFFFFFFFFFFFFFFF), &o,
+                                  pt.data());
     }
     if (rc != 0) throw std::runtime_error("hip error " + std::to_string(rc));
     py::dict d;
@@ -602,6 +604,9 @@ static py::dict mm1_gpu(uint64_t ntrials, uint64_t num_objects, double arr_rate,
     d["avg_system_time"] = o.total_objs ? o.total_wait / (double)o.total_objs : 0.0;
     d["events_per_sec"] =
         o.elapsed_ms > 0 ? (double)o.total_events / (o.elapsed_ms * 1e-3) : 0.0;
+    py::list pl;
+    for (double v : pt) pl.append(v);
+    d["per_trial_avg"] = pl;
     return d;
 }
 
