@@ -1260,6 +1260,7 @@ struct Engine {
         Rng rng;
         int32_t evn;
         int32_t pad_;
+        EvEntry evtop;  // heap-top register cache (valid when evn > 0)
     };
 
     CMB_FORCEINLINE void ctx_save(Ctx& c) const {
@@ -1272,6 +1273,7 @@ struct Engine {
         c.trial_index = trial_index;
         c.rng = rng;
         c.evn = evq.n;
+        c.evtop = evq.top_c;
     }
 
     CMB_FORCEINLINE void ctx_load(const Ctx& c, const Params* p) {
@@ -1285,6 +1287,7 @@ struct Engine {
         params = p;
         rng = c.rng;
         evq.n = c.evn;
+        evq.top_c = c.evtop;
     }
 
     // Identify the code path the NEXT dispatch_one() will take:

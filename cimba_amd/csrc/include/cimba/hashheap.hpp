@@ -85,6 +85,11 @@ struct HashHeap {
     EvMapSlot* map;  // back-map slots (storage-owned when MAP)
     int32_t cap2;    // usable spill entries (0 until attached)
     int32_t n;
+    // register-cached copy of e[0] (valid while n > 0): the heap top is
+    // read by every peek and every pop — the hottest dependent load in
+    // the dispatch chain — and every mutation flows through place(), so
+    // the cache is maintained by construction
+    EvEntry top_c;
 
     CMB_FORCEINLINE explicit HashHeap(EvEntry (&buf)[CAP])
         : e(buf), e2(nullptr), map(nullptr), cap2(0), n(0) {}
@@ -147,9 +152,11 @@ struct HashHeap {
         }
     }
 
-    // placement: every heap move goes through here so the map tracks it
+    // placement: every heap move goes through here so the map AND the
+    // top cache track it
     CMB_FORCEINLINE void place(int32_t i, const EvEntry& ev) {
         at(i) = ev;
+        if (i == 0) top_c = ev;
         map_set(ev.handle, i);
     }
 
@@ -184,7 +191,7 @@ struct HashHeap {
     }
     CMB_FORCEINLINE bool empty() const { return n == 0; }
     CMB_FORCEINLINE bool full() const { return n == capacity(); }
-    CMB_FORCEINLINE const EvEntry& top() const { return e[0]; }
+    CMB_FORCEINLINE const EvEntry& top() const { return top_c; }
 
     CMB_FORCEINLINE void sift_up(int32_t i) {
         EvEntry tmp = at(i);
@@ -220,7 +227,7 @@ struct HashHeap {
     }
 
     CMB_FORCEINLINE EvEntry pop() {
-        EvEntry out = e[0];
+        EvEntry out = top_c;
         map_erase(out.handle);
         --n;
         if (n > 0) {
