@@ -61,10 +61,10 @@ int cimba_mm1_gpu_run_pt(uint64_t ntrials, double arr_mean, double srv_mean,
     if (use_lane) {
         const char* lb = getenv("CIMBA_MM1_LANE_BLOCKS");
         const uint32_t blocks = lb ? (uint32_t)atoi(lb) : 2048u;
-        // measured (gpurun r2 sweeps): plain scratch wins at the bench
-        // batch (4.0 G ev/s at N>=524288); vote-gated conv only matched
-        // the occupancy-tuned scratch on M/M/1, so scratch stays default
-        const int lane_mode = lane ? atoi(lane) : 2;
+        // measured (profiles/r02_conv_divergence.md): after the heap-top
+        // register cache the vote-gated conv kernel wins for M/M/1 too
+        // (4.69 vs 4.55 G ev/s at the bench batch)
+        const int lane_mode = lane ? atoi(lane) : 3;
         if (lane_mode == 1)  // explicit HBM-lane variant
             rc = run_trials_gpu_lane<MM1, 1>(P, ntrials, seed, trial_base, until,
                                              max_events, &out->elapsed_ms,
