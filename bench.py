@@ -179,7 +179,10 @@ def main():
                 "arrival_rate": 0.9,
                 "service_rate": 1.0,
                 "parallelism": f"trial-parallel dp{n_gpus}",
-                "engine": "trial-per-lane (64/wave) for large batches, trial-per-wavefront + LDS otherwise",
+                "engine": "vote-gated converged lane kernel (64 trials/wave, "
+                          "wave-voted dispatch paths, heap-top register "
+                          "cache); trial-per-wavefront + LDS for small "
+                          "batches",
                 "device": "gpu" if use_gpu else "cpu-host-debug",
             },
         }

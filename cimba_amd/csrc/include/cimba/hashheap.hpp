@@ -21,12 +21,14 @@ namespace cmb {
 // similar), c = auxiliary u32 (guard id / timer slot), b = 64-bit payload
 // (signal value, object word)).
 struct EvEntry {
+    // field order groups the PEEK-hot head {t, kind, a, handle} into the
+    // first 16-byte chunk (one dwordx4 load covers a path peek)
     double t;        // activation time
-    uint64_t pseq;   // (INT16_MAX - priority) << 48 | seq  → min == next
-    uint64_t b;      // payload
-    uint32_t handle; // unique id for cancel/reschedule
     uint16_t kind;
     uint16_t a;
+    uint32_t handle; // unique id for cancel/reschedule
+    uint64_t pseq;   // (INT16_MAX - priority) << 48 | seq  → min == next
+    uint64_t b;      // payload
     uint32_t c;
     uint32_t pad_;
 };
