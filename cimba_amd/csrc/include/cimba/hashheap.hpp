@@ -21,14 +21,17 @@ namespace cmb {
 // similar), c = auxiliary u32 (guard id / timer slot), b = 64-bit payload
 // (signal value, object word)).
 struct EvEntry {
-    // field order groups the PEEK-hot head {t, kind, a, handle} into the
-    // first 16-byte chunk (one dwordx4 load covers a path peek)
+    // NOTE: a "peek-hot head first" reorder ({t, kind, a, handle} in the
+    // leading 16 B) was measured 2-4% SLOWER across all three lane
+    // models: with the heap-top register cache a peek costs no memory at
+    // all, and splitting the (t, pseq) comparator pair across chunks
+    // taxes every sift compare instead.  Comparator-pair-first stays.
     double t;        // activation time
-    uint16_t kind;
-    uint16_t a;
-    uint32_t handle; // unique id for cancel/reschedule
     uint64_t pseq;   // (INT16_MAX - priority) << 48 | seq  → min == next
     uint64_t b;      // payload
+    uint32_t handle; // unique id for cancel/reschedule
+    uint16_t kind;
+    uint16_t a;
     uint32_t c;
     uint32_t pad_;
 };
