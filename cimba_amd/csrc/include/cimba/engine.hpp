@@ -1244,51 +1244,8 @@ struct Engine {
         return true;
     }
 
-    // ---- context spill/restore + path peek: support for the
-    // path-converged lane kernel (hip/deskernel.hip conv_lane_kernel).
-    // A lane time-multiplexes K trials; the hot scalar context of a
-    // parked trial lives in this POD alongside its Storage. ----
-
-    struct Ctx {
-        double now;
-        uint64_t ev_dispatched;
-        uint64_t seq;
-        uint32_t next_handle;
-        int32_t status;
-        int32_t n_event_waiters;
-        uint32_t trial_index;
-        Rng rng;
-        int32_t evn;
-        int32_t pad_;
-        EvEntry evtop;  // heap-top register cache (valid when evn > 0)
-    };
-
-    CMB_FORCEINLINE void ctx_save(Ctx& c) const {
-        c.now = now;
-        c.ev_dispatched = ev_dispatched;
-        c.seq = seq;
-        c.next_handle = next_handle;
-        c.status = status;
-        c.n_event_waiters = n_event_waiters;
-        c.trial_index = trial_index;
-        c.rng = rng;
-        c.evn = evq.n;
-        c.evtop = evq.top_c;
-    }
-
-    CMB_FORCEINLINE void ctx_load(const Ctx& c, const Params* p) {
-        now = c.now;
-        ev_dispatched = c.ev_dispatched;
-        seq = c.seq;
-        next_handle = c.next_handle;
-        status = c.status;
-        n_event_waiters = c.n_event_waiters;
-        trial_index = c.trial_index;
-        params = p;
-        rng = c.rng;
-        evq.n = c.evn;
-        evq.top_c = c.evtop;
-    }
+    // ---- path peek: support for the vote-gated converged lane kernel
+    // (hip/deskernel_impl.hpp conv_lane_kernel) ----
 
     // Identify the code path the NEXT dispatch_one() will take:
     // (event kind, owning process function, its resume pc).  Lanes whose
