@@ -157,3 +157,25 @@ def test_bench_self_launch_host():
     assert rep["n_gpus"] == 2
     assert rep["value"] > 0
     assert rep["config"]["parallelism"] == "trial-parallel dp2"
+
+
+@pytest.mark.parametrize("model", ["mg1", "jobshop"])
+def test_bench_model_flags_self_launch_host(model):
+    """BASELINE configs 3-4 through the same self-launch contract."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.pop("WORLD_SIZE", None)
+    env.pop("RANK", None)
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "bench.py"), "--gpus", "2",
+         "--host", "--model", model, "--steps", "1", "--warmup", "0",
+         "--trials", "4", "--objects", "200"],
+        capture_output=True, text=True, timeout=600, env=env, cwd=root)
+    assert out.returncode == 0, out.stderr[-2000:]
+    rep = json.loads([l for l in out.stdout.splitlines()
+                      if l.startswith("{")][-1])
+    assert rep["n_gpus"] == 2 and rep["value"] > 0
