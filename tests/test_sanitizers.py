@@ -131,3 +131,31 @@ def test_tsan_prodcons_stress(tmp_path):
                          timeout=900, env=env)
     assert out.returncode == 0, (out.stdout[-1000:], out.stderr[-3000:])
     assert "prodcons fuzz OK" in out.stdout
+
+
+def test_asan_ubsan_spill_and_backmap(tmp_path):
+    """The round-2 memory machinery under ASan+UBSan: heap/queue spill
+    crossings (two-tier indexing, slab claim) and the handle back-map
+    (open addressing, backward-shift deletion) — exactly the code where
+    an off-by-one would corrupt silently in a release build."""
+    import re
+
+    spf = open(os.path.join(ROOT, "tests", "test_spill.py")).read()
+    HARNESS = re.search(r'HARNESS = r"""(.*?)"""', spf, re.S).group(1)
+    src = tmp_path / "spill_san.cpp"
+    src.write_text(HARNESS)
+    exe = str(tmp_path / "spill_san")
+    r = subprocess.run(
+        [CLANGXX, "-std=c++17", "-O1", "-g",
+         "-fsanitize=address,undefined", "-fno-omit-frame-pointer",
+         "-I", os.path.join(ROOT, "cimba_amd", "csrc", "include"),
+         str(src),
+         os.path.join(ROOT, "cimba_amd", "csrc", "host", "support.cpp"),
+         "-o", exe, "-lpthread"], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr[-3000:]
+    out = subprocess.run([exe], capture_output=True, text=True, timeout=900,
+                         env={**os.environ,
+                              "UBSAN_OPTIONS": "halt_on_error=1",
+                              "ASAN_OPTIONS": "detect_leaks=1"})
+    assert out.returncode == 0, (out.stdout[-1000:], out.stderr[-3000:])
+    assert "spill suite OK" in out.stdout

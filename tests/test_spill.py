@@ -121,6 +121,26 @@ struct CfgSpill {
     static constexpr int PQCAP = 1;
     static constexpr int NUM_COND = 0;
 };
+// map-enabled variant: the handle->index back-map plus the spill tier
+// together (the host C API shape), for the open-addressing/backward-
+// shift-deletion storm below
+struct CfgMap {
+    static constexpr int MAX_PROC = 2;
+    static constexpr int MAX_EV = 32;
+    static constexpr int SPILL_EV = 480;
+    static constexpr bool EV_MAP = true;
+    static constexpr int TIMERS = 1;
+    static constexpr int NUM_QUEUES = 1;
+    static constexpr int QCAP = 8;
+    static constexpr int SPILL_Q = 120;
+    static constexpr int NUM_RES = 0;
+    static constexpr int NUM_POOLS = 0;
+    static constexpr int NUM_BUFS = 0;
+    static constexpr int NUM_PQ = 0;
+    static constexpr int PQCAP = 1;
+    static constexpr int NUM_COND = 0;
+};
+
 struct CfgBig {
     static constexpr int MAX_PROC = 2;
     static constexpr int MAX_EV = 72;
@@ -247,11 +267,70 @@ static int test_exhaustion_abort() {
     return 0;
 }
 
+// 5) back-map storm: schedule / random-cancel / reschedule across the
+// tier boundary with the map ON; a scan oracle checks every lookup
+static int test_backmap_storm() {
+    using MMap = Burst<CfgMap>;
+    using EMap = cmb::Engine<MMap>;
+    auto store = std::make_unique<EMap::Storage>();
+    auto slab = std::make_unique<EMap::Spill>();
+    EMap E(*store);
+    E.set_spill(slab.get());
+    MMap::Params P{0};
+    E.init(&P, 99, 0);
+    cmb::Rng r;
+    r.seed(4242);
+    uint32_t h[512] = {0};
+    int live = 0;
+    for (int round = 0; round < 4000; ++round) {
+        const uint64_t op = r.next() % 100;
+        if (op < 55 && live < 512) {  // schedule
+            for (int i = 0; i < 512; ++i)
+                if (!h[i]) {
+                    h[i] = E.schedule(cmb::EV_USER, 0, 0, (uint64_t)i,
+                                      r.uniform(0.0, 1e6), 0);
+                    CHECK(h[i] != 0);
+                    ++live;
+                    break;
+                }
+        } else if (op < 80 && live > 0) {  // cancel a random live handle
+            int k = (int)(r.next() % 512);
+            while (!h[k]) k = (k + 1) % 512;
+            CHECK(E.event_cancel(h[k]));
+            CHECK(!E.event_cancel(h[k]));  // double-cancel must miss
+            h[k] = 0;
+            --live;
+        } else if (live > 0) {  // reschedule
+            int k = (int)(r.next() % 512);
+            while (!h[k]) k = (k + 1) % 512;
+            CHECK(E.event_reschedule(h[k], r.uniform(0.0, 1e6), 0));
+        }
+        CHECK(E.evq.n == live);
+        // oracle: the map agrees with a full scan for one random handle
+        if (live > 0) {
+            int k = (int)(r.next() % 512);
+            while (!h[k]) k = (k + 1) % 512;
+            const int32_t idx = E.evq.find_index(h[k]);
+            CHECK(idx >= 0 && E.evq.at(idx).handle == h[k]);
+        }
+    }
+    // drain in nondecreasing time order
+    double last = -1.0;
+    while (!E.evq.empty()) {
+        cmb::EvEntry ev = E.evq.pop();
+        CHECK(ev.t >= last);
+        last = ev.t;
+    }
+    std::puts("backmap storm OK");
+    return 0;
+}
+
 int main() {
     if (test_queue_fifo()) return 1;
     if (test_heap_spill()) return 1;
     if (test_equivalence()) return 1;
     if (test_exhaustion_abort()) return 1;
+    if (test_backmap_storm()) return 1;
     std::puts("spill suite OK");
     return 0;
 }
