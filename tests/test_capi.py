@@ -196,6 +196,13 @@ def _build_and_run(tmp_path, src_name, exe_name, timeout=600, args=()):
     return out
 
 
+def test_capi_hello_tutorial(libcimba, tmp_path):
+    out = _build_and_run(tmp_path, "hello_capi.c", "hello")
+    assert out.returncode == 0, (out.stdout, out.stderr)
+    assert "hello from hello" in out.stdout
+    assert "simulation ended" in out.stdout
+
+
 def test_capi_harbor_tutorial(libcimba, tmp_path):
     """Harbor-with-abandoned-trials (reference tut_4_3 counterpart):
     condition waits, multi-unit pool acquires, logger_error -> trial
