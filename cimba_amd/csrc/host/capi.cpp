@@ -76,6 +76,7 @@ struct CModel : ModelBase {
         static constexpr int NUM_BUFS = CIMBA_C_NUM_BUFS;
         static constexpr int NUM_PQ = CIMBA_C_NUM_PQ;
         static constexpr int PQCAP = 4096;
+        static constexpr int SPILL_PQ = 12288;   // PQs grow to 16K total
         static constexpr int NUM_COND = CIMBA_C_NUM_COND;
     };
     struct Params {

