@@ -124,6 +124,11 @@ struct ProcRec {
     uint64_t wseq;      // FIFO tie-break within a guard; 64-bit so an
                         // EV_GRANT can never alias a new wait on the same
                         // guard, whatever the trial length
+    uint64_t grank;     // packed wait rank (32767-priority)<<48 | wseq:
+                        // wseq is monotone in dispatch order, so entry_t
+                        // never decides (it ties exactly when wseq does),
+                        // and the guard-front scan loads ONE word per
+                        // waiter instead of three fields
     double entry_t;     // guard entry time (ordering: pri desc, entry asc, seq asc)
     uint32_t timers[NT];  // pending timer event handles; 0 = free slot
     int16_t waiters_head;  // procs waiting for me to finish (wait_process)
@@ -584,7 +589,11 @@ struct Engine {
     }
 
     CMB_FORCEINLINE void proc_priority_set(int pidx, int priority) {
-        procs[pidx].priority = (int16_t)priority;
+        ProcT& p = procs[pidx];
+        p.priority = (int16_t)priority;
+        if (p.await_kind == AW_GUARD)  // live reprioritization of a wait
+            p.grank = ((uint64_t)(uint16_t)(32767 - p.priority) << 48) |
+                      (p.wseq & UINT64_C(0xFFFFFFFFFFFF));
     }
 
     // allocate a process slot, recycling finished ones — the device-native
@@ -789,6 +798,8 @@ struct Engine {
         p.demand_ctx = ctx;
         p.entry_t = now;
         p.wseq = seq++;
+        p.grank = ((uint64_t)(uint16_t)(32767 - p.priority) << 48) |
+                  (p.wseq & UINT64_C(0xFFFFFFFFFFFF));
         if constexpr (GUARD_MASK) {
             guards[gid].wmask |= (uint64_t)1 << pidx_of(&p);
         } else {
@@ -822,17 +833,15 @@ struct Engine {
         int best = -1;
         if constexpr (GUARD_MASK) {
             uint64_t m = guards[gid].wmask;
+            uint64_t bestr = ~UINT64_C(0);
             while (m) {
                 const int i = __builtin_ctzll(m);
                 m &= m - 1;
-                if (best < 0) { best = i; continue; }
-                const ProcT& a = procs[i];
-                const ProcT& b = procs[best];
-                if (a.priority > b.priority ||
-                    (a.priority == b.priority &&
-                     (a.entry_t < b.entry_t ||
-                      (a.entry_t == b.entry_t && a.wseq < b.wseq))))
+                const uint64_t r = procs[i].grank;
+                if (r < bestr) {
+                    bestr = r;
                     best = i;
+                }
             }
             return best;
         }
