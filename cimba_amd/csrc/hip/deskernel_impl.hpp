@@ -192,7 +192,7 @@ __global__ __launch_bounds__(256, MINW) void lane_scratch_kernel(
 // distinct candidates (__shfl from the first lane of each unseen group),
 // and ONLY the lanes on the most-popular path dispatch.  Idle minority
 // lanes cost less than serializing all paths: measured 3.93 G ev/s vs
-// 3.33 G for the ungated scratch kernel (gpurun_out/r2_ab1.log).
+// 3.33 G for the ungated scratch kernel (profiles/logs/r2_ab1.log).
 //
 // Trial refill is wave-synchronous via the stride loop — all lanes of a
 // wave start their next trial together, so the low-utilization tail is
@@ -200,7 +200,7 @@ __global__ __launch_bounds__(256, MINW) void lane_scratch_kernel(
 // K parked trials per lane with per-lane atomic refill — measured 21%
 // WORSE at any batch that needs refill: staggered per-lane trial ages
 // leave each wave grinding out stragglers at ~50% liveness after the
-// pool empties.  gpurun_out/r2_sweep3 data; kept here as the measured
+// pool empties.  profiles/logs (r2 sweep data); kept here as the measured
 // justification for the no-refill shape.)
 // ---------------------------------------------------------------------------
 template <class Model, int MINW = 1>
