@@ -17,6 +17,7 @@
 #include "../models/jobshop.hpp"
 #include "../models/awacs.hpp"
 #include "../models/scenarios.hpp"
+#include "../models/spillprobe.hpp"
 
 #include <map>
 #include <string>
@@ -49,6 +50,9 @@ int cimba_mm1_gpu_run_pt(uint64_t ntrials, double arr_mean, double srv_mean,
 int cimba_gpu_device_count(int* n);
 int cimba_gpu_sync(void);
 int cimba_scenario_gpu_run(int which, void* result_out);
+int cimba_spillprobe_gpu_run(uint64_t ntrials, uint64_t num_objects,
+                             uint64_t seed, uint64_t trial_base, int device,
+                             double* elapsed_ms, void* results_out);
 int cimba_mg1_gpu_run(uint64_t ntrials, const void* params, uint64_t seed,
                       uint64_t trial_base, int device, double* elapsed_ms,
                       void* results_out);
@@ -901,6 +905,46 @@ PYBIND11_MODULE(_C, m) {
         return d;
     }, py::arg("ntargets") = 64, py::arg("master_seed") = 11ULL,
        py::arg("device") = 0);
+    m.def("spillprobe_run", [](uint64_t ntrials, uint64_t num_objects,
+                               uint64_t seed, int device, bool gpu) {
+        using SP = cmb_models::SpillProbe;
+        std::vector<SP::Result> res(ntrials);
+        int rc = 0;
+        double ms = -1.0;
+        {
+            py::gil_scoped_release nogil;
+            if (gpu) {
+                rc = cimba_spillprobe_gpu_run(ntrials, num_objects, seed, 0,
+                                              device, &ms, res.data());
+            } else {
+                SP::Params p{num_objects};
+                run_host<SP>(p, seed, ntrials, 0, res.data());
+            }
+        }
+        if (rc != 0)
+            throw std::runtime_error("hip error " + std::to_string(rc));
+        uint64_t ev = 0, ok = 0;
+        double wait = 0.0;
+        int32_t bad = 0;
+        py::list per_trial;
+        for (auto& r : res) {
+            ev += r.events;
+            wait += r.sum_wait;
+            if (r.status == 0)
+                ++ok;
+            else if (!bad)
+                bad = r.status;
+            per_trial.append(py::make_tuple(r.events, r.sum_wait, r.status));
+        }
+        py::dict d;
+        d["total_events"] = ev;
+        d["total_wait"] = wait;
+        d["trials_ok"] = ok;
+        d["first_bad_status"] = bad;
+        d["per_trial"] = per_trial;
+        return d;
+    }, py::arg("ntrials"), py::arg("num_objects"), py::arg("seed") = 1,
+       py::arg("device") = 0, py::arg("gpu") = true);
     m.def("scenario_host", &scenario_host, py::arg("which"));
     m.def("scenario_gpu", &scenario_gpu, py::arg("which"));
     m.def("scenario_run_host", &scenario_run_host, py::arg("which"),

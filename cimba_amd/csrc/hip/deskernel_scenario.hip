@@ -4,9 +4,11 @@
 #include "deskernel_impl.hpp"
 
 #include "../models/scenarios.hpp"
+#include "../models/spillprobe.hpp"
 
 using cmb::Engine;
 using cmb_models::Scenario;
+using cmb_models::SpillProbe;
 using namespace cmb_dk;
 
 namespace {
@@ -38,6 +40,20 @@ int cimba_scenario_gpu_run(int which, void* result_out) {
                       hipMemcpyDeviceToHost));
     HIP_TRY(hipFree(d_out));
     return 0;
+}
+
+// spill-probe entry (models/spillprobe.hpp): every trial crosses the
+// fast->slab boundary, exercising lazy claim atomics + the two-tier
+// heap/queue paths on the device; runs through the default conv kernel
+int cimba_spillprobe_gpu_run(uint64_t ntrials, uint64_t num_objects,
+                             uint64_t seed, uint64_t trial_base, int device,
+                             double* elapsed_ms, void* results_out) {
+    HIP_TRY(hipSetDevice(device));
+    SpillProbe::Params P{num_objects};
+    return run_trials_gpu_conv<SpillProbe, 4>(
+        P, ntrials, seed, trial_base, 1.0e308,
+        UINT64_C(0xFFFFFFFFFFFFFFFF), elapsed_ms,
+        (SpillProbe::Result*)results_out, 0u);
 }
 
 }  // extern "C"

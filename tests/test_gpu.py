@@ -132,3 +132,30 @@ def test_native_rccl_multigpu_runner():
     assert r["total_events"] == s["total_events"]
     assert abs(r["mean_system_time"] - s["avg_system_time"]) < 1e-9
     assert 8.0 < r["mean_system_time"] < 12.0
+
+
+def test_device_spill_probe():
+    """DEVICE spill machinery exercised for real (the production models
+    overflow their fast tier only once per ~5e5 trials): every spill-
+    probe trial crosses the scratch->HBM slab boundary hundreds of
+    times.  GPU results must match the host engine bitwise, and slab
+    POOL exhaustion must abort the unlucky trials cleanly."""
+    import os
+
+    h = ca._C.spillprobe_run(ntrials=512, num_objects=150, seed=31,
+                             gpu=False)
+    g = ca._C.spillprobe_run(ntrials=512, num_objects=150, seed=31,
+                             gpu=True, device=0)
+    assert h["trials_ok"] == g["trials_ok"] == 512
+    assert g["per_trial"] == h["per_trial"]  # bitwise, incl. sum_wait
+
+    # pool exhaustion: 16 slabs for 512 always-spilling trials -> exactly
+    # the 16 claimants finish; the rest abort with a clean status
+    os.environ["CIMBA_SPILL_SLABS"] = "16"
+    try:
+        e = ca._C.spillprobe_run(ntrials=512, num_objects=150, seed=31,
+                                 gpu=True, device=0)
+    finally:
+        del os.environ["CIMBA_SPILL_SLABS"]
+    assert e["trials_ok"] == 16, e["trials_ok"]
+    assert e["first_bad_status"] in (1, 2)  # ST_HEAP_FULL / ST_QUEUE_FULL

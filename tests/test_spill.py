@@ -407,3 +407,13 @@ def test_mg1_spill_config_available():
     r = ca.mg1_host(ntrials=32, num_objects=4000, arr_rate=0.8,
                     srv_mean=1.0, srv_scv=4.0, dist=2, seed=11, threads=0)
     assert r["trials_ok"] == 32, r
+
+
+def test_spillprobe_host_path():
+    """The spill-probe model on the host engine: every trial crosses the
+    8-entry fast tier into the slab hundreds of times."""
+    import cimba_amd as ca
+
+    r = ca._C.spillprobe_run(ntrials=32, num_objects=150, seed=11, gpu=False)
+    assert r["trials_ok"] == 32, r
+    assert r["total_events"] > 32 * 250
