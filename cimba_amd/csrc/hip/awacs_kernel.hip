@@ -503,6 +503,11 @@ __device__ void dwell_physics_block(
         cnts[wv] = mycnt;
         ull_acc[wv] = illum_local;
     }
+    // drain this wave's x/y/alt stores BEFORE the barrier: s_barrier does
+    // not order vector-memory stores, and the survivor pipeline reads
+    // targets from OTHER waves' triage chunks (same CU -> same L1, so a
+    // landed store is visible; vmcnt(0) guarantees it landed)
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     // concatenate the per-chunk lists (ascending t preserved)
     int off = 0, nsurv = 0;
