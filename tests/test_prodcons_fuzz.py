@@ -30,7 +30,16 @@ struct ProdCons : cmb::ModelBase {
         static constexpr int MAX_EV = 64;
         static constexpr int TIMERS = 2;
         static constexpr int NUM_QUEUES = 2;
+#ifdef CMB_FUZZ_SPILL
+        // spill variant: a 4-slot ring under the same logical limit (8),
+        // so every full-queue episode crosses the slab boundary while
+        // the two-sided contention the fuzz targets is unchanged —
+        // results must be BITWISE identical to the plain build
+        static constexpr int QCAP = 4;
+        static constexpr int SPILL_Q = 16;
+#else
         static constexpr int QCAP = 8;
+#endif
         static constexpr int NUM_RES = 0;
         static constexpr int NUM_POOLS = 0;
         static constexpr int NUM_BUFS = 0;
@@ -123,8 +132,8 @@ struct ProdCons : cmb::ModelBase {
     CMB_FORCEINLINE static void setup(E_& E) {
         E.globals = Globals{};
         E.globals.order_ok = 1;
-        E.queues[0].limit = Cfg::QCAP;
-        E.queues[1].limit = Cfg::QCAP;
+        E.queues[0].limit = 8;  // logical capacity (ring + spill tiers)
+        E.queues[1].limit = 8;
         for (int i = 0; i < NPROD + NCONS; ++i) {
             const int pidx = E.proc_alloc();
             // random priorities shuffle the guard grant order per trial
@@ -175,20 +184,29 @@ int main(int argc, char** argv) {
 """
 
 
-@pytest.fixture(scope="module")
-def harness(tmp_path_factory):
-    d = tmp_path_factory.mktemp("pcfuzz")
-    src = d / "pcfuzz.cpp"
+def _compile(d, extra):
+    src = d / f"pcfuzz{len(extra)}.cpp"
     src.write_text(HARNESS)
-    exe = str(d / "pcfuzz")
+    exe = str(d / f"pcfuzz{len(extra)}")
     r = subprocess.run(
-        ["g++", "-std=c++17", "-O2", "-g",
+        ["g++", "-std=c++17", "-O2", "-g", *extra,
          "-I", os.path.join(ROOT, "cimba_amd", "csrc", "include"),
          str(src),
          os.path.join(ROOT, "cimba_amd", "csrc", "host", "support.cpp"),
          "-o", exe, "-lpthread"], capture_output=True, text=True)
     assert r.returncode == 0, r.stderr[-3000:]
     return exe
+
+
+@pytest.fixture(scope="module")
+def harness(tmp_path_factory):
+    return _compile(tmp_path_factory.mktemp("pcfuzz"), [])
+
+
+@pytest.fixture(scope="module")
+def harness_spill(tmp_path_factory):
+    return _compile(tmp_path_factory.mktemp("pcfuzzsp"),
+                    ["-DCMB_FUZZ_SPILL"])
 
 
 @pytest.mark.parametrize("seed", [1, 99, 0xFEED])
@@ -200,3 +218,16 @@ def test_prodcons_fuzz(harness, seed):
     # the timeout path must actually fire across the batch
     timeouts = int(r.stdout.split("trials,")[1].split()[0])
     assert timeouts > 0
+
+
+@pytest.mark.parametrize("seed", [1, 99, 0xFEED])
+def test_prodcons_fuzz_spill_bitwise(harness, harness_spill, seed):
+    """The spill build (4-slot ring + slab under the same logical limit)
+    must reproduce the plain build's randomized schedules BITWISE:
+    physical placement of queue entries may never change semantics."""
+    a = subprocess.run([harness, str(seed), "400"], capture_output=True,
+                       text=True, timeout=600)
+    b = subprocess.run([harness_spill, str(seed), "400"],
+                       capture_output=True, text=True, timeout=600)
+    assert a.returncode == 0 and b.returncode == 0, (a.stderr, b.stderr)
+    assert a.stdout == b.stdout
