@@ -45,6 +45,15 @@ extern "C" {
 
 #define CMB_UNLIMITED INT32_C(0x7FFFFFFF)
 
+/* ---- engine envelope (defaults; rebuild libcimba with
+ * -DCIMBA_C_MAX_PROC=N etc. to change — cimba_amd/csrc/host/capi.cpp).
+ * Within a trial: up to 16384 live processes, 64 each of object queues /
+ * resources / pools / buffers / conditions, 32 priority queues, 65536
+ * pending user events.  The event heap (16K fast + 48K growth tier),
+ * object queues (8K + 24K) and priority queues (4K + 12K) grow into a
+ * spill tier automatically; a trial aborts with a clean status only on
+ * true exhaustion. ---- */
+
 typedef struct cmb_sim cmb_sim;           /* per-trial engine (opaque) */
 typedef struct cmb_process cmb_process;   /* process handle (opaque) */
 typedef struct cmb_objectqueue cmb_objectqueue;
