@@ -46,9 +46,25 @@ struct Naive {
 int main(int argc, char** argv) {
     const uint64_t seed = argc > 1 ? strtoull(argv[1], nullptr, 0) : 1;
     const uint64_t nops = argc > 2 ? strtoull(argv[2], nullptr, 0) : 20000;
+#ifdef CMB_HEAP_TIERED
+    // two-tier + back-map instantiation (the host C API shape, scaled
+    // down): an 8-entry fast tier under a 504-entry spill slab, so the
+    // fuzz crosses the boundary constantly, with the open-addressing
+    // handle map live on every op
+    constexpr int CAP = 8;
+    constexpr int SCAP = 504;
+    static EvEntry buf[CAP];
+    static EvEntry slab[SCAP];
+    static cmb::EvMapSlot mapbuf[HashHeap<CAP, SCAP, true>::MSIZE];
+    HashHeap<CAP, SCAP, true> h(buf);
+    h.map = mapbuf;
+    h.map_clear();
+    h.attach_spill(slab, SCAP);
+#else
     constexpr int CAP = 512;
     static EvEntry buf[CAP];
     HashHeap<CAP> h(buf);
+#endif
     Naive m;
     cmb::Rng r;
     r.seed(seed);
@@ -141,23 +157,43 @@ int main(int argc, char** argv) {
 """
 
 
-@pytest.fixture(scope="module")
-def harness(tmp_path_factory):
-    d = tmp_path_factory.mktemp("heapfuzz")
-    src = d / "heapfuzz.cpp"
+def _compile(d, extra):
+    src = d / f"heapfuzz{len(extra)}.cpp"
     src.write_text(HARNESS)
-    exe = str(d / "heapfuzz")
+    exe = str(d / f"heapfuzz{len(extra)}")
     r = subprocess.run(
-        ["g++", "-std=c++17", "-O2", "-g",
+        ["g++", "-std=c++17", "-O2", "-g", *extra,
          "-I", os.path.join(ROOT, "cimba_amd", "csrc", "include"),
          str(src), "-o", exe], capture_output=True, text=True)
     assert r.returncode == 0, r.stderr[-3000:]
     return exe
 
 
+@pytest.fixture(scope="module")
+def harness(tmp_path_factory):
+    return _compile(tmp_path_factory.mktemp("heapfuzz"), [])
+
+
+@pytest.fixture(scope="module")
+def harness_tiered(tmp_path_factory):
+    return _compile(tmp_path_factory.mktemp("heapfuzzt"),
+                    ["-DCMB_HEAP_TIERED"])
+
+
 @pytest.mark.parametrize("seed", [1, 2, 0xDEADBEEF, 12345, 777])
 def test_heap_fuzz(harness, seed):
     r = subprocess.run([harness, str(seed), "20000"], capture_output=True,
                        text=True, timeout=300)
+    assert r.returncode == 0, (r.stdout, r.stderr[-2000:])
+    assert "heap fuzz OK" in r.stdout
+
+
+@pytest.mark.parametrize("seed", [1, 2, 0xDEADBEEF, 12345, 777])
+def test_heap_fuzz_tiered_with_backmap(harness_tiered, seed):
+    """Same model-check over the two-tier + handle-map instantiation:
+    every push/pop/cancel/reschedule/pattern op crosses or probes the
+    fast/slab boundary and the open-addressing map simultaneously."""
+    r = subprocess.run([harness_tiered, str(seed), "20000"],
+                       capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, (r.stdout, r.stderr[-2000:])
     assert "heap fuzz OK" in r.stdout
