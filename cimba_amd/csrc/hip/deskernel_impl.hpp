@@ -191,8 +191,10 @@ __global__ __launch_bounds__(256, MINW) void lane_scratch_kernel(
 // lanes peek their next dispatch path (Engine::peek_path), poll up to 4
 // distinct candidates (__shfl from the first lane of each unseen group),
 // and ONLY the lanes on the most-popular path dispatch.  Idle minority
-// lanes cost less than serializing all paths: measured 3.93 G ev/s vs
-// 3.33 G for the ungated scratch kernel (profiles/logs/r2_ab1.log).
+// lanes cost less than serializing all paths.  Measurement history and
+// the shipped-default matrix: profiles/r02_conv_divergence.md (first
+// A/B: 3.93 vs 3.33 G ev/s, profiles/logs/r2_ab1.log; final with the
+// heap-top cache: 4.69 G and conv wins for every lane model).
 //
 // Trial refill is wave-synchronous via the stride loop — all lanes of a
 // wave start their next trial together, so the low-utilization tail is
