@@ -51,7 +51,6 @@ struct trial {
 };
 
 struct target {
-    struct world* w;
     double x, y, vx, vy, rcs;
     uint8_t tracked;
 };
@@ -158,7 +157,6 @@ static void trial_fn(cmb_sim* sim, void* exp_slot) {
 
     for (unsigned i = 0; i < N_TARGETS; ++i) {
         struct target* t = &w->tgt[i];
-        t->w = w;
         t->x = cmb_random_uniform(sim, -area_m, area_m);
         t->y = cmb_random_uniform(sim, -area_m, area_m);
         const double ang = cmb_random_uniform(sim, 0.0, 2.0 * M_PI);

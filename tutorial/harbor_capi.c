@@ -59,12 +59,10 @@ enum ship_size { SMALL = 0, LARGE = 1 };
 
 struct trial {
     double lambda;     /* in: arrivals per hour */
-    uint64_t seed;     /* in */
     double avg_in_harbor;
     double tug_util;
     uint64_t served;
     uint64_t arrived;
-    int abandoned;     /* set by the trial cleanup hook */
 };
 
 struct ship;
