@@ -19,6 +19,7 @@
 #include "../models/scenarios.hpp"
 #include "../models/spillprobe.hpp"
 
+#include <atomic>
 #include <map>
 #include <string>
 #include <vector>
@@ -531,7 +532,8 @@ static py::dict scenario_gpu(int which) {
 static py::dict scenario_run_host(int which, uint64_t ntrials, int threads) {
     Scenario::Params p{which};
     std::vector<Scenario::Result> res(ntrials);
-    int thread_inits = 0, thread_exits = 0, cleanups = 0;
+    // hooks fire on worker THREADS: the counters must be atomic
+    std::atomic<int> thread_inits{0}, thread_exits{0}, cleanups{0};
     RunHooks hooks;
     hooks.thread_init = [&](int) { ++thread_inits; };
     hooks.thread_exit = [&](int) { ++thread_exits; };
@@ -542,9 +544,9 @@ static py::dict scenario_run_host(int which, uint64_t ntrials, int threads) {
     d["trials"] = rep.trials;
     d["failed"] = rep.failed;
     d["abandoned"] = rep.abandoned;
-    d["thread_inits"] = thread_inits;
-    d["thread_exits"] = thread_exits;
-    d["cleanups"] = cleanups;
+    d["thread_inits"] = thread_inits.load();
+    d["thread_exits"] = thread_exits.load();
+    d["cleanups"] = cleanups.load();
     d["first_status"] = ntrials ? res[0].status : 0;
     return d;
 }
