@@ -134,6 +134,29 @@ def test_trial_base_world_invariance():
     assert (a["per_trial_avg"] + b["per_trial_avg"]) == full["per_trial_avg"]
 
 
+def test_trial_base_world_invariance_all_models():
+    """The same shard/full bit-equality for MG1 and JobShop (the other
+    BASELINE configs the scaling bench can run)."""
+    import cimba_amd as ca
+
+    full = ca.mg1_host(ntrials=8, num_objects=300, srv_scv=2.0, dist=2,
+                       seed=77, threads=1)
+    a = ca.mg1_host(ntrials=4, num_objects=300, srv_scv=2.0, dist=2,
+                    seed=77, threads=1, trial_base=0)
+    b = ca.mg1_host(ntrials=4, num_objects=300, srv_scv=2.0, dist=2,
+                    seed=77, threads=1, trial_base=4)
+    assert a["total_events"] + b["total_events"] == full["total_events"]
+    assert (a["per_trial_avg"] + b["per_trial_avg"]) == full["per_trial_avg"]
+
+    fullj = ca.jobshop_host(ntrials=6, entities=200, njobs=12, seed=5,
+                            threads=1)
+    aj = ca.jobshop_host(ntrials=3, entities=200, njobs=12, seed=5,
+                         threads=1, trial_base=0)
+    bj = ca.jobshop_host(ntrials=3, entities=200, njobs=12, seed=5,
+                         threads=1, trial_base=3)
+    assert aj["total_events"] + bj["total_events"] == fullj["total_events"]
+
+
 def test_bench_self_launch_host():
     """`python bench.py --gpus 2 --host` must self-launch 2 ranks (no
     external torch.distributed.run) and report the ACTUAL world size
